@@ -1,0 +1,123 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: MNIST LeNet-5 CNN synchronous data-parallel SGD.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched under `torch.distributed.run --nnodes=1 --nproc-per-node N`, one
+rank per GPU over RCCL.  W untimed warmup steps, then EXACTLY K timed steps
+bracketed by barrier + torch.cuda.synchronize on both sides; elapsed is the
+MAX over ranks; rank 0 prints ONE JSON line.
+
+Metric (BASELINE.json): images/sec whole node + p50 step time, MNIST CNN
+sync-SGD, synthetic MNIST-shaped data, random-init weights, bf16 compute on
+GPU.  Weak scaling: per-GPU batch fixed (default 1024 -> global 8192 at
+N=8, the reference's CDF-config batch, cfg/time_cdf_cfgs/*:62).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=40)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch_size", type=int, default=1024,
+                    help="per-GPU batch (weak scaling)")
+    ap.add_argument("--model", default="lenet")
+    ap.add_argument("--mode", default="full_sync")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    on_gpu = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if on_gpu else "cpu")
+    if on_gpu:
+        torch.cuda.set_device(device)
+    if world > 1:
+        dist.init_process_group("nccl" if on_gpu else "gloo",
+                                rank=rank, world_size=world)
+
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+
+    tflags = build_train_parser().parse_args([
+        "--synthetic_data", "--model", args.model,
+        "--batch_size", str(args.batch_size),
+        "--train_dir", "/tmp/dmnist_bench",
+        "--save_interval_secs", "100000",
+        "--max_steps", str(args.steps + args.warmup + 1),
+    ])
+    trainer = Trainer(tflags, device=device, rank=rank, world=world,
+                      local_rank=local_rank)
+    trainer._num_examples = 60000
+    ds = make_dataset(tflags, rank, world, trainer.device, trainer.compute_dtype)
+
+    def one_step():
+        x, y = ds.next_batch(args.batch_size)
+        trainer.train_step(x, y)
+
+    for _ in range(args.warmup):
+        one_step()
+
+    if world > 1:
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    step_t = np.zeros(args.steps)
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        ts = time.perf_counter()
+        one_step()
+        step_t[i] = time.perf_counter() - ts
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        dist.barrier()
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if world > 1 else args.gpus
+    ms_per_step = elapsed / args.steps * 1000.0
+    images_per_sec = args.batch_size * n_gpus * args.steps / elapsed
+    dtype = "bf16" if trainer.compute_dtype == torch.bfloat16 else "fp32"
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole node), MNIST CNN sync-SGD",
+            "value": round(images_per_sec, 1),
+            "unit": "images/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 4),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch_size * n_gpus,
+                "seq_len": None,
+                "parallelism": f"dp{n_gpus}",
+                "p50_ms_per_step": round(float(np.percentile(step_t, 50)) * 1000, 4),
+                "p95_ms_per_step": round(float(np.percentile(step_t, 95)) * 1000, 4),
+                "p99_ms_per_step": round(float(np.percentile(step_t, 99)) * 1000, 4),
+                "mode": args.mode,
+            },
+        }), flush=True)
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

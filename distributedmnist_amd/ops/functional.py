@@ -1,0 +1,162 @@
+"""Autograd wiring for the fused primitives, dispatching CPU-ref vs HIP ext.
+
+Dispatch rule: CUDA tensors -> the in-tree HIP extension (fails loudly if the
+.so is missing on a GPU box — _C.ext()); CPU tensors -> ops/cpu_ref.py fp32
+reference.  Both paths share autograd semantics so the engine/ and models/
+code is device-agnostic.
+
+Mixed precision: parameters are fp32 masters; on GPU the model holds bf16
+shadow copies used for compute (passed as the non-differentiable `w_comp` /
+`b_comp` args).  Weight gradients are always produced in fp32 so they can
+accumulate straight into the flat fp32 all-reduce bucket
+(parallel/sync.py).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .. import _C
+from . import cpu_ref
+
+
+def _use_hip(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+class ConvPoolFn(torch.autograd.Function):
+    """Fused conv5x5-SAME + bias + ReLU + maxpool2x2s2 (mnist.py:107-127)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, w_comp, b_comp, need_dx: bool):
+        if _use_hip(x):
+            y, amax = _C.ext().conv_pool_fwd(x, w_comp, b_comp)
+        else:
+            y, amax = cpu_ref.conv_pool_fwd(x, w_comp, b_comp)
+        ctx.save_for_backward(x, w_comp, y, amax)
+        ctx.need_dx = need_dx
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w_comp, y, amax = ctx.saved_tensors
+        if _use_hip(dy):
+            dx, dw, db = _C.ext().conv_pool_bwd(
+                dy.contiguous(), x, w_comp, y, amax, ctx.need_dx)
+        else:
+            dx, dw, db = cpu_ref.conv_pool_bwd(dy, x, w_comp, y, amax)
+        if not ctx.need_dx:
+            dx = None
+        return dx, dw, db, None, None, None
+
+
+def conv_pool(x, w, b, w_comp=None, b_comp=None, need_dx=True):
+    return ConvPoolFn.apply(x, w, b, w_comp if w_comp is not None else w,
+                            b_comp if b_comp is not None else b, need_dx)
+
+
+class LinearActFn(torch.autograd.Function):
+    """x @ W + b [+ReLU] [+TF-dropout(1-p_keep)] (mnist.py:136-145).
+
+    Dropout is folded into the epilogue; the backward recovers BOTH the relu
+    and the dropout mask from sign(y): dropped or relu-clipped positions have
+    y == 0, kept positions are scaled by 1/p_keep.
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, b, w_comp, b_comp, relu: bool, p_keep: float,
+                seed: int, offset: int):
+        if _use_hip(x):
+            y = _C.ext().linear_act_fwd(x, w_comp, b_comp, relu, p_keep,
+                                        seed, offset)
+        else:
+            y = cpu_ref.linear_fwd(x, w_comp, b_comp, relu)
+            if p_keep < 1.0:
+                gen = torch.Generator(device="cpu")
+                gen.manual_seed((seed * 0x9E3779B97F4A7C15 + offset) % (2**63))
+                mask = (torch.rand(y.shape, generator=gen) < p_keep)
+                y = y * mask.to(y.dtype) / p_keep
+        ctx.save_for_backward(x, w_comp, y)
+        ctx.relu = relu
+        ctx.p_keep = p_keep
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w_comp, y = ctx.saved_tensors
+        relu, p_keep = ctx.relu, ctx.p_keep
+        if _use_hip(dy):
+            dx, dw, db = _C.ext().linear_act_bwd(
+                dy.contiguous(), x, w_comp, y, relu, p_keep,
+                ctx.needs_input_grad[0])
+        else:
+            dyf = dy.float()
+            if relu or p_keep < 1.0:
+                dyf = dyf * (y > 0).float()
+                if p_keep < 1.0:
+                    dyf = dyf / p_keep
+            dx = (dyf @ w_comp.float().t()).to(x.dtype)
+            dw = x.float().t() @ dyf
+            db = dyf.sum(dim=0)
+        if not ctx.needs_input_grad[0]:
+            dx = None
+        return dx, dw, db, None, None, None, None, None, None
+
+
+def linear_act(x, w, b, w_comp=None, b_comp=None, relu=False, p_keep=1.0,
+               seed=0, offset=0):
+    return LinearActFn.apply(x, w, b, w_comp if w_comp is not None else w,
+                             b_comp if b_comp is not None else b, relu,
+                             float(p_keep), int(seed), int(offset))
+
+
+class SoftmaxXentFn(torch.autograd.Function):
+    """Fused mean sparse-softmax-CE + top-1-correct count (mnist.py:149-164).
+
+    Forward also produces dlogits = (softmax - onehot)/B so backward is a
+    saved-tensor multiply — one fused GPU kernel total.
+    Returns (loss, correct_count); correct_count is non-differentiable.
+    """
+
+    @staticmethod
+    def forward(ctx, logits, labels):
+        if _use_hip(logits):
+            loss, correct, dl = _C.ext().softmax_xent_fwd(logits, labels)
+        else:
+            loss, correct, dl = cpu_ref.softmax_xent_fwd(logits, labels)
+        ctx.save_for_backward(dl)
+        ctx.mark_non_differentiable(correct)
+        return loss, correct
+
+    @staticmethod
+    def backward(ctx, dloss, _dcorrect):
+        (dl,) = ctx.saved_tensors
+        return dl * dloss, None
+
+
+def softmax_xent(logits, labels):
+    return SoftmaxXentFn.apply(logits, labels)
+
+
+def sgd_step(master: torch.Tensor, grad: torch.Tensor, lr: float,
+             grad_scale: float = 1.0, drop_connect_keep=None,
+             seed: int = 0, offset: int = 0, shadow=None):
+    """Fused flat SGD apply: master -= lr*scale*(grad [* bernoulli(keep)]).
+
+    Optionally refreshes the bf16 `shadow` copy (one kernel on GPU).
+    Reference semantics: GradientDescentOptimizer + drop_connect mask with NO
+    rescale (distributed_train.py:176,414-416).
+    """
+    if master.is_cuda:
+        _C.ext().sgd_step(master, grad,
+                          shadow if shadow is not None else master,
+                          shadow is not None, float(lr), float(grad_scale),
+                          -1.0 if drop_connect_keep is None else float(drop_connect_keep),
+                          int(seed), int(offset))
+    else:
+        gen = None
+        if drop_connect_keep is not None:
+            gen = torch.Generator(device="cpu")
+            gen.manual_seed((seed * 0x9E3779B97F4A7C15 + offset) % (2**63))
+        cpu_ref.sgd_step(master, grad, lr, grad_scale,
+                         drop_connect_keep, gen, shadow)

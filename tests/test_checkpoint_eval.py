@@ -1,0 +1,80 @@
+"""Checkpoint layout + evaluator tests (SURVEY.md sections 5.4, 3.4)."""
+
+import os
+import re
+import subprocess
+import sys
+
+import torch
+
+from distributedmnist_amd.engine.supervisor import Supervisor
+from distributedmnist_amd.engine.train import Trainer, make_dataset
+from distributedmnist_amd.utils.flags import build_train_parser
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _train(tmp_path, steps=4):
+    flags = build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", str(tmp_path / "train"),
+         "--batch_size", "16", "--max_steps", str(steps), "--model", "mlp",
+         "--device", "cpu"])
+    t = Trainer(flags)
+    ds = make_dataset(flags, 0, 1, t.device, t.compute_dtype)
+    t.train(ds)
+    return flags
+
+
+def test_checkpoint_layout_tf_compatible(tmp_path):
+    flags = _train(tmp_path)
+    d = flags.train_dir
+    idx = os.path.join(d, "checkpoint")
+    assert os.path.exists(idx)
+    content = open(idx).read()
+    m = re.search(r'model_checkpoint_path: "(model\.ckpt-\d+)"', content)
+    assert m, content
+    assert os.path.exists(os.path.join(d, m.group(1)))
+    step, payload = Supervisor.restore(d)
+    assert step == 4
+    assert "flat_master" in payload and "model_state" in payload
+
+
+def test_eval_entrypoint_run_once(tmp_path):
+    flags = _train(tmp_path, steps=3)
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "src", "mnist_eval.py"),
+         "--checkpoint_dir", flags.train_dir,
+         "--eval_dir", str(tmp_path / "eval"),
+         "--run_once", "--synthetic_data", "--model", "mlp",
+         "--device", "cpu"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    # scraper contract: benchmark.py extract_times_losses_precision regex
+    m = re.search(r"Num examples: ([0-9]*)  Precision @ 1: ([\.0-9]*) "
+                  r"Loss: ([\.0-9]*) Time: ([\.0-9]*)", out.stdout)
+    assert m, out.stdout
+    assert int(m.group(1)) == 10000
+    assert 0.0 <= float(m.group(2)) <= 1.0
+    sm = re.search(r"at step=(\d+)", out.stdout)
+    assert sm and int(sm.group(1)) == 3
+
+
+def test_train_entrypoint_subprocess(tmp_path):
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "src", "mnist_distributed_train.py"),
+         "--synthetic_data", "--train_dir", str(tmp_path / "t"),
+         "--batch_size", "8", "--max_steps", "3", "--model", "mlp",
+         "--device", "cpu"],
+        capture_output=True, text=True, timeout=180)
+    assert out.returncode == 0, out.stderr
+    # per-step line matches the reference scraper (benchmark.py:31)
+    steps = re.findall(r".*step ([0-9]*),.*", out.stdout)
+    assert steps and max(int(s) for s in steps) >= 2
+    assert "loss = " in out.stdout and "examples/sec" in out.stdout
+
+
+def test_ps_role_exits(tmp_path):
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "src", "mnist_distributed_train.py"),
+         "--job_name", "ps"], capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0

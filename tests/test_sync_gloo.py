@@ -1,0 +1,154 @@
+"""Multi-process (gloo, world_size=2) tests of the sync engine — the
+distributed path is made correct-by-construction here on CPU; the same code
+runs over RCCL on the 8-GPU node (SURVEY.md section 4 item c)."""
+
+import os
+import socket
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from distributedmnist_amd.parallel.sync import SyncEngine
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _init(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+
+def _run_full_sync(rank, world, port, q):
+    _init(rank, world, port)
+    g = torch.full((10,), float(rank + 1))
+    eng = SyncEngine(g, mode="full_sync", rank=rank, world_size=world)
+    eng.step_begin(0)
+    applied, grad, contributors = eng.reduce(0, 0.01)
+    q.put((rank, applied, grad[0].item(), contributors))
+    dist.destroy_process_group()
+
+
+def _run_k_of_n(rank, world, port, q):
+    _init(rank, world, port)
+    g = torch.full((10,), float(rank + 1))
+    eng = SyncEngine(g, mode="k_of_n", replicas_to_aggregate=1,
+                     rank=rank, world_size=world)
+    eng.step_begin(0)
+    # rank 1 is the slow one -> excluded with K=1
+    applied, grad, contributors = eng.reduce(0, compute_time_s=0.01 + rank)
+    q.put((rank, applied, grad[0].item(), contributors))
+    dist.destroy_process_group()
+
+
+def _run_interval(rank, world, port, q):
+    _init(rank, world, port)
+    g = torch.full((4,), 1.0)
+    eng = SyncEngine(g, mode="interval", interval_ms=0.0,
+                     rank=rank, world_size=world)
+    # interval_ms=0 -> fires on every step; 2 ranks x grad=1 accumulated once
+    eng.step_begin(0)
+    applied, grad, contributors = eng.reduce(0, 0.01)
+    q.put((rank, applied, None if grad is None else grad[0].item(), contributors))
+    dist.destroy_process_group()
+
+
+def _run_timeout(rank, world, port, q):
+    _init(rank, world, port)
+    g = torch.full((4,), float(rank + 1))
+    eng = SyncEngine(g, mode="full_sync", straggler_timeout_ms=100.0,
+                     rank=rank, world_size=world)
+    eng.step_begin(0)
+    # rank 1 exceeds the 100ms deadline -> drops its own contribution
+    ct = 0.01 if rank == 0 else 0.5
+    applied, grad, contributors = eng.reduce(0, compute_time_s=ct)
+    q.put((rank, applied, grad[0].item(), contributors))
+    dist.destroy_process_group()
+
+
+def _spawn(fn, world=2):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = free_port()
+    procs = [ctx.Process(target=fn, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        r, *vals = q.get()
+        results[r] = vals
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+    return results
+
+
+@pytest.mark.timeout(120)
+def test_full_sync_sums_and_counts():
+    res = _spawn(_run_full_sync)
+    for r in (0, 1):
+        applied, gval, contributors = res[r]
+        assert applied and contributors == 2
+        assert gval == 3.0  # 1 + 2 summed; engine returns SUM
+
+
+@pytest.mark.timeout(120)
+def test_k_of_n_drops_slowest():
+    res = _spawn(_run_k_of_n)
+    for r in (0, 1):
+        applied, gval, contributors = res[r]
+        assert applied and contributors == 1
+        assert gval == 1.0  # only rank 0 (fast) contributed
+
+
+@pytest.mark.timeout(120)
+def test_interval_fires_and_averages():
+    res = _spawn(_run_interval)
+    for r in (0, 1):
+        applied, gval, contributors = res[r]
+        assert applied and contributors == 2
+        assert gval == 2.0  # both accumulators (1 step each) summed
+
+
+@pytest.mark.timeout(120)
+def test_straggler_timeout_drops_self():
+    res = _spawn(_run_timeout)
+    for r in (0, 1):
+        applied, gval, contributors = res[r]
+        assert applied and contributors == 1
+        assert gval == 1.0  # rank 1's grad zeroed
+
+
+def _run_trainer_e2e(rank, world, port, q):
+    _init(rank, world, port)
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+    flags = build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", f"/tmp/dmnist_e2e_test_{port}",
+         "--batch_size", "16", "--max_steps", "5", "--model", "mlp",
+         "--device", "cpu"])
+    t = Trainer(flags, rank=rank, world=world, local_rank=rank)
+    ds = make_dataset(flags, rank, world, t.device, t.compute_dtype)
+    t.train(ds)
+    q.put((rank, t.fp.flat_master.sum().item(),
+           t.fp.flat_master[:5].tolist()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_trainer_e2e_params_identical_across_ranks():
+    res = _spawn(_run_trainer_e2e)
+    assert res[0] == res[1], "ranks diverged after 5 sync steps"
