@@ -1,0 +1,235 @@
+// PyTorch bindings for the gfx950 kernel library (_dmnist_hip).
+// Every op validates dtype/contiguity and fails loudly — no silent
+// fallbacks (the GPU path must run these kernels, _C.py policy).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include "kernels.h"
+
+namespace {
+
+#define CHECK_BF16(t) TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16")
+#define CHECK_F32(t) TORCH_CHECK((t).scalar_type() == at::kFloat, #t " must be fp32")
+#define CHECK_CONTIG(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+#define CHECK_CUDA(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+
+const unsigned short* bf16_ptr(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+unsigned short* bf16_mut(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+int cdiv(int a, int b) { return (a + b - 1) / b; }
+
+int pick_splitk(int mtiles, int ntiles, int ksteps) {
+  int tiles = mtiles * ntiles;
+  if (tiles >= 256 || ksteps <= 1) return 1;
+  int want = std::min(ksteps, std::max(1, 512 / tiles));
+  return want;
+}
+
+// --------------------------------------------------------------------------
+torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                             bool relu, double p_keep, int64_t seed,
+                             int64_t offset) {
+  CHECK_CUDA(x); CHECK_BF16(x); CHECK_CONTIG(x);
+  CHECK_BF16(w); CHECK_CONTIG(w);
+  CHECK_F32(b); CHECK_CONTIG(b);
+  int M = x.size(0), K = x.size(1), N = w.size(1);
+  TORCH_CHECK(w.size(0) == K, "w/x shape mismatch");
+  auto y = torch::empty({M, N}, x.options());
+  GemmParams p{};
+  p.A = bf16_ptr(x); p.B = bf16_ptr(w);
+  p.bias = b.data_ptr<float>();
+  p.C = y.data_ptr(); p.amax = nullptr;
+  p.M = M; p.N = N; p.K = K;
+  p.lda = K; p.ldb = N; p.ldc = N;
+  p.splitk = 1;
+  p.p_keep = (float)p_keep; p.seed = (uint64_t)seed; p.offset = (uint64_t)offset;
+  bool drop = p_keep < 1.0;
+  TORCH_CHECK(!drop || relu, "dropout path requires relu epilogue");
+  bool big = cdiv(M, 128) * cdiv(N, 128) >= 128;
+  auto s = cur_stream();
+  if (drop) (big ? gemm_fwd_drop_128 : gemm_fwd_drop_64)(p, s);
+  else if (relu) (big ? gemm_fwd_relu_128 : gemm_fwd_relu_64)(p, s);
+  else (big ? gemm_fwd_bias_128 : gemm_fwd_bias_64)(p, s);
+  return y;
+}
+
+std::vector<torch::Tensor> linear_act_bwd(torch::Tensor dy, torch::Tensor x,
+                                          torch::Tensor w, torch::Tensor y,
+                                          bool relu, double p_keep,
+                                          bool need_dx) {
+  CHECK_CUDA(dy); CHECK_BF16(dy); CHECK_CONTIG(dy);
+  CHECK_BF16(x); CHECK_CONTIG(x);
+  CHECK_BF16(w); CHECK_CONTIG(w);
+  int B = x.size(0), K = x.size(1), N = w.size(1);
+  auto s = cur_stream();
+  auto db = torch::zeros({N}, x.options().dtype(at::kFloat));
+  torch::Tensor dyeff;
+  bool mask = relu || p_keep < 1.0;
+  if (mask) {
+    dyeff = torch::empty_like(dy);
+    launch_relu_drop_bwd(bf16_ptr(dy), bf16_ptr(y), bf16_mut(dyeff),
+                         db.data_ptr<float>(), B, N,
+                         (float)(1.0 / p_keep), 1, s);
+  } else {
+    dyeff = dy;
+    launch_relu_drop_bwd(bf16_ptr(dy), bf16_ptr(y), bf16_mut(dyeff),
+                         db.data_ptr<float>(), B, N, 1.0f, 0, s);
+  }
+  // dW[K,N] = x^T @ dyeff  (fp32 accumulate into the grad bucket add)
+  auto dw = torch::zeros({K, N}, x.options().dtype(at::kFloat));
+  {
+    GemmParams p{};
+    p.A = bf16_ptr(x); p.B = bf16_ptr(dyeff);
+    p.C = dw.data_ptr();
+    p.M = K; p.N = N; p.K = B;
+    p.lda = K;  // A_T: A[m][k'] = x[k'*lda + m]
+    p.ldb = N; p.ldc = N;
+    bool big = cdiv(K, 128) * cdiv(N, 128) >= 128;
+    int bm = big ? 128 : 64;
+    p.splitk = pick_splitk(cdiv(K, bm), cdiv(N, bm), cdiv(B, 32));
+    (big ? gemm_dw_128 : gemm_dw_64)(p, s);
+  }
+  torch::Tensor dx;
+  if (need_dx) {
+    dx = torch::empty({B, K}, x.options());
+    GemmParams p{};
+    p.A = bf16_ptr(dyeff); p.B = bf16_ptr(w);
+    p.C = dx.data_ptr();
+    p.M = B; p.N = K; p.K = N;
+    p.lda = N; p.ldb = N; p.ldc = K;  // B_NMAJ: Bs[n'][k'] = w[n'*N + k']
+    p.splitk = 1;
+    bool big = cdiv(B, 128) * cdiv(K, 128) >= 128;
+    (big ? gemm_dx_128 : gemm_dx_64)(p, s);
+  } else {
+    dx = torch::empty({0}, x.options());
+  }
+  return {dx, dw, db};
+}
+
+// --------------------------------------------------------------------------
+std::vector<torch::Tensor> conv_pool_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b) {
+  CHECK_CUDA(x); CHECK_BF16(x); CHECK_CONTIG(x);
+  CHECK_BF16(w); CHECK_CONTIG(w);
+  CHECK_F32(b); CHECK_CONTIG(b);
+  int NB = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  TORCH_CHECK(w.dim() == 4 && w.size(0) == 5 && w.size(1) == 5 &&
+              w.size(2) == Cin, "conv weight must be [5,5,Cin,Cout]");
+  int Cout = w.size(3);
+  int Ho = H / 2, Wo = W / 2;
+  auto y = torch::empty({NB, Ho, Wo, Cout}, x.options());
+  auto amax = torch::empty({NB, Ho, Wo, Cout}, x.options().dtype(at::kByte));
+  GemmParams p{};
+  p.A = bf16_ptr(x); p.B = bf16_ptr(w);
+  p.bias = b.data_ptr<float>();
+  p.C = y.data_ptr(); p.amax = amax.data_ptr<uint8_t>();
+  p.M = NB * Ho * Wo * 4; p.N = Cout; p.K = 25 * Cin;
+  p.ldb = Cout; p.ldc = Cout;
+  p.splitk = 1;
+  p.CB = NB; p.CH = H; p.CW = W; p.CHo = Ho; p.CWo = Wo;
+  p.Cin = Cin; p.Cout = Cout;
+  auto s = cur_stream();
+  if (Cin == 1) conv1_fwd_pool(p, s);
+  else {
+    TORCH_CHECK(Cin % 8 == 0, "conv requires Cin==1 or Cin%8==0");
+    conv_fwd_pool(p, s);
+  }
+  return {y, amax};
+}
+
+std::vector<torch::Tensor> conv_pool_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor y,
+                                         torch::Tensor amax, bool need_dx) {
+  CHECK_CUDA(dy); CHECK_BF16(dy); CHECK_CONTIG(dy);
+  CHECK_BF16(x); CHECK_CONTIG(x); CHECK_BF16(w); CHECK_CONTIG(w);
+  int NB = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int Cout = w.size(3);
+  int Ho = H / 2, Wo = W / 2;
+  auto s = cur_stream();
+  // 1) scatter pooled grad (relu-masked) back to conv-output positions
+  auto dact = torch::empty({NB, H, W, Cout}, x.options());
+  auto db = torch::zeros({Cout}, x.options().dtype(at::kFloat));
+  launch_pool_bwd_scatter(bf16_ptr(dy), bf16_ptr(y),
+                          amax.data_ptr<uint8_t>(), bf16_mut(dact),
+                          db.data_ptr<float>(), NB * Ho * Wo, Cout, H, W, Wo,
+                          s);
+  // 2) dW[(khkw,ci), co] = sum_pixels x_shift * dact  (im2col^T GEMM)
+  auto dw = torch::zeros({5, 5, Cin, Cout}, x.options().dtype(at::kFloat));
+  {
+    GemmParams p{};
+    p.A = bf16_ptr(x); p.B = bf16_ptr(dact);
+    p.C = dw.data_ptr();
+    p.M = 25 * Cin; p.N = Cout; p.K = NB * H * W;
+    p.ldb = Cout; p.ldc = Cout;
+    p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
+    p.splitk = pick_splitk(cdiv(p.M, 64), cdiv(p.N, 64), cdiv(p.K, 32));
+    conv_dw_gemm(p, s);
+  }
+  torch::Tensor dx;
+  if (need_dx) {
+    TORCH_CHECK(Cout % 32 == 0, "conv_dx requires Cout%32==0");
+    dx = torch::empty({NB, H, W, Cin}, x.options());
+    GemmParams p{};
+    p.A = bf16_ptr(dact); p.B = bf16_ptr(w);
+    p.C = dx.data_ptr();
+    p.M = NB * H * W; p.N = Cin; p.K = 25 * Cout;
+    p.ldc = Cin;
+    p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
+    p.splitk = 1;
+    conv_dx_gemm(p, s);
+  } else {
+    dx = torch::empty({0}, x.options());
+  }
+  return {dx, dw, db};
+}
+
+// --------------------------------------------------------------------------
+std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
+                                            torch::Tensor labels) {
+  CHECK_CUDA(logits); CHECK_BF16(logits); CHECK_CONTIG(logits);
+  TORCH_CHECK(labels.scalar_type() == at::kLong, "labels must be int64");
+  CHECK_CONTIG(labels);
+  int B = logits.size(0), C = logits.size(1);
+  TORCH_CHECK(C <= 16, "softmax_xent kernel supports C<=16");
+  auto dl = torch::empty_like(logits);
+  auto out = torch::zeros({2}, logits.options().dtype(at::kFloat));
+  launch_softmax_xent(bf16_ptr(logits), labels.data_ptr<long>(),
+                      bf16_mut(dl), out.data_ptr<float>(), B, C,
+                      cur_stream());
+  auto loss = out.select(0, 0);
+  auto correct = out.select(0, 1);
+  return {loss, correct, dl};
+}
+
+void sgd_step(torch::Tensor master, torch::Tensor grad, torch::Tensor shadow,
+              bool has_shadow, double lr, double scale, double dc_keep,
+              int64_t seed, int64_t offset) {
+  CHECK_CUDA(master); CHECK_F32(master); CHECK_CONTIG(master);
+  CHECK_F32(grad); CHECK_CONTIG(grad);
+  launch_sgd_step(master.data_ptr<float>(), grad.data_ptr<float>(),
+                  has_shadow ? bf16_mut(shadow) : nullptr,
+                  has_shadow ? 1 : 0, master.numel(),
+                  (float)(lr * scale), (float)dc_keep, (uint64_t)seed,
+                  (uint64_t)offset, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("linear_act_fwd", &linear_act_fwd, "fused linear+bias+relu+dropout");
+  m.def("linear_act_bwd", &linear_act_bwd, "linear backward (dx, dw, db)");
+  m.def("conv_pool_fwd", &conv_pool_fwd, "fused conv5x5+bias+relu+maxpool");
+  m.def("conv_pool_bwd", &conv_pool_bwd, "conv+pool backward (dx, dw, db)");
+  m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax-CE (+grad)");
+  m.def("sgd_step", &sgd_step, "fused flat SGD apply");
+}

@@ -1,0 +1,314 @@
+// Implicit-GEMM MFMA tile kernel for gfx950 (CDNA4) — the single compute
+// template behind every matmul-shaped op in the model:
+//   fc fwd (bias/relu/dropout epilogue), fc bwd dX / dW,
+//   conv5x5-SAME fwd fused with bias+ReLU+maxpool2x2 (pool-grouped M
+//   ordering: m = pooled_pixel*4 + window_pos, so pooling is a per-lane
+//   4-register max in the MFMA epilogue — zero cross-lane traffic),
+//   conv bwd dX and dW (im2col gathers in the LDS staging stage).
+//
+// Structure: 256 threads = 4 waves (2x2), v_mfma_f32_16x16x32_bf16 tiles,
+// A/B tiles staged to LDS as [outer][K+8] (the +8 bf16 row pad makes the
+// 16-lane ds_read_b128 fragment reads bank-conflict-free: row stride 80B,
+// (a/4)%64 distinct across the group). Single-buffered, 2 barriers/K-step
+// (v1 — see cdna_hip_programming.md §5 ladder for the pipelining headroom).
+
+#include "common.h"
+#include "kernels.h"
+
+// ---- modes ---------------------------------------------------------------
+enum AMode {
+  A_N = 0,         // A[m][k] = Asrc[m*lda + k]
+  A_T = 1,         // A[m][k] = Asrc[k*lda + m]           (fc dW: x^T)
+  A_CONV_FWD = 2,  // im2col gather, pool-grouped m, Cin % 8 == 0
+  A_CONV1_FWD = 3, // im2col gather, Cin == 1 (K = 25, one K-step)
+  A_CONV_DX = 4,   // dact gather: A[m=in pixel][k=(khkw,co)]
+  A_CONV_DW = 5,   // x gather transposed: A[m=(khkw,ci)][k=pixel]
+};
+enum BMode {
+  B_KMAJ = 0,      // Bmat[k][n] = Bsrc[k*ldb + n] (transpose-staged)
+  B_NMAJ = 1,      // Bmat[k][n] = Bsrc[n*ldb + k] (row-staged; fc dX: W rows)
+  B_CONV_DX_W = 2, // Bmat[k=(khkw,co)][n=ci] = w[((khkw)*Cin+ci)*Cout + co]
+};
+enum Epi { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_RELU = 2, EPI_BIAS_RELU_DROP = 3,
+           EPI_POOL = 4 };
+enum OutKind { OUT_BF16 = 0, OUT_F32_ATOMIC = 1 };
+
+#define NTHREADS 256
+
+// load up to 8 bf16 from p[0..vcnt), zero-fill the rest; vector fast path
+DEV short8 loadRow8(const ushort_t* p, int vcnt) {
+  short8 v;
+  if (vcnt >= 8 && ((uintptr_t)p & 15) == 0) {
+    v = *reinterpret_cast<const short8*>(p);
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = (j < vcnt) ? (short)p[j] : (short)0;
+  }
+  return v;
+}
+
+// decode pool-grouped output index m -> conv-output pixel (n, oy, ox)
+DEV bool decode_pool_m(const GemmParams& p, int m, int& n, int& oy, int& ox) {
+  int q = m >> 2, pos = m & 3;
+  int wo = q % p.CWo;
+  int t = q / p.CWo;
+  int ho = t % p.CHo;
+  n = t / p.CHo;
+  oy = ho * 2 + (pos >> 1);
+  ox = wo * 2 + (pos & 1);
+  return n < p.CB;
+}
+
+template <int BM, int BN, int AMODE>
+DEV void stageA(const GemmParams& p, ushort_t (*As)[40], int m0, int kt,
+                int kend, int tid) {
+  constexpr int BK = 32;
+  if (AMODE == A_N || AMODE == A_CONV_FWD || AMODE == A_CONV_DX) {
+    // row-contiguous sources: chunks of 8 bf16 along k
+    for (int c = tid; c < BM * (BK / 8); c += NTHREADS) {
+      int i = c / (BK / 8);
+      int kc = (c % (BK / 8)) * 8;
+      int m = m0 + i, k = kt + kc;
+      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      int vcnt = min(8, kend - k);
+      if (m < p.M && vcnt > 0) {
+        if (AMODE == A_N) {
+          v = loadRow8(p.A + (size_t)m * p.lda + k, vcnt);
+        } else if (AMODE == A_CONV_FWD) {
+          // k = khkw*Cin + ci ; chunk stays inside one khkw (Cin % 8 == 0)
+          int khkw = k / p.Cin, ci = k % p.Cin;
+          int kh = khkw / 5, kw = khkw % 5;
+          int n, oy, ox;
+          if (decode_pool_m(p, m, n, oy, ox)) {
+            int y = oy + kh - 2, x = ox + kw - 2;
+            if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+              v = loadRow8(p.A + (((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci,
+                           vcnt);
+          }
+        } else {  // A_CONV_DX: m = input pixel, k = khkw*Cout + co
+          int wi = m % p.CW, t2 = m / p.CW;
+          int hi = t2 % p.CH, n = t2 / p.CH;
+          int khkw = k / p.Cout, co = k % p.Cout;
+          int kh = khkw / 5, kw = khkw % 5;
+          int y = hi - kh + 2, x = wi - kw + 2;
+          if (n < p.CB && y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+            v = loadRow8(p.A + (((size_t)n * p.CH + y) * p.CW + x) * p.Cout + co,
+                         vcnt);
+        }
+      }
+      *reinterpret_cast<short8*>(&As[i][kc]) = v;
+    }
+  } else if (AMODE == A_T) {
+    // source rows are k-major: load 8 consecutive m, scatter-transpose
+    for (int c = tid; c < BK * (BM / 8); c += NTHREADS) {
+      int kk = c / (BM / 8);
+      int i0 = (c % (BM / 8)) * 8;
+      int k = kt + kk, m = m0 + i0;
+      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (k < kend) v = loadRow8(p.A + (size_t)k * p.lda + m, min(8, p.M - m));
+#pragma unroll
+      for (int j = 0; j < 8; ++j) As[i0 + j][kk] = v[j];
+    }
+  } else if (AMODE == A_CONV1_FWD) {
+    // Cin==1, K==25: scalar gather (tiny)
+    for (int c = tid; c < BM * BK; c += NTHREADS) {
+      int i = c / BK, kk = c % BK;
+      int m = m0 + i, k = kt + kk;
+      ushort_t v = 0;
+      int n, oy, ox;
+      if (m < p.M && k < kend && k < 25 && decode_pool_m(p, m, n, oy, ox)) {
+        int kh = k / 5, kw = k % 5;
+        int y = oy + kh - 2, x = ox + kw - 2;
+        if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+          v = p.A[((size_t)n * p.CH + y) * p.CW + x];
+      }
+      As[i][kk] = v;
+    }
+  } else {  // A_CONV_DW: m = (khkw, ci), k = conv-output pixel; scalar
+    for (int c = tid; c < BM * BK; c += NTHREADS) {
+      int i = c / BK, kk = c % BK;
+      int m = m0 + i, k = kt + kk;
+      ushort_t v = 0;
+      if (m < p.M && k < kend) {
+        int khkw = m / p.Cin, ci = m % p.Cin;
+        int kh = khkw / 5, kw = khkw % 5;
+        int w_ = k % p.CW, t2 = k / p.CW;
+        int h_ = t2 % p.CH, n = t2 / p.CH;
+        int y = h_ + kh - 2, x = w_ + kw - 2;
+        if (n < p.CB && y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+          v = p.A[(((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci];
+      }
+      As[i][kk] = v;
+    }
+  }
+}
+
+template <int BN, int BMODE>
+DEV void stageB(const GemmParams& p, ushort_t (*Bs)[40], int n0, int kt,
+                int kend, int tid) {
+  constexpr int BK = 32;
+  if (BMODE == B_KMAJ) {
+    for (int c = tid; c < BK * (BN / 8); c += NTHREADS) {
+      int kk = c / (BN / 8);
+      int j0 = (c % (BN / 8)) * 8;
+      int k = kt + kk, n = n0 + j0;
+      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (k < kend) v = loadRow8(p.B + (size_t)k * p.ldb + n, min(8, p.N - n));
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Bs[j0 + j][kk] = v[j];
+    }
+  } else if (BMODE == B_NMAJ) {
+    for (int c = tid; c < BN * (BK / 8); c += NTHREADS) {
+      int i = c / (BK / 8);
+      int kc = (c % (BK / 8)) * 8;
+      int n = n0 + i, k = kt + kc;
+      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      int vcnt = min(8, kend - k);
+      if (n < p.N && vcnt > 0)
+        v = loadRow8(p.B + (size_t)n * p.ldb + k, vcnt);
+      *reinterpret_cast<short8*>(&Bs[i][kc]) = v;
+    }
+  } else {  // B_CONV_DX_W: Bs[ci][kk] = w[((khkw)*Cin + ci)*Cout + co(k)]
+    for (int c = tid; c < BN * (BK / 8); c += NTHREADS) {
+      int i = c / (BK / 8);
+      int kc = (c % (BK / 8)) * 8;
+      int n = n0 + i, k = kt + kc;  // n = ci, k = khkw*Cout + co
+      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (n < p.N && k < kend) {
+        int khkw = k / p.Cout, co = k % p.Cout;
+        v = loadRow8(p.B + ((size_t)khkw * p.Cin + n) * p.Cout + co,
+                     min(8, kend - k));
+      }
+      *reinterpret_cast<short8*>(&Bs[i][kc]) = v;
+    }
+  }
+}
+
+template <int BM, int BN, int AMODE, int BMODE, int EPI, int OUT>
+__global__ __launch_bounds__(NTHREADS)
+void gemm_tile_kernel(GemmParams p) {
+  constexpr int BK = 32;
+  constexpr int WM = BM / 2, WN = BN / 2;
+  constexpr int MI = WM / 16, NI = WN / 16;
+  __shared__ __align__(16) ushort_t As[BM][BK + 8];
+  __shared__ __align__(16) ushort_t Bs[BN][BK + 8];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  // K-range of this split-K slice
+  int ksteps_total = (p.K + BK - 1) / BK;
+  int steps_per = (ksteps_total + p.splitk - 1) / p.splitk;
+  int kbeg = (int)blockIdx.z * steps_per * BK;
+  int kend = min(p.K, kbeg + steps_per * BK);
+
+  f32x4 acc[MI][NI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int lrow = lane & 15;
+  const int kq = (lane >> 4) * 8;
+
+  for (int kt = kbeg; kt < kend; kt += BK) {
+    stageA<BM, BN, AMODE>(p, As, m0, kt, kend, tid);
+    stageB<BN, BMODE>(p, Bs, n0, kt, kend, tid);
+    __syncthreads();
+    short8 af[MI], bf[NI];
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+      af[mi] = *reinterpret_cast<const short8*>(&As[wr * WM + mi * 16 + lrow][kq]);
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni)
+      bf[ni] = *reinterpret_cast<const short8*>(&Bs[wc * WN + ni * 16 + lrow][kq]);
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    __syncthreads();
+  }
+
+  // ---- epilogue ----------------------------------------------------------
+  const int frow = (lane >> 4) * 4;  // C/D: row=(lane>>4)*4+reg, col=lane&15
+  const int fcol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) {
+      int gc = n0 + wc * WN + ni * 16 + fcol;
+      if (gc >= p.N) continue;
+      float bias_v = (EPI != EPI_NONE && p.bias) ? p.bias[gc] : 0.f;
+      if (EPI == EPI_POOL) {
+        // rows (q*4 .. q*4+3) of this lane group are one pool window
+        int gq = (m0 + wr * WM + mi * 16 + frow) >> 2;
+        if (gq * 4 >= p.M) continue;
+        float best = -1.0f / 0.0f;
+        int barg = 0;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[mi][ni][r] + bias_v;
+          v = v > 0.f ? v : 0.f;  // conv bias + relu BEFORE pool
+          if (v > best) { best = v; barg = r; }
+        }
+        ushort_t* Cb = reinterpret_cast<ushort_t*>(p.C);
+        Cb[(size_t)gq * p.ldc + gc] = f2bf(best);
+        if (p.amax) p.amax[(size_t)gq * p.ldc + gc] = (uint8_t)barg;
+      } else {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int gr = m0 + wr * WM + mi * 16 + frow + r;
+          if (gr >= p.M) continue;
+          float v = acc[mi][ni][r];
+          if (EPI != EPI_NONE) v += bias_v;
+          if (EPI == EPI_BIAS_RELU || EPI == EPI_BIAS_RELU_DROP)
+            v = v > 0.f ? v : 0.f;
+          if (EPI == EPI_BIAS_RELU_DROP) {
+            float u = philox_uniform(p.seed, p.offset,
+                                     (uint64_t)gr * p.N + gc);
+            v = (u < p.p_keep) ? v / p.p_keep : 0.f;
+          }
+          if (OUT == OUT_BF16) {
+            reinterpret_cast<ushort_t*>(p.C)[(size_t)gr * p.ldc + gc] = f2bf(v);
+          } else {
+            atomicAdd(reinterpret_cast<float*>(p.C) + (size_t)gr * p.ldc + gc, v);
+          }
+        }
+      }
+    }
+  }
+}
+
+// ---- host-side launch table ----------------------------------------------
+#include <hip/hip_runtime_api.h>
+
+static inline int cdiv_host(int a, int b) { return (a + b - 1) / b; }
+
+// Explicit entry points used by bindings.cpp (keeps instantiations bounded).
+#define GEMM_ENTRY(name, BM, BN, AM, BMo, EPI, OUT)                          \
+  void name(const GemmParams& p, hipStream_t s) {                            \
+    dim3 grid(cdiv_host(p.M, BM), cdiv_host(p.N, BN), p.splitk);             \
+    hipLaunchKernelGGL((gemm_tile_kernel<BM, BN, AM, BMo, EPI, OUT>), grid,  \
+                       dim3(NTHREADS), 0, s, p);                             \
+  }
+
+GEMM_ENTRY(gemm_fwd_bias_128, 128, 128, A_N, B_KMAJ, EPI_BIAS, OUT_BF16)
+GEMM_ENTRY(gemm_fwd_bias_64, 64, 64, A_N, B_KMAJ, EPI_BIAS, OUT_BF16)
+GEMM_ENTRY(gemm_fwd_relu_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU, OUT_BF16)
+GEMM_ENTRY(gemm_fwd_relu_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU, OUT_BF16)
+GEMM_ENTRY(gemm_fwd_drop_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16)
+GEMM_ENTRY(gemm_fwd_drop_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16)
+GEMM_ENTRY(gemm_dx_128, 128, 128, A_N, B_NMAJ, EPI_NONE, OUT_BF16)
+GEMM_ENTRY(gemm_dx_64, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_BF16)
+GEMM_ENTRY(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC)
+GEMM_ENTRY(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC)
+GEMM_ENTRY(conv_fwd_pool, 128, 64, A_CONV_FWD, B_KMAJ, EPI_POOL, OUT_BF16)
+GEMM_ENTRY(conv1_fwd_pool, 128, 64, A_CONV1_FWD, B_KMAJ, EPI_POOL, OUT_BF16)
+GEMM_ENTRY(conv_dx_gemm, 128, 32, A_CONV_DX, B_CONV_DX_W, EPI_NONE, OUT_BF16)
+GEMM_ENTRY(conv_dw_gemm, 64, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC)

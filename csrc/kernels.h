@@ -1,0 +1,49 @@
+// Host-visible API of the HIP kernel library (implemented in *.hip).
+#pragma once
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+struct GemmParams {
+  const unsigned short* A;
+  const unsigned short* B;
+  const float* bias;
+  void* C;
+  uint8_t* amax;
+  int M, N, K;
+  int lda, ldb, ldc;
+  int splitk;
+  int CB, CH, CW, CHo, CWo, Cin, Cout;
+  float p_keep;
+  uint64_t seed, offset;
+};
+
+// gemm_tile.hip — implicit-GEMM MFMA entry points
+void gemm_fwd_bias_128(const GemmParams&, hipStream_t);
+void gemm_fwd_bias_64(const GemmParams&, hipStream_t);
+void gemm_fwd_relu_128(const GemmParams&, hipStream_t);
+void gemm_fwd_relu_64(const GemmParams&, hipStream_t);
+void gemm_fwd_drop_128(const GemmParams&, hipStream_t);
+void gemm_fwd_drop_64(const GemmParams&, hipStream_t);
+void gemm_dx_128(const GemmParams&, hipStream_t);
+void gemm_dx_64(const GemmParams&, hipStream_t);
+void gemm_dw_128(const GemmParams&, hipStream_t);
+void gemm_dw_64(const GemmParams&, hipStream_t);
+void conv_fwd_pool(const GemmParams&, hipStream_t);
+void conv1_fwd_pool(const GemmParams&, hipStream_t);
+void conv_dx_gemm(const GemmParams&, hipStream_t);
+void conv_dw_gemm(const GemmParams&, hipStream_t);
+
+// ops_misc.hip — host wrappers
+void launch_relu_drop_bwd(const unsigned short* dy, const unsigned short* y,
+                          unsigned short* dyeff, float* db, int B, int N,
+                          float inv_keep, int apply_mask, hipStream_t);
+void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
+                             const uint8_t* amax, unsigned short* dact,
+                             float* db, int Mpool, int C, int H, int W, int Wo,
+                             hipStream_t);
+void launch_softmax_xent(const unsigned short* logits, const long* labels,
+                         unsigned short* dlogits, float* out, int B, int C,
+                         hipStream_t);
+void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
+                     int has_shadow, long n, float lr_scale, float dc_keep,
+                     uint64_t seed, uint64_t offset, hipStream_t);

@@ -1,0 +1,221 @@
+// Non-GEMM fused kernels: relu/dropout gradient mask (+ bias-grad column
+// sums), maxpool backward scatter (+ conv bias grad), fused softmax-CE
+// (+top-1 correct count), and the fused flat SGD apply (+drop-connect mask,
+// + bf16 shadow refresh).  All memory-bound: bf16 payloads, grid-stride.
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// dyeff[b,n] = dy * (y>0) / p_keep ; db[n] += column sums (fp32)
+// Recovers BOTH the relu and the dropout mask from sign(y)
+// (ops/functional.py LinearActFn docstring).
+// Block: 256 threads as 4 rows x 64 cols; grid (col_groups, row_slices).
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256)
+void relu_drop_bwd_kernel(const ushort_t* dy, const ushort_t* y,
+                          ushort_t* dyeff, float* db, int B, int N,
+                          float inv_keep, int apply_mask, int rows_per_slice) {
+  __shared__ float partial[64];
+  int cg = blockIdx.x;           // 64-column group
+  int rs = blockIdx.y;           // row slice
+  int c = (threadIdx.x & 63);
+  int rlane = threadIdx.x >> 6;  // 0..3
+  int n = cg * 64 + c;
+  float sum = 0.f;
+  if (n < N) {
+    int r0 = rs * rows_per_slice;
+    int r1 = min(B, r0 + rows_per_slice);
+    for (int r = r0 + rlane; r < r1; r += 4) {
+      size_t idx = (size_t)r * N + n;
+      float g = bf2f(dy[idx]);
+      if (apply_mask) g = (bf2f(y[idx]) > 0.f) ? g * inv_keep : 0.f;
+      dyeff[idx] = f2bf(g);
+      sum += g;
+    }
+  }
+  // reduce 4 row-lanes per column through LDS
+  if (rlane == 0) partial[c] = 0.f;
+  __syncthreads();
+  atomicAdd(&partial[c], sum);
+  __syncthreads();
+  if (rlane == 0 && n < N && db) atomicAdd(&db[n], partial[c]);
+}
+
+// ---------------------------------------------------------------------------
+// Maxpool 2x2 backward scatter: route dy (masked by relu: pooled y > 0)
+// to the argmax position of each window; other 3 positions zero.
+// Also accumulates db[c] (conv bias grad = sum over dact).
+// One thread per pooled element (q, c); dact is [B, H, W, C] bf16.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256)
+void pool_bwd_scatter_kernel(const ushort_t* dy, const ushort_t* y,
+                             const uint8_t* amax, ushort_t* dact, float* db,
+                             int Mpool, int C, int H, int W, int Wo) {
+  __shared__ float partial[64];
+  if ((threadIdx.x >> 6) == 0) partial[threadIdx.x & 63] = 0.f;
+  __syncthreads();
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  float g = 0.f;
+  if (i < (size_t)Mpool * C) {
+    int c = i % C;
+    size_t q = i / C;
+    int wo = q % Wo;
+    size_t t = q / Wo;
+    int ho = t % (H / 2);
+    int n = t / (H / 2);
+    g = bf2f(dy[i]);
+    if (!(bf2f(y[i]) > 0.f)) g = 0.f;
+    int pos = amax[i];
+    ushort_t gb = f2bf(g);
+#pragma unroll
+    for (int r = 0; r < 2; ++r)
+#pragma unroll
+      for (int cx = 0; cx < 2; ++cx) {
+        size_t o = (((size_t)n * H + ho * 2 + r) * W + wo * 2 + cx) * C + c;
+        dact[o] = (r * 2 + cx == pos) ? gb : (ushort_t)0;
+      }
+    atomicAdd(&partial[c & 63], g);
+  }
+  __syncthreads();
+  if (db && (threadIdx.x >> 6) == 0) {
+    int c = threadIdx.x & 63;
+    // C is 32 or 64; for C==32 lanes 32..63 alias columns 0..31 of partial
+    if (C >= 64) {
+      atomicAdd(&db[c], partial[c]);
+    } else if (c < C) {
+      atomicAdd(&db[c], partial[c] + partial[c + 32 < 64 ? c + 32 : c]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused softmax cross-entropy fwd(+grad) + top-1 correct count.
+// One thread per row (C <= 16); out[0] += sum(loss)/B ; out[1] += correct.
+// dlogits = (softmax - onehot)/B, bf16.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256)
+void softmax_xent_kernel(const ushort_t* logits, const long* labels,
+                         ushort_t* dlogits, float* out, int B, int C) {
+  __shared__ float red[2];
+  if (threadIdx.x == 0) { red[0] = 0.f; red[1] = 0.f; }
+  __syncthreads();
+  int b = blockIdx.x * blockDim.x + threadIdx.x;
+  float loss = 0.f, correct = 0.f;
+  if (b < B) {
+    float v[16];
+    float mx = -1e30f;
+    int arg = 0;
+    for (int c = 0; c < C; ++c) {
+      v[c] = bf2f(logits[(size_t)b * C + c]);
+      if (v[c] > mx) { mx = v[c]; arg = c; }
+    }
+    float se = 0.f;
+    for (int c = 0; c < C; ++c) { v[c] = __expf(v[c] - mx); se += v[c]; }
+    float inv_se = 1.f / se;
+    long lab = labels[b];
+    float invB = 1.f / (float)B;
+    for (int c = 0; c < C; ++c) {
+      float p = v[c] * inv_se;
+      float d = (p - (c == (int)lab ? 1.f : 0.f)) * invB;
+      dlogits[(size_t)b * C + c] = f2bf(d);
+    }
+    loss = -(__logf(v[(int)lab] * inv_se)) * invB;
+    correct = (arg == (int)lab) ? 1.f : 0.f;
+  }
+  atomicAdd(&red[0], loss);
+  atomicAdd(&red[1], correct);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    atomicAdd(&out[0], red[0]);
+    atomicAdd(&out[1], red[1]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused flat SGD apply: master -= lr*scale*(g [* bernoulli(keep)]);
+// shadow = bf16(master).  Drop-connect = reference distributed_train.py:414
+// (mask, NO rescale).  Vectorized float4 path + scalar tail.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256)
+void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
+                     int has_shadow, long n, float lr_scale, float dc_keep,
+                     uint64_t seed, uint64_t offset) {
+  long i = (long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i * 4 < n; i += stride) {
+    long base = i * 4;
+    if (base + 3 < n) {
+      float4 g = *reinterpret_cast<const float4*>(grad + base);
+      float4 m = *reinterpret_cast<const float4*>(master + base);
+      if (dc_keep > 0.f) {
+        Philox4 ph = philox4x32(seed, offset, (uint64_t)i);
+        const float s = 1.0f / 4294967296.0f;
+        g.x *= (ph.x * s < dc_keep) ? 1.f : 0.f;
+        g.y *= (ph.y * s < dc_keep) ? 1.f : 0.f;
+        g.z *= (ph.z * s < dc_keep) ? 1.f : 0.f;
+        g.w *= (ph.w * s < dc_keep) ? 1.f : 0.f;
+      }
+      m.x -= lr_scale * g.x; m.y -= lr_scale * g.y;
+      m.z -= lr_scale * g.z; m.w -= lr_scale * g.w;
+      *reinterpret_cast<float4*>(master + base) = m;
+      if (has_shadow) {
+        shadow[base] = f2bf(m.x); shadow[base + 1] = f2bf(m.y);
+        shadow[base + 2] = f2bf(m.z); shadow[base + 3] = f2bf(m.w);
+      }
+    } else {
+      Philox4 ph = philox4x32(seed, offset, (uint64_t)i);
+      const float s = 1.0f / 4294967296.0f;
+      float u[4] = {ph.x * s, ph.y * s, ph.z * s, ph.w * s};
+      for (long j = base; j < n; ++j) {
+        float g = grad[j];
+        if (dc_keep > 0.f) g *= (u[j - base] < dc_keep) ? 1.f : 0.f;
+        float m = master[j] - lr_scale * g;
+        master[j] = m;
+        if (has_shadow) shadow[j] = f2bf(m);
+      }
+    }
+  }
+}
+
+// ---- host wrappers --------------------------------------------------------
+#include "kernels.h"
+
+static inline int cdivh(long a, long b) { return (int)((a + b - 1) / b); }
+
+void launch_relu_drop_bwd(const unsigned short* dy, const unsigned short* y,
+                          unsigned short* dyeff, float* db, int B, int N,
+                          float inv_keep, int apply_mask, hipStream_t s) {
+  int rows_per_slice = 256;
+  dim3 grid(cdivh(N, 64), cdivh(B, rows_per_slice));
+  hipLaunchKernelGGL(relu_drop_bwd_kernel, grid, dim3(256), 0, s, dy, y,
+                     dyeff, db, B, N, inv_keep, apply_mask, rows_per_slice);
+}
+
+void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
+                             const uint8_t* amax, unsigned short* dact,
+                             float* db, int Mpool, int C, int H, int W, int Wo,
+                             hipStream_t s) {
+  long total = (long)Mpool * C;
+  dim3 grid(cdivh(total, 256));
+  hipLaunchKernelGGL(pool_bwd_scatter_kernel, grid, dim3(256), 0, s, dy, y,
+                     amax, dact, db, Mpool, C, H, W, Wo);
+}
+
+void launch_softmax_xent(const unsigned short* logits, const long* labels,
+                         unsigned short* dlogits, float* out, int B, int C,
+                         hipStream_t s) {
+  dim3 grid(cdivh(B, 256));
+  hipLaunchKernelGGL(softmax_xent_kernel, grid, dim3(256), 0, s, logits,
+                     labels, dlogits, out, B, C);
+}
+
+void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
+                     int has_shadow, long n, float lr_scale, float dc_keep,
+                     uint64_t seed, uint64_t offset, hipStream_t s) {
+  long groups = (n + 3) / 4;
+  int blocks = (int)min((long)2048, (groups + 255) / 256);
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(sgd_step_kernel, dim3(blocks), dim3(256), 0, s, master,
+                     grad, shadow, has_shadow, n, lr_scale, dc_keep, seed,
+                     offset);
+}

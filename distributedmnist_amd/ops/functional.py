@@ -131,7 +131,7 @@ class SoftmaxXentFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss, _dcorrect):
         (dl,) = ctx.saved_tensors
-        return dl * dloss, None
+        return (dl * dloss).to(dl.dtype), None
 
 
 def softmax_xent(logits, labels):
