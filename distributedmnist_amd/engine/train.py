@@ -106,6 +106,7 @@ class Trainer:
         self.is_chief = rank == 0
         self.step = 0
         self.num_contributors = world
+        self._num_examples = 60000  # overwritten by train(); LR-schedule default
         # parameter-init parity across ranks (broadcast once)
         if world > 1 and dist.is_initialized():
             dist.broadcast(self.fp.flat_master, src=0)
