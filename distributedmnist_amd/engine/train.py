@@ -167,12 +167,35 @@ class Trainer:
                 log.info("Restored checkpoint at step %d", step0)
         time_acc_list = []
         begin_time = time.time()
+        writer = None
+        if flags.should_summarize and self.is_chief:
+            try:
+                from torch.utils.tensorboard import SummaryWriter
+                writer = SummaryWriter(flags.train_dir)
+            except Exception:
+                log.info("tensorboard unavailable; --should_summarize ignored")
+        next_summary_time = time.time() + flags.save_summaries_secs
         while self.step < max_steps:
             start_time = time.time()
             images, labels = dataset.next_batch(flags.batch_size)
             images, labels = self.to_device(images, labels)
-            applied, loss_v, acc_v, _ct = self.train_step(images, labels)
-            finish_time = time.time()
+            if flags.timeline_logging and 5 <= self.step <= 7:
+                # chrome trace per step (reference FULL_TRACE timelines,
+                # distributed_train.py:317-358) via torch.profiler/kineto
+                from torch.profiler import ProfilerActivity, profile
+                acts = [ProfilerActivity.CPU]
+                if self.device.type == "cuda":
+                    acts.append(ProfilerActivity.CUDA)
+                with profile(activities=acts) as prof:
+                    applied, loss_v, acc_v, _ct = self.train_step(images, labels)
+                os.makedirs(flags.train_dir, exist_ok=True)
+                prof.export_chrome_trace(os.path.join(
+                    flags.train_dir,
+                    f"worker={self.rank}_timeline_iter={self.step}.json"))
+                finish_time = time.time()
+            else:
+                applied, loss_v, acc_v, _ct = self.train_step(images, labels)
+                finish_time = time.time()
             duration = finish_time - start_time
             examples_per_sec = flags.batch_size / duration
             # per-step line: scraper contract (benchmark.py:31 'step (\d+),')
@@ -189,7 +212,14 @@ class Trainer:
                     np.save(path, np.array(time_acc_list, dtype=np.float64))
                 except OSError:
                     pass
+            if writer is not None and time.time() > next_summary_time:
+                writer.add_scalar("Train Loss", loss_v, self.step)
+                writer.add_scalar("Train Accuracy", acc_v, self.step)
+                writer.add_scalar("Examples/sec", examples_per_sec, self.step)
+                next_summary_time += flags.save_summaries_secs
             sv.maybe_save(self.step, self.checkpoint_payload())
+        if writer is not None:
+            writer.close()
         if self.is_chief:
             log.info("Elapsed Time: %f", time.time() - begin_time)
             sv.save(self.step, self.checkpoint_payload())
