@@ -63,16 +63,21 @@ torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
   return y;
 }
 
-std::vector<torch::Tensor> linear_act_bwd(torch::Tensor dy, torch::Tensor x,
-                                          torch::Tensor w, torch::Tensor y,
-                                          bool relu, double p_keep,
-                                          bool need_dx) {
+// core backward; when dw/db are passed they are accumulated into
+// (pre-zeroed fp32 bucket views — the direct-grad path); else allocated.
+std::vector<torch::Tensor> linear_act_bwd_impl(torch::Tensor dy, torch::Tensor x,
+                                               torch::Tensor w, torch::Tensor y,
+                                               bool relu, double p_keep,
+                                               bool need_dx,
+                                               torch::Tensor dw, torch::Tensor db) {
   CHECK_CUDA(dy); CHECK_BF16(dy); CHECK_CONTIG(dy);
   CHECK_BF16(x); CHECK_CONTIG(x);
   CHECK_BF16(w); CHECK_CONTIG(w);
   int B = x.size(0), K = x.size(1), N = w.size(1);
   auto s = cur_stream();
-  auto db = torch::zeros({N}, x.options().dtype(at::kFloat));
+  if (!db.defined())
+    db = torch::zeros({N}, x.options().dtype(at::kFloat));
+  CHECK_F32(db); CHECK_CONTIG(db);
   torch::Tensor dyeff;
   bool mask = relu || p_keep < 1.0;
   if (mask) {
@@ -85,8 +90,10 @@ std::vector<torch::Tensor> linear_act_bwd(torch::Tensor dy, torch::Tensor x,
     launch_relu_drop_bwd(bf16_ptr(dy), bf16_ptr(y), bf16_mut(dyeff),
                          db.data_ptr<float>(), B, N, 1.0f, 0, s);
   }
-  // dW[K,N] = x^T @ dyeff  (fp32 accumulate into the grad bucket add)
-  auto dw = torch::zeros({K, N}, x.options().dtype(at::kFloat));
+  // dW[K,N] = x^T @ dyeff  (fp32 split-K atomics)
+  if (!dw.defined())
+    dw = torch::zeros({K, N}, x.options().dtype(at::kFloat));
+  CHECK_F32(dw); CHECK_CONTIG(dw);
   {
     GemmParams p{};
     p.A = bf16_ptr(x); p.B = bf16_ptr(dyeff);
@@ -114,6 +121,23 @@ std::vector<torch::Tensor> linear_act_bwd(torch::Tensor dy, torch::Tensor x,
     dx = torch::empty({0}, x.options());
   }
   return {dx, dw, db};
+}
+
+std::vector<torch::Tensor> linear_act_bwd(torch::Tensor dy, torch::Tensor x,
+                                          torch::Tensor w, torch::Tensor y,
+                                          bool relu, double p_keep,
+                                          bool need_dx) {
+  return linear_act_bwd_impl(dy, x, w, y, relu, p_keep, need_dx,
+                             torch::Tensor(), torch::Tensor());
+}
+
+torch::Tensor linear_act_bwd_into(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor w, torch::Tensor y,
+                                  bool relu, double p_keep, bool need_dx,
+                                  torch::Tensor dw_out, torch::Tensor db_out) {
+  auto r = linear_act_bwd_impl(dy, x, w, y, relu, p_keep, need_dx,
+                               dw_out.view({x.size(1), w.size(1)}), db_out);
+  return r[0];
 }
 
 // --------------------------------------------------------------------------
@@ -147,9 +171,10 @@ std::vector<torch::Tensor> conv_pool_fwd(torch::Tensor x, torch::Tensor w,
   return {y, amax};
 }
 
-std::vector<torch::Tensor> conv_pool_bwd(torch::Tensor dy, torch::Tensor x,
-                                         torch::Tensor w, torch::Tensor y,
-                                         torch::Tensor amax, bool need_dx) {
+std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
+                                              torch::Tensor w, torch::Tensor y,
+                                              torch::Tensor amax, bool need_dx,
+                                              torch::Tensor dw, torch::Tensor db) {
   CHECK_CUDA(dy); CHECK_BF16(dy); CHECK_CONTIG(dy);
   CHECK_BF16(x); CHECK_CONTIG(x); CHECK_BF16(w); CHECK_CONTIG(w);
   int NB = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
@@ -158,13 +183,17 @@ std::vector<torch::Tensor> conv_pool_bwd(torch::Tensor dy, torch::Tensor x,
   auto s = cur_stream();
   // 1) scatter pooled grad (relu-masked) back to conv-output positions
   auto dact = torch::empty({NB, H, W, Cout}, x.options());
-  auto db = torch::zeros({Cout}, x.options().dtype(at::kFloat));
+  if (!db.defined())
+    db = torch::zeros({Cout}, x.options().dtype(at::kFloat));
+  CHECK_F32(db); CHECK_CONTIG(db);
   launch_pool_bwd_scatter(bf16_ptr(dy), bf16_ptr(y),
                           amax.data_ptr<uint8_t>(), bf16_mut(dact),
                           db.data_ptr<float>(), NB * Ho * Wo, Cout, H, W, Wo,
                           s);
   // 2) dW[(khkw,ci), co] = sum_pixels x_shift * dact  (im2col^T GEMM)
-  auto dw = torch::zeros({5, 5, Cin, Cout}, x.options().dtype(at::kFloat));
+  if (!dw.defined())
+    dw = torch::zeros({5, 5, Cin, Cout}, x.options().dtype(at::kFloat));
+  CHECK_F32(dw); CHECK_CONTIG(dw);
   {
     GemmParams p{};
     p.A = bf16_ptr(x); p.B = bf16_ptr(dact);
@@ -191,6 +220,22 @@ std::vector<torch::Tensor> conv_pool_bwd(torch::Tensor dy, torch::Tensor x,
     dx = torch::empty({0}, x.options());
   }
   return {dx, dw, db};
+}
+
+std::vector<torch::Tensor> conv_pool_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor y,
+                                         torch::Tensor amax, bool need_dx) {
+  return conv_pool_bwd_impl(dy, x, w, y, amax, need_dx, torch::Tensor(),
+                            torch::Tensor());
+}
+
+torch::Tensor conv_pool_bwd_into(torch::Tensor dy, torch::Tensor x,
+                                 torch::Tensor w, torch::Tensor y,
+                                 torch::Tensor amax, bool need_dx,
+                                 torch::Tensor dw_out, torch::Tensor db_out) {
+  auto r = conv_pool_bwd_impl(dy, x, w, y, amax, need_dx,
+                              dw_out.view_as(w), db_out);
+  return r[0];
 }
 
 // --------------------------------------------------------------------------
@@ -228,8 +273,12 @@ void sgd_step(torch::Tensor master, torch::Tensor grad, torch::Tensor shadow,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_act_fwd", &linear_act_fwd, "fused linear+bias+relu+dropout");
   m.def("linear_act_bwd", &linear_act_bwd, "linear backward (dx, dw, db)");
+  m.def("linear_act_bwd_into", &linear_act_bwd_into,
+        "linear backward accumulating dw/db into bucket views");
   m.def("conv_pool_fwd", &conv_pool_fwd, "fused conv5x5+bias+relu+maxpool");
   m.def("conv_pool_bwd", &conv_pool_bwd, "conv+pool backward (dx, dw, db)");
+  m.def("conv_pool_bwd_into", &conv_pool_bwd_into,
+        "conv backward accumulating dw/db into bucket views");
   m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax-CE (+grad)");
   m.def("sgd_step", &sgd_step, "fused flat SGD apply");
 }

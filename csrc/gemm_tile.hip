@@ -124,21 +124,44 @@ DEV void stageA(const GemmParams& p, ushort_t (*As)[40], int m0, int kt,
       }
       As[i][kk] = v;
     }
-  } else {  // A_CONV_DW: m = (khkw, ci), k = conv-output pixel; scalar
-    for (int c = tid; c < BM * BK; c += NTHREADS) {
-      int i = c / BK, kk = c % BK;
-      int m = m0 + i, k = kt + kk;
-      ushort_t v = 0;
-      if (m < p.M && k < kend) {
-        int khkw = m / p.Cin, ci = m % p.Cin;
-        int kh = khkw / 5, kw = khkw % 5;
-        int w_ = k % p.CW, t2 = k / p.CW;
-        int h_ = t2 % p.CH, n = t2 / p.CH;
-        int y = h_ + kh - 2, x = w_ + kw - 2;
-        if (n < p.CB && y >= 0 && y < p.CH && x >= 0 && x < p.CW)
-          v = p.A[(((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci];
+  } else {  // A_CONV_DW: m = (khkw, ci), k = conv-output pixel
+    if (p.Cin % 8 == 0) {
+      // 8 consecutive m stay inside one khkw (ci0 aligned to 8):
+      // one 16B load of x[pixel][ci0..ci0+7], transpose-scatter to LDS
+      for (int c = tid; c < BK * (BM / 8); c += NTHREADS) {
+        int kk = c / (BM / 8);
+        int i0 = (c % (BM / 8)) * 8;
+        int k = kt + kk, m = m0 + i0;
+        short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        if (m < p.M && k < kend) {
+          int khkw = m / p.Cin, ci = m % p.Cin;
+          int kh = khkw / 5, kw = khkw % 5;
+          int w_ = k % p.CW, t2 = k / p.CW;
+          int h_ = t2 % p.CH, n = t2 / p.CH;
+          int y = h_ + kh - 2, x = w_ + kw - 2;
+          if (n < p.CB && y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+            v = loadRow8(p.A + (((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci,
+                         min(8, p.M - m));
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) As[i0 + j][kk] = v[j];
       }
-      As[i][kk] = v;
+    } else {  // Cin == 1 (conv1): scalar, tiny M
+      for (int c = tid; c < BM * BK; c += NTHREADS) {
+        int i = c / BK, kk = c % BK;
+        int m = m0 + i, k = kt + kk;
+        ushort_t v = 0;
+        if (m < p.M && k < kend) {
+          int khkw = m / p.Cin, ci = m % p.Cin;
+          int kh = khkw / 5, kw = khkw % 5;
+          int w_ = k % p.CW, t2 = k / p.CW;
+          int h_ = t2 % p.CH, n = t2 / p.CH;
+          int y = h_ + kh - 2, x = w_ + kw - 2;
+          if (n < p.CB && y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+            v = p.A[(((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci];
+        }
+        As[i][kk] = v;
+      }
     }
   }
 }

@@ -51,19 +51,26 @@ extern "C" __global__ __launch_bounds__(256)
 void pool_bwd_scatter_kernel(const ushort_t* dy, const ushort_t* y,
                              const uint8_t* amax, ushort_t* dact, float* db,
                              int Mpool, int C, int H, int W, int Wo) {
+  // fixed-size grid, grid-stride; per-column fp32 partials accumulated in
+  // LDS across the whole loop, ONE global atomic per column per block
+  // (a naive per-element global atomicAdd on db was 1.6M atomics on 64
+  // addresses — 300+ us; this form is ~64 atomics per block).
   __shared__ float partial[64];
-  if ((threadIdx.x >> 6) == 0) partial[threadIdx.x & 63] = 0.f;
+  if (threadIdx.x < 64) partial[threadIdx.x] = 0.f;
   __syncthreads();
-  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-  float g = 0.f;
-  if (i < (size_t)Mpool * C) {
+  size_t total = (size_t)Mpool * C;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  float local = 0.f;
+  int mycol = -1;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
     int c = i % C;
     size_t q = i / C;
     int wo = q % Wo;
     size_t t = q / Wo;
     int ho = t % (H / 2);
     int n = t / (H / 2);
-    g = bf2f(dy[i]);
+    float g = bf2f(dy[i]);
     if (!(bf2f(y[i]) > 0.f)) g = 0.f;
     int pos = amax[i];
     ushort_t gb = f2bf(g);
@@ -74,16 +81,20 @@ void pool_bwd_scatter_kernel(const ushort_t* dy, const ushort_t* y,
         size_t o = (((size_t)n * H + ho * 2 + r) * W + wo * 2 + cx) * C + c;
         dact[o] = (r * 2 + cx == pos) ? gb : (ushort_t)0;
       }
-    atomicAdd(&partial[c & 63], g);
+    // stride is a multiple of C in practice (C | 256*grid): column per
+    // thread is FIXED across iterations; accumulate in a register
+    if (mycol < 0) mycol = c;
+    local += g;
   }
+  if (mycol >= 0) atomicAdd(&partial[mycol & 63], local);
   __syncthreads();
-  if (db && (threadIdx.x >> 6) == 0) {
-    int c = threadIdx.x & 63;
-    // C is 32 or 64; for C==32 lanes 32..63 alias columns 0..31 of partial
+  if (db && threadIdx.x < 64) {
+    float v = partial[threadIdx.x];
     if (C >= 64) {
-      atomicAdd(&db[c], partial[c]);
-    } else if (c < C) {
-      atomicAdd(&db[c], partial[c] + partial[c + 32 < 64 ? c + 32 : c]);
+      if (v != 0.f) atomicAdd(&db[threadIdx.x], v);
+    } else if ((int)threadIdx.x < C) {
+      float v2 = partial[threadIdx.x + 32];
+      if (v + v2 != 0.f) atomicAdd(&db[threadIdx.x], v + v2);
     }
   }
 }
@@ -185,7 +196,11 @@ static inline int cdivh(long a, long b) { return (int)((a + b - 1) / b); }
 void launch_relu_drop_bwd(const unsigned short* dy, const unsigned short* y,
                           unsigned short* dyeff, float* db, int B, int N,
                           float inv_keep, int apply_mask, hipStream_t s) {
-  int rows_per_slice = 256;
+  // enough row-slices to fill the chip (256 CUs); 64 atomics per block on db
+  int rows_per_slice = 32;
+  while ((long)cdivh(N, 64) * cdivh(B, rows_per_slice) > 4096 &&
+         rows_per_slice < B)
+    rows_per_slice *= 2;
   dim3 grid(cdivh(N, 64), cdivh(B, rows_per_slice));
   hipLaunchKernelGGL(relu_drop_bwd_kernel, grid, dim3(256), 0, s, dy, y,
                      dyeff, db, B, N, inv_keep, apply_mask, rows_per_slice);
@@ -196,9 +211,10 @@ void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
                              float* db, int Mpool, int C, int H, int W, int Wo,
                              hipStream_t s) {
   long total = (long)Mpool * C;
-  dim3 grid(cdivh(total, 256));
-  hipLaunchKernelGGL(pool_bwd_scatter_kernel, grid, dim3(256), 0, s, dy, y,
-                     amax, dact, db, Mpool, C, H, W, Wo);
+  int blocks = cdivh(total, 256);
+  if (blocks > 2048) blocks = 2048;  // grid-stride; bounds db atomics
+  hipLaunchKernelGGL(pool_bwd_scatter_kernel, dim3(blocks), dim3(256), 0, s,
+                     dy, y, amax, dact, db, Mpool, C, H, W, Wo);
 }
 
 void launch_softmax_xent(const unsigned short* logits, const long* labels,

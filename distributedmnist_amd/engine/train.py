@@ -123,12 +123,25 @@ class Trainer:
         labels = labels.to(device=self.device, non_blocking=True)
         return images, labels
 
+    @property
+    def _needs_step_timing(self) -> bool:
+        """Host-synchronizing per-step timers are only needed for the
+        instrumented/straggler modes; the full-sync hot path stays async."""
+        return (self.mode in ("k_of_n", "cdf")
+                or self.engine.timeout_s is not None)
+
     def train_step(self, images, labels):
-        """One synchronous step. Returns (applied, loss, acc, compute_time)."""
+        """One synchronous step.
+
+        Returns (applied, loss, acc, compute_time); loss/acc are 0-dim
+        DEVICE tensors on the hot path (no forced host sync — call float()
+        only when logging)."""
         flags = self.flags
         self.model.set_step(self.step)
         self.engine.step_begin(self.step)
-        self.timer.start()
+        timed = self._needs_step_timing
+        if timed:
+            self.timer.start()
         if flags.inject_slow_rank == self.rank and flags.inject_slow_ms > 0:
             time.sleep(flags.inject_slow_ms / 1000.0)
         self.fp.zero_grad()
@@ -136,7 +149,7 @@ class Trainer:
         loss, acc = self.model.loss_and_accuracy(logits, labels)
         loss.backward()
         self.fp.fix_grad_views()
-        compute_time = self.timer.stop()
+        compute_time = self.timer.stop() if timed else 0.0
         applied, grad, contributors = self.engine.reduce(self.step, compute_time)
         if applied:
             lr = lr_at(self.step, flags, self._num_examples,
@@ -149,7 +162,7 @@ class Trainer:
                         shadow=self.fp.flat_shadow)
             self.num_contributors = contributors
         self.step += 1
-        return applied, float(loss.detach().float()), float(acc.detach().float()), compute_time
+        return applied, loss.detach(), acc.detach(), compute_time
 
     # ------------------------------------------------------------------
     def train(self, dataset, max_steps=None):
@@ -188,6 +201,7 @@ class Trainer:
                     acts.append(ProfilerActivity.CUDA)
                 with profile(activities=acts) as prof:
                     applied, loss_v, acc_v, _ct = self.train_step(images, labels)
+                torch.cuda.synchronize() if self.device.type == "cuda" else None
                 os.makedirs(flags.train_dir, exist_ok=True)
                 prof.export_chrome_trace(os.path.join(
                     flags.train_dir,
@@ -196,6 +210,7 @@ class Trainer:
             else:
                 applied, loss_v, acc_v, _ct = self.train_step(images, labels)
                 finish_time = time.time()
+            loss_v, acc_v = float(loss_v), float(acc_v)
             duration = finish_time - start_time
             examples_per_sec = flags.batch_size / duration
             # per-step line: scraper contract (benchmark.py:31 'step (\d+),')

@@ -54,6 +54,7 @@ class LeNet5(nn.Module):
         self.shadows: dict[str, torch.Tensor] = {}
         self.dropout_seed = seed
         self._step = 0
+        self.training_direct_grads = True
 
     def set_step(self, step: int):
         """Dropout offset — keeps masks deterministic per (seed, step)."""
@@ -62,20 +63,36 @@ class LeNet5(nn.Module):
     def _comp(self, name: str) -> torch.Tensor:
         return self.shadows.get(name, getattr(self, name))
 
+    def _gout(self, name: str):
+        """Direct-grad bucket view (GPU bf16 path): backward kernels
+        accumulate into flat_grad without autograd add glue."""
+        if not self.shadows or not self.training_direct_grads:
+            return None
+        g = getattr(self, name).grad
+        return g if (g is not None and g.is_cuda) else None
+
     def forward(self, x: torch.Tensor, train: bool = True) -> torch.Tensor:
         """x: [B,28,28,1] (compute dtype) -> logits [B,10]."""
         h = Fx.conv_pool(x, self.conv1_w, self.conv1_b,
-                         self._comp("conv1_w"), self.conv1_b, need_dx=False)
+                         self._comp("conv1_w"), self.conv1_b, need_dx=False,
+                         dw_out=self._gout("conv1_w"),
+                         db_out=self._gout("conv1_b"))
         h = Fx.conv_pool(h, self.conv2_w, self.conv2_b,
-                         self._comp("conv2_w"), self.conv2_b, need_dx=True)
+                         self._comp("conv2_w"), self.conv2_b, need_dx=True,
+                         dw_out=self._gout("conv2_w"),
+                         db_out=self._gout("conv2_b"))
         h = h.reshape(h.shape[0], 7 * 7 * 64)
         p_keep = 0.5 if train else 1.0
         h = Fx.linear_act(h, self.fc1_w, self.fc1_b,
                           self._comp("fc1_w"), self.fc1_b,
                           relu=True, p_keep=p_keep,
-                          seed=self.dropout_seed, offset=self._step)
+                          seed=self.dropout_seed, offset=self._step,
+                          dw_out=self._gout("fc1_w"),
+                          db_out=self._gout("fc1_b"))
         logits = Fx.linear_act(h, self.fc2_w, self.fc2_b,
-                               self._comp("fc2_w"), self.fc2_b, relu=False)
+                               self._comp("fc2_w"), self.fc2_b, relu=False,
+                               dw_out=self._gout("fc2_w"),
+                               db_out=self._gout("fc2_b"))
         return logits
 
     def loss_and_accuracy(self, logits, labels):

@@ -31,6 +31,7 @@ class MLP(nn.Module):
         self.shadows: dict[str, torch.Tensor] = {}
         self.dropout_seed = seed
         self._step = 0
+        self.training_direct_grads = True
 
     def set_step(self, step: int):
         self._step = int(step)
@@ -38,12 +39,22 @@ class MLP(nn.Module):
     def _comp(self, name: str) -> torch.Tensor:
         return self.shadows.get(name, getattr(self, name))
 
+    def _gout(self, name: str):
+        if not self.shadows or not self.training_direct_grads:
+            return None
+        g = getattr(self, name).grad
+        return g if (g is not None and g.is_cuda) else None
+
     def forward(self, x: torch.Tensor, train: bool = True) -> torch.Tensor:
         h = x.reshape(x.shape[0], -1)
         h = Fx.linear_act(h, self.fc1_w, self.fc1_b,
-                          self._comp("fc1_w"), self.fc1_b, relu=True)
+                          self._comp("fc1_w"), self.fc1_b, relu=True,
+                          dw_out=self._gout("fc1_w"),
+                          db_out=self._gout("fc1_b"))
         return Fx.linear_act(h, self.fc2_w, self.fc2_b,
-                             self._comp("fc2_w"), self.fc2_b, relu=False)
+                             self._comp("fc2_w"), self.fc2_b, relu=False,
+                             dw_out=self._gout("fc2_w"),
+                             db_out=self._gout("fc2_b"))
 
     def loss_and_accuracy(self, logits, labels):
         loss, correct = Fx.softmax_xent(logits, labels)

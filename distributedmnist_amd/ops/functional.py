@@ -25,34 +25,51 @@ def _use_hip(t: torch.Tensor) -> bool:
 
 
 class ConvPoolFn(torch.autograd.Function):
-    """Fused conv5x5-SAME + bias + ReLU + maxpool2x2s2 (mnist.py:107-127)."""
+    """Fused conv5x5-SAME + bias + ReLU + maxpool2x2s2 (mnist.py:107-127).
+
+    GPU direct-grad mode: when dw_out/db_out (fp32 views into the flat
+    all-reduce bucket, pre-zeroed each step) are passed, the backward
+    kernels accumulate straight into them and None is returned to autograd
+    for w/b — no per-param zeros() or AccumulateGrad adds on the hot path.
+    """
 
     @staticmethod
-    def forward(ctx, x, w, b, w_comp, b_comp, need_dx: bool):
+    def forward(ctx, x, w, b, w_comp, b_comp, need_dx: bool,
+                dw_out, db_out):
         if _use_hip(x):
             y, amax = _C.ext().conv_pool_fwd(x, w_comp, b_comp)
         else:
             y, amax = cpu_ref.conv_pool_fwd(x, w_comp, b_comp)
         ctx.save_for_backward(x, w_comp, y, amax)
         ctx.need_dx = need_dx
+        ctx.outs = (dw_out, db_out) if (dw_out is not None and _use_hip(x)) else None
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x, w_comp, y, amax = ctx.saved_tensors
         if _use_hip(dy):
-            dx, dw, db = _C.ext().conv_pool_bwd(
-                dy.contiguous(), x, w_comp, y, amax, ctx.need_dx)
+            if ctx.outs is not None:
+                dw_out, db_out = ctx.outs
+                dx = _C.ext().conv_pool_bwd_into(
+                    dy.contiguous(), x, w_comp, y, amax, ctx.need_dx,
+                    dw_out, db_out)
+                dw = db = None
+            else:
+                dx, dw, db = _C.ext().conv_pool_bwd(
+                    dy.contiguous(), x, w_comp, y, amax, ctx.need_dx)
         else:
             dx, dw, db = cpu_ref.conv_pool_bwd(dy, x, w_comp, y, amax)
         if not ctx.need_dx:
             dx = None
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None, None
 
 
-def conv_pool(x, w, b, w_comp=None, b_comp=None, need_dx=True):
+def conv_pool(x, w, b, w_comp=None, b_comp=None, need_dx=True,
+              dw_out=None, db_out=None):
     return ConvPoolFn.apply(x, w, b, w_comp if w_comp is not None else w,
-                            b_comp if b_comp is not None else b, need_dx)
+                            b_comp if b_comp is not None else b, need_dx,
+                            dw_out, db_out)
 
 
 class LinearActFn(torch.autograd.Function):
@@ -65,7 +82,7 @@ class LinearActFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b, w_comp, b_comp, relu: bool, p_keep: float,
-                seed: int, offset: int):
+                seed: int, offset: int, dw_out, db_out):
         if _use_hip(x):
             y = _C.ext().linear_act_fwd(x, w_comp, b_comp, relu, p_keep,
                                         seed, offset)
@@ -79,6 +96,7 @@ class LinearActFn(torch.autograd.Function):
         ctx.save_for_backward(x, w_comp, y)
         ctx.relu = relu
         ctx.p_keep = p_keep
+        ctx.outs = (dw_out, db_out) if (dw_out is not None and _use_hip(x)) else None
         return y
 
     @staticmethod
@@ -86,9 +104,16 @@ class LinearActFn(torch.autograd.Function):
         x, w_comp, y = ctx.saved_tensors
         relu, p_keep = ctx.relu, ctx.p_keep
         if _use_hip(dy):
-            dx, dw, db = _C.ext().linear_act_bwd(
-                dy.contiguous(), x, w_comp, y, relu, p_keep,
-                ctx.needs_input_grad[0])
+            if ctx.outs is not None:
+                dw_out, db_out = ctx.outs
+                dx = _C.ext().linear_act_bwd_into(
+                    dy.contiguous(), x, w_comp, y, relu, p_keep,
+                    ctx.needs_input_grad[0], dw_out, db_out)
+                dw = db = None
+            else:
+                dx, dw, db = _C.ext().linear_act_bwd(
+                    dy.contiguous(), x, w_comp, y, relu, p_keep,
+                    ctx.needs_input_grad[0])
         else:
             dyf = dy.float()
             if relu or p_keep < 1.0:
@@ -100,14 +125,15 @@ class LinearActFn(torch.autograd.Function):
             db = dyf.sum(dim=0)
         if not ctx.needs_input_grad[0]:
             dx = None
-        return dx, dw, db, None, None, None, None, None, None
+        return (dx, dw, db, None, None, None, None, None, None, None, None)
 
 
 def linear_act(x, w, b, w_comp=None, b_comp=None, relu=False, p_keep=1.0,
-               seed=0, offset=0):
+               seed=0, offset=0, dw_out=None, db_out=None):
     return LinearActFn.apply(x, w, b, w_comp if w_comp is not None else w,
                              b_comp if b_comp is not None else b, relu,
-                             float(p_keep), int(seed), int(offset))
+                             float(p_keep), int(seed), int(offset),
+                             dw_out, db_out)
 
 
 class SoftmaxXentFn(torch.autograd.Function):

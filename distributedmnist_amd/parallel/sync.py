@@ -105,6 +105,14 @@ class SyncEngine:
         if self.mode == "interval":
             return self._reduce_interval(step)
 
+        # hot path: pure synchronous all-reduce, no staging copies, no
+        # host round-trips — the collective IS the barrier
+        if self.mode == "full_sync" and self.timeout_s is None:
+            if self.distributed:
+                dist.all_reduce(self.flat_grad, op=dist.ReduceOp.SUM,
+                                group=self.group)
+            return True, self.flat_grad, self.world
+
         contribute = True
         if self.timeout_s is not None and compute_time_s > self.timeout_s:
             contribute = False

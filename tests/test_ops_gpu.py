@@ -257,6 +257,6 @@ def test_train_steps_reduce_loss_gpu(ext):
     for _ in range(30):
         xb, yb = ds.next_batch(256)
         _, loss, acc, _ = t.train_step(xb, yb)
-        losses.append(loss)
+        losses.append(float(loss))
     assert all(np.isfinite(losses))
     assert np.mean(losses[-10:]) < np.mean(losses[:10]), losses
