@@ -62,7 +62,7 @@ def main():
 
     def one_step():
         x, y = ds.next_batch(args.batch_size)
-        trainer.train_step(x, y)
+        trainer.graph_or_eager_step(x, y)
 
     for _ in range(args.warmup):
         one_step()

@@ -36,9 +36,10 @@ int pick_splitk(int mtiles, int ntiles, int ksteps) {
 }
 
 // --------------------------------------------------------------------------
-torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
-                             bool relu, double p_keep, int64_t seed,
-                             int64_t offset) {
+torch::Tensor linear_act_fwd_impl(torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor b, bool relu, double p_keep,
+                                  int64_t seed, int64_t offset,
+                                  const long* offset_dev) {
   CHECK_CUDA(x); CHECK_BF16(x); CHECK_CONTIG(x);
   CHECK_BF16(w); CHECK_CONTIG(w);
   CHECK_F32(b); CHECK_CONTIG(b);
@@ -53,6 +54,7 @@ torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
   p.lda = K; p.ldb = N; p.ldc = N;
   p.splitk = 1;
   p.p_keep = (float)p_keep; p.seed = (uint64_t)seed; p.offset = (uint64_t)offset;
+  p.offset_dev = offset_dev;
   bool drop = p_keep < 1.0;
   TORCH_CHECK(!drop || relu, "dropout path requires relu epilogue");
   bool big = cdiv(M, 128) * cdiv(N, 128) >= 128;
@@ -61,6 +63,22 @@ torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
   else if (relu) (big ? gemm_fwd_relu_128 : gemm_fwd_relu_64)(p, s);
   else (big ? gemm_fwd_bias_128 : gemm_fwd_bias_64)(p, s);
   return y;
+}
+
+torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                             bool relu, double p_keep, int64_t seed,
+                             int64_t offset) {
+  return linear_act_fwd_impl(x, w, b, relu, p_keep, seed, offset, nullptr);
+}
+
+// hipGraph-capturable variant: dropout RNG offset read from device memory
+torch::Tensor linear_act_fwd_dev(torch::Tensor x, torch::Tensor w,
+                                 torch::Tensor b, bool relu, double p_keep,
+                                 int64_t seed, torch::Tensor offset_dev) {
+  TORCH_CHECK(offset_dev.scalar_type() == at::kLong && offset_dev.is_cuda(),
+              "offset_dev must be a cuda int64 tensor");
+  return linear_act_fwd_impl(x, w, b, relu, p_keep, seed, 0,
+                             offset_dev.data_ptr<long>());
 }
 
 // core backward; when dw/db are passed they are accumulated into
@@ -268,6 +286,30 @@ void sgd_step(torch::Tensor master, torch::Tensor grad, torch::Tensor shadow,
                   (uint64_t)offset, cur_stream());
 }
 
+void sgd_step_dev(torch::Tensor master, torch::Tensor grad,
+                  torch::Tensor shadow, bool has_shadow,
+                  torch::Tensor lr_scale_dev, double dc_keep, int64_t seed,
+                  torch::Tensor offset_dev) {
+  CHECK_CUDA(master); CHECK_F32(master); CHECK_CONTIG(master);
+  CHECK_F32(grad); CHECK_CONTIG(grad);
+  CHECK_F32(lr_scale_dev);
+  launch_sgd_step_dev(master.data_ptr<float>(), grad.data_ptr<float>(),
+                      has_shadow ? bf16_mut(shadow) : nullptr,
+                      has_shadow ? 1 : 0, master.numel(),
+                      lr_scale_dev.data_ptr<float>(), (float)dc_keep,
+                      (uint64_t)seed, offset_dev.data_ptr<long>(),
+                      cur_stream());
+}
+
+void step_advance(torch::Tensor step_dev, torch::Tensor lr_scale_dev,
+                  double lr0, double decay, int64_t decay_steps,
+                  double inv_contrib) {
+  launch_step_advance(step_dev.data_ptr<long>(),
+                      lr_scale_dev.data_ptr<float>(), (float)lr0,
+                      (float)decay, (int)decay_steps, (float)inv_contrib,
+                      cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -281,4 +323,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "conv backward accumulating dw/db into bucket views");
   m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax-CE (+grad)");
   m.def("sgd_step", &sgd_step, "fused flat SGD apply");
+  m.def("sgd_step_dev", &sgd_step_dev,
+        "SGD apply with device-side lr/offset (hipGraph-capturable)");
+  m.def("step_advance", &step_advance,
+        "device-side staircase LR + step increment (inside the graph)");
+  m.def("linear_act_fwd_dev", &linear_act_fwd_dev,
+        "linear fwd with device-side dropout offset");
 }

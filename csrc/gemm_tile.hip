@@ -6,11 +6,13 @@
 //   4-register max in the MFMA epilogue — zero cross-lane traffic),
 //   conv bwd dX and dW (im2col gathers in the LDS staging stage).
 //
-// Structure: 256 threads = 4 waves (2x2), v_mfma_f32_16x16x32_bf16 tiles,
-// A/B tiles staged to LDS as [outer][K+8] (the +8 bf16 row pad makes the
-// 16-lane ds_read_b128 fragment reads bank-conflict-free: row stride 80B,
-// (a/4)%64 distinct across the group). Single-buffered, 2 barriers/K-step
-// (v1 — see cdna_hip_programming.md §5 ladder for the pipelining headroom).
+// Structure: 256 threads = 4 waves (2x2), v_mfma_f32_16x16x32_bf16,
+// BK=64, double-buffered LDS, ONE barrier per K-step with the T14
+// issue-early/write-late split (cdna_hip_programming.md §6 G15): the next
+// tile's global loads are issued BEFORE the current tile's MFMAs so HBM
+// latency hides under compute; the ds_write lands after the barrier.
+// LDS images are [outer][K+8] (the +8 bf16 row pad makes the 16-lane
+// ds_read_b128 fragment reads conflict-free: row stride 144B).
 
 #include "common.h"
 #include "kernels.h"
@@ -34,6 +36,8 @@ enum Epi { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_RELU = 2, EPI_BIAS_RELU_DROP = 3
 enum OutKind { OUT_BF16 = 0, OUT_F32_ATOMIC = 1 };
 
 #define NTHREADS 256
+#define BK 64
+#define LDK (BK + 8)
 
 // load up to 8 bf16 from p[0..vcnt), zero-fill the rest; vector fast path
 DEV short8 loadRow8(const ushort_t* p, int vcnt) {
@@ -59,23 +63,26 @@ DEV bool decode_pool_m(const GemmParams& p, int m, int& n, int& oy, int& ox) {
   return n < p.CB;
 }
 
-template <int BM, int BN, int AMODE>
-DEV void stageA(const GemmParams& p, ushort_t (*As)[40], int m0, int kt,
-                int kend, int tid) {
-  constexpr int BK = 32;
-  if (AMODE == A_N || AMODE == A_CONV_FWD || AMODE == A_CONV_DX) {
-    // row-contiguous sources: chunks of 8 bf16 along k
-    for (int c = tid; c < BM * (BK / 8); c += NTHREADS) {
+// ---- staging: load phase (global -> regs), mode-specific addressing ------
+// Chunk c covers either 8 k-elements of one row (rows16 layout) or 8 outer
+// elements of one k (trans layout).  CH chunks per thread.
+
+template <int BM, int AMODE, int CH>
+DEV void loadA(const GemmParams& p, int m0, int kt, int kend, int tid,
+               short8 (&regs)[CH]) {
+#pragma unroll
+  for (int j = 0; j < CH; ++j) {
+    int c = tid + j * NTHREADS;
+    short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+    if (AMODE == A_N || AMODE == A_CONV_FWD || AMODE == A_CONV_DX) {
       int i = c / (BK / 8);
       int kc = (c % (BK / 8)) * 8;
       int m = m0 + i, k = kt + kc;
-      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
       int vcnt = min(8, kend - k);
       if (m < p.M && vcnt > 0) {
         if (AMODE == A_N) {
           v = loadRow8(p.A + (size_t)m * p.lda + k, vcnt);
         } else if (AMODE == A_CONV_FWD) {
-          // k = khkw*Cin + ci ; chunk stays inside one khkw (Cin % 8 == 0)
           int khkw = k / p.Cin, ci = k % p.Cin;
           int kh = khkw / 5, kw = khkw % 5;
           int n, oy, ox;
@@ -83,9 +90,9 @@ DEV void stageA(const GemmParams& p, ushort_t (*As)[40], int m0, int kt,
             int y = oy + kh - 2, x = ox + kw - 2;
             if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
               v = loadRow8(p.A + (((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci,
-                           vcnt);
+                           min(vcnt, p.Cin - ci));
           }
-        } else {  // A_CONV_DX: m = input pixel, k = khkw*Cout + co
+        } else {  // A_CONV_DX
           int wi = m % p.CW, t2 = m / p.CW;
           int hi = t2 % p.CH, n = t2 / p.CH;
           int khkw = k / p.Cout, co = k % p.Cout;
@@ -93,128 +100,155 @@ DEV void stageA(const GemmParams& p, ushort_t (*As)[40], int m0, int kt,
           int y = hi - kh + 2, x = wi - kw + 2;
           if (n < p.CB && y >= 0 && y < p.CH && x >= 0 && x < p.CW)
             v = loadRow8(p.A + (((size_t)n * p.CH + y) * p.CW + x) * p.Cout + co,
-                         vcnt);
+                         min(vcnt, p.Cout - co));
         }
       }
-      *reinterpret_cast<short8*>(&As[i][kc]) = v;
-    }
-  } else if (AMODE == A_T) {
-    // source rows are k-major: load 8 consecutive m, scatter-transpose
-    for (int c = tid; c < BK * (BM / 8); c += NTHREADS) {
+    } else if (AMODE == A_T) {
       int kk = c / (BM / 8);
       int i0 = (c % (BM / 8)) * 8;
       int k = kt + kk, m = m0 + i0;
-      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
       if (k < kend) v = loadRow8(p.A + (size_t)k * p.lda + m, min(8, p.M - m));
-#pragma unroll
-      for (int j = 0; j < 8; ++j) As[i0 + j][kk] = v[j];
-    }
-  } else if (AMODE == A_CONV1_FWD) {
-    // Cin==1, K==25: scalar gather (tiny)
-    for (int c = tid; c < BM * BK; c += NTHREADS) {
-      int i = c / BK, kk = c % BK;
-      int m = m0 + i, k = kt + kk;
-      ushort_t v = 0;
+    } else if (AMODE == A_CONV1_FWD) {
+      // Cin==1, K==25: 8 scalar gathers per chunk (tiny op)
+      int i = c / (BK / 8);
+      int kc = (c % (BK / 8)) * 8;
+      int m = m0 + i;
       int n, oy, ox;
-      if (m < p.M && k < kend && k < 25 && decode_pool_m(p, m, n, oy, ox)) {
-        int kh = k / 5, kw = k % 5;
-        int y = oy + kh - 2, x = ox + kw - 2;
-        if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
-          v = p.A[((size_t)n * p.CH + y) * p.CW + x];
-      }
-      As[i][kk] = v;
-    }
-  } else {  // A_CONV_DW: m = (khkw, ci), k = conv-output pixel
-    if (p.Cin % 8 == 0) {
-      // 8 consecutive m stay inside one khkw (ci0 aligned to 8):
-      // one 16B load of x[pixel][ci0..ci0+7], transpose-scatter to LDS
-      for (int c = tid; c < BK * (BM / 8); c += NTHREADS) {
-        int kk = c / (BM / 8);
-        int i0 = (c % (BM / 8)) * 8;
-        int k = kt + kk, m = m0 + i0;
-        short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
-        if (m < p.M && k < kend) {
-          int khkw = m / p.Cin, ci = m % p.Cin;
-          int kh = khkw / 5, kw = khkw % 5;
-          int w_ = k % p.CW, t2 = k / p.CW;
-          int h_ = t2 % p.CH, n = t2 / p.CH;
-          int y = h_ + kh - 2, x = w_ + kw - 2;
-          if (n < p.CB && y >= 0 && y < p.CH && x >= 0 && x < p.CW)
-            v = loadRow8(p.A + (((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci,
-                         min(8, p.M - m));
-        }
+      if (m < p.M && kc < 25 && decode_pool_m(p, m, n, oy, ox)) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) As[i0 + j][kk] = v[j];
-      }
-    } else {  // Cin == 1 (conv1): scalar, tiny M
-      for (int c = tid; c < BM * BK; c += NTHREADS) {
-        int i = c / BK, kk = c % BK;
-        int m = m0 + i, k = kt + kk;
-        ushort_t v = 0;
-        if (m < p.M && k < kend) {
-          int khkw = m / p.Cin, ci = m % p.Cin;
-          int kh = khkw / 5, kw = khkw % 5;
-          int w_ = k % p.CW, t2 = k / p.CW;
-          int h_ = t2 % p.CH, n = t2 / p.CH;
-          int y = h_ + kh - 2, x = w_ + kw - 2;
-          if (n < p.CB && y >= 0 && y < p.CH && x >= 0 && x < p.CW)
-            v = p.A[(((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci];
+        for (int e = 0; e < 8; ++e) {
+          int k = kt + kc + e;
+          if (k < kend && k < 25) {
+            int kh = k / 5, kw = k % 5;
+            int y = oy + kh - 2, x = ox + kw - 2;
+            if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+              v[e] = p.A[((size_t)n * p.CH + y) * p.CW + x];
+          }
         }
-        As[i][kk] = v;
       }
+    } else {  // A_CONV_DW: A[m=(khkw,ci)][k=pixel]
+      int kk = c / (BM / 8);
+      int i0 = (c % (BM / 8)) * 8;
+      int k = kt + kk, m = m0 + i0;
+      if (k < kend && m < p.M) {
+        int w_ = k % p.CW, t2 = k / p.CW;
+        int h_ = t2 % p.CH, n = t2 / p.CH;
+        if (n < p.CB) {
+          if (p.Cin % 8 == 0) {
+            // 8 consecutive m share one khkw (ci0 aligned to 8): vector load
+            int khkw = m / p.Cin, ci = m % p.Cin;
+            int kh = khkw / 5, kw = khkw % 5;
+            int y = h_ + kh - 2, x = w_ + kw - 2;
+            if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+              v = loadRow8(p.A + (((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci,
+                           min(8, p.M - m));
+          } else {
+            // Cin==1: every m is a different filter tap — per-element gather
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              int me = m + e;
+              if (me >= p.M) break;
+              int khkw = me / p.Cin, ci = me % p.Cin;
+              int kh = khkw / 5, kw = khkw % 5;
+              int y = h_ + kh - 2, x = w_ + kw - 2;
+              if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+                v[e] = p.A[(((size_t)n * p.CH + y) * p.CW + x) * p.Cin + ci];
+            }
+          }
+        }
+      }
+    }
+    regs[j] = v;
+  }
+}
+
+template <int BM, int AMODE, int CH>
+DEV void writeA(ushort_t (*As)[LDK], int tid, const short8 (&regs)[CH]) {
+#pragma unroll
+  for (int j = 0; j < CH; ++j) {
+    int c = tid + j * NTHREADS;
+    if (AMODE == A_N || AMODE == A_CONV_FWD || AMODE == A_CONV_DX ||
+        AMODE == A_CONV1_FWD) {
+      int i = c / (BK / 8);
+      int kc = (c % (BK / 8)) * 8;
+      *reinterpret_cast<short8*>(&As[i][kc]) = regs[j];
+    } else {  // trans layouts scatter 8 outer rows at one k column
+      int kk = c / (BM / 8);
+      int i0 = (c % (BM / 8)) * 8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) As[i0 + e][kk] = regs[j][e];
     }
   }
 }
 
-template <int BN, int BMODE>
-DEV void stageB(const GemmParams& p, ushort_t (*Bs)[40], int n0, int kt,
-                int kend, int tid) {
-  constexpr int BK = 32;
-  if (BMODE == B_KMAJ) {
-    for (int c = tid; c < BK * (BN / 8); c += NTHREADS) {
+template <int BN, int BMODE, int CH>
+DEV void loadB(const GemmParams& p, int n0, int kt, int kend, int tid,
+               short8 (&regs)[CH]) {
+#pragma unroll
+  for (int j = 0; j < CH; ++j) {
+    int c = tid + j * NTHREADS;
+    short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+    if (BMODE == B_KMAJ) {
       int kk = c / (BN / 8);
       int j0 = (c % (BN / 8)) * 8;
       int k = kt + kk, n = n0 + j0;
-      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
       if (k < kend) v = loadRow8(p.B + (size_t)k * p.ldb + n, min(8, p.N - n));
-#pragma unroll
-      for (int j = 0; j < 8; ++j) Bs[j0 + j][kk] = v[j];
-    }
-  } else if (BMODE == B_NMAJ) {
-    for (int c = tid; c < BN * (BK / 8); c += NTHREADS) {
+    } else if (BMODE == B_NMAJ) {
       int i = c / (BK / 8);
       int kc = (c % (BK / 8)) * 8;
       int n = n0 + i, k = kt + kc;
-      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
       int vcnt = min(8, kend - k);
-      if (n < p.N && vcnt > 0)
-        v = loadRow8(p.B + (size_t)n * p.ldb + k, vcnt);
-      *reinterpret_cast<short8*>(&Bs[i][kc]) = v;
-    }
-  } else {  // B_CONV_DX_W: Bs[ci][kk] = w[((khkw)*Cin + ci)*Cout + co(k)]
-    for (int c = tid; c < BN * (BK / 8); c += NTHREADS) {
+      if (n < p.N && vcnt > 0) v = loadRow8(p.B + (size_t)n * p.ldb + k, vcnt);
+    } else {  // B_CONV_DX_W
       int i = c / (BK / 8);
       int kc = (c % (BK / 8)) * 8;
-      int n = n0 + i, k = kt + kc;  // n = ci, k = khkw*Cout + co
-      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      int n = n0 + i, k = kt + kc;
       if (n < p.N && k < kend) {
         int khkw = k / p.Cout, co = k % p.Cout;
         v = loadRow8(p.B + ((size_t)khkw * p.Cin + n) * p.Cout + co,
-                     min(8, kend - k));
+                     min(min(8, kend - k), p.Cout - co));
       }
-      *reinterpret_cast<short8*>(&Bs[i][kc]) = v;
+    }
+    regs[j] = v;
+  }
+}
+
+template <int BN, int BMODE, int CH>
+DEV void writeB(ushort_t (*Bs)[LDK], int tid, const short8 (&regs)[CH]) {
+#pragma unroll
+  for (int j = 0; j < CH; ++j) {
+    int c = tid + j * NTHREADS;
+    if (BMODE == B_KMAJ) {
+      int kk = c / (BN / 8);
+      int j0 = (c % (BN / 8)) * 8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) Bs[j0 + e][kk] = regs[j][e];
+    } else {
+      int i = c / (BK / 8);
+      int kc = (c % (BK / 8)) * 8;
+      *reinterpret_cast<short8*>(&Bs[i][kc]) = regs[j];
     }
   }
 }
 
-template <int BM, int BN, int AMODE, int BMODE, int EPI, int OUT>
+// ---------------------------------------------------------------------------
+// PIPE=1: double-buffered LDS, issue-early/write-late, 1 barrier/K-step —
+//   wins for small grids (fc GEMMs) where occupancy is grid-limited anyway.
+// PIPE=0: single buffer, 2 barriers/K-step, half the LDS — wins for the
+//   huge-grid conv gathers where 4-5 blocks/CU of TLP hide latency better
+//   than the in-wave pipeline (measured: conv2 fwd 103us PIPE0 vs 120us
+//   PIPE1; fc1 fwd 94us PIPE0 vs 59us PIPE1).
+template <int BM, int BN, int AMODE, int BMODE, int EPI, int OUT, int PIPE>
 __global__ __launch_bounds__(NTHREADS)
 void gemm_tile_kernel(GemmParams p) {
-  constexpr int BK = 32;
   constexpr int WM = BM / 2, WN = BN / 2;
   constexpr int MI = WM / 16, NI = WN / 16;
-  __shared__ __align__(16) ushort_t As[BM][BK + 8];
-  __shared__ __align__(16) ushort_t Bs[BN][BK + 8];
+  constexpr int CHA = (BM * BK / 8) / NTHREADS;
+  constexpr int CHB = (BN * BK / 8) / NTHREADS;
+  constexpr int DB = PIPE ? 2 : 1;
+  static_assert(CHA >= 1 && CHB >= 1, "tile too small for 256 threads");
+  __shared__ __align__(16) ushort_t As[DB][BM][LDK];
+  __shared__ __align__(16) ushort_t Bs[DB][BN][LDK];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -223,11 +257,12 @@ void gemm_tile_kernel(GemmParams p) {
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
 
-  // K-range of this split-K slice
   int ksteps_total = (p.K + BK - 1) / BK;
   int steps_per = (ksteps_total + p.splitk - 1) / p.splitk;
   int kbeg = (int)blockIdx.z * steps_per * BK;
   int kend = min(p.K, kbeg + steps_per * BK);
+  int nt = (kend - kbeg + BK - 1) / BK;
+  if (nt <= 0) nt = 0;
 
   f32x4 acc[MI][NI];
 #pragma unroll
@@ -238,24 +273,61 @@ void gemm_tile_kernel(GemmParams p) {
   const int lrow = lane & 15;
   const int kq = (lane >> 4) * 8;
 
-  for (int kt = kbeg; kt < kend; kt += BK) {
-    stageA<BM, BN, AMODE>(p, As, m0, kt, kend, tid);
-    stageB<BN, BMODE>(p, Bs, n0, kt, kend, tid);
+  short8 ra[CHA], rb[CHB];
+  int cur = 0;
+  if (PIPE && nt > 0) {
+    loadA<BM, AMODE, CHA>(p, m0, kbeg, kend, tid, ra);
+    loadB<BN, BMODE, CHB>(p, n0, kbeg, kend, tid, rb);
+    writeA<BM, AMODE, CHA>(As[0], tid, ra);
+    writeB<BN, BMODE, CHB>(Bs[0], tid, rb);
     __syncthreads();
-    short8 af[MI], bf[NI];
+  }
+
+  for (int t = 0; t < nt; ++t) {
+    const bool have_next = PIPE && (t + 1) < nt;
+    if (PIPE) {
+      if (have_next) {
+        int kn = kbeg + (t + 1) * BK;
+        loadA<BM, AMODE, CHA>(p, m0, kn, kend, tid, ra);  // issue early (T14)
+        loadB<BN, BMODE, CHB>(p, n0, kn, kend, tid, rb);
+      }
+    } else {
+      int kt = kbeg + t * BK;
+      loadA<BM, AMODE, CHA>(p, m0, kt, kend, tid, ra);
+      loadB<BN, BMODE, CHB>(p, n0, kt, kend, tid, rb);
+      writeA<BM, AMODE, CHA>(As[0], tid, ra);
+      writeB<BN, BMODE, CHB>(Bs[0], tid, rb);
+      __syncthreads();
+    }
+    // compute current tile: BK=64 = two K=32 MFMA sub-steps
 #pragma unroll
-    for (int mi = 0; mi < MI; ++mi)
-      af[mi] = *reinterpret_cast<const short8*>(&As[wr * WM + mi * 16 + lrow][kq]);
+    for (int kh = 0; kh < 2; ++kh) {
+      short8 af[MI], bf[NI];
 #pragma unroll
-    for (int ni = 0; ni < NI; ++ni)
-      bf[ni] = *reinterpret_cast<const short8*>(&Bs[wc * WN + ni * 16 + lrow][kq]);
-#pragma unroll
-    for (int mi = 0; mi < MI; ++mi)
+      for (int mi = 0; mi < MI; ++mi)
+        af[mi] = *reinterpret_cast<const short8*>(
+            &As[cur][wr * WM + mi * 16 + lrow][kh * 32 + kq]);
 #pragma unroll
       for (int ni = 0; ni < NI; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
-    __syncthreads();
+        bf[ni] = *reinterpret_cast<const short8*>(
+            &Bs[cur][wc * WN + ni * 16 + lrow][kh * 32 + kq]);
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    if (PIPE) {
+      if (have_next) {
+        writeA<BM, AMODE, CHA>(As[cur ^ 1], tid, ra);  // write after MFMAs
+        writeB<BN, BMODE, CHB>(Bs[cur ^ 1], tid, rb);
+        __syncthreads();
+        cur ^= 1;
+      }
+    } else {
+      __syncthreads();
+    }
   }
 
   // ---- epilogue ----------------------------------------------------------
@@ -293,8 +365,8 @@ void gemm_tile_kernel(GemmParams p) {
           if (EPI == EPI_BIAS_RELU || EPI == EPI_BIAS_RELU_DROP)
             v = v > 0.f ? v : 0.f;
           if (EPI == EPI_BIAS_RELU_DROP) {
-            float u = philox_uniform(p.seed, p.offset,
-                                     (uint64_t)gr * p.N + gc);
+            uint64_t off = p.offset_dev ? (uint64_t)*p.offset_dev : p.offset;
+            float u = philox_uniform(p.seed, off, (uint64_t)gr * p.N + gc);
             v = (u < p.p_keep) ? v / p.p_keep : 0.f;
           }
           if (OUT == OUT_BF16) {
@@ -314,24 +386,26 @@ void gemm_tile_kernel(GemmParams p) {
 static inline int cdiv_host(int a, int b) { return (a + b - 1) / b; }
 
 // Explicit entry points used by bindings.cpp (keeps instantiations bounded).
-#define GEMM_ENTRY(name, BM, BN, AM, BMo, EPI, OUT)                          \
+#define GEMM_ENTRY(name, BM, BN, AM, BMo, EPI, OUT, PIPE)                   \
   void name(const GemmParams& p, hipStream_t s) {                            \
     dim3 grid(cdiv_host(p.M, BM), cdiv_host(p.N, BN), p.splitk);             \
-    hipLaunchKernelGGL((gemm_tile_kernel<BM, BN, AM, BMo, EPI, OUT>), grid,  \
-                       dim3(NTHREADS), 0, s, p);                             \
+    hipLaunchKernelGGL((gemm_tile_kernel<BM, BN, AM, BMo, EPI, OUT, PIPE>),  \
+                       grid, dim3(NTHREADS), 0, s, p);                       \
   }
 
-GEMM_ENTRY(gemm_fwd_bias_128, 128, 128, A_N, B_KMAJ, EPI_BIAS, OUT_BF16)
-GEMM_ENTRY(gemm_fwd_bias_64, 64, 64, A_N, B_KMAJ, EPI_BIAS, OUT_BF16)
-GEMM_ENTRY(gemm_fwd_relu_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU, OUT_BF16)
-GEMM_ENTRY(gemm_fwd_relu_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU, OUT_BF16)
-GEMM_ENTRY(gemm_fwd_drop_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16)
-GEMM_ENTRY(gemm_fwd_drop_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16)
-GEMM_ENTRY(gemm_dx_128, 128, 128, A_N, B_NMAJ, EPI_NONE, OUT_BF16)
-GEMM_ENTRY(gemm_dx_64, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_BF16)
-GEMM_ENTRY(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC)
-GEMM_ENTRY(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC)
-GEMM_ENTRY(conv_fwd_pool, 128, 64, A_CONV_FWD, B_KMAJ, EPI_POOL, OUT_BF16)
-GEMM_ENTRY(conv1_fwd_pool, 128, 64, A_CONV1_FWD, B_KMAJ, EPI_POOL, OUT_BF16)
-GEMM_ENTRY(conv_dx_gemm, 128, 32, A_CONV_DX, B_CONV_DX_W, EPI_NONE, OUT_BF16)
-GEMM_ENTRY(conv_dw_gemm, 64, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC)
+// fc GEMMs: grids of 32-400 WGs (occupancy grid-limited) -> PIPE=1
+GEMM_ENTRY(gemm_fwd_bias_128, 128, 128, A_N, B_KMAJ, EPI_BIAS, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_bias_64, 64, 64, A_N, B_KMAJ, EPI_BIAS, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_relu_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_relu_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_drop_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_drop_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
+GEMM_ENTRY(gemm_dx_128, 128, 128, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
+GEMM_ENTRY(gemm_dx_64, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
+GEMM_ENTRY(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+GEMM_ENTRY(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+// conv gathers: thousands of WGs -> single-buffer, TLP hides latency
+GEMM_ENTRY(conv_fwd_pool, 128, 64, A_CONV_FWD, B_KMAJ, EPI_POOL, OUT_BF16, 0)
+GEMM_ENTRY(conv1_fwd_pool, 128, 64, A_CONV1_FWD, B_KMAJ, EPI_POOL, OUT_BF16, 0)
+GEMM_ENTRY(conv_dx_gemm, 128, 32, A_CONV_DX, B_CONV_DX_W, EPI_NONE, OUT_BF16, 0)
+GEMM_ENTRY(conv_dw_gemm, 64, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)

@@ -15,6 +15,8 @@ struct GemmParams {
   int CB, CH, CW, CHo, CWo, Cin, Cout;
   float p_keep;
   uint64_t seed, offset;
+  const long* offset_dev;  // when set, RNG offset is read from device memory
+                           // (hipGraph replay: host args are frozen)
 };
 
 // gemm_tile.hip — implicit-GEMM MFMA entry points
@@ -47,3 +49,11 @@ void launch_softmax_xent(const unsigned short* logits, const long* labels,
 void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
                      uint64_t seed, uint64_t offset, hipStream_t);
+// graph-capturable variants: lr_scale / RNG offset read from device memory
+void launch_sgd_step_dev(float* master, const float* grad,
+                         unsigned short* shadow, int has_shadow, long n,
+                         const float* lr_scale_dev, float dc_keep,
+                         uint64_t seed, const long* offset_dev, hipStream_t);
+void launch_step_advance(long* step_dev, float* lr_scale_dev, float lr0,
+                         float decay, int decay_steps, float inv_contrib,
+                         hipStream_t);

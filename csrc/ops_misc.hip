@@ -150,7 +150,10 @@ void softmax_xent_kernel(const ushort_t* logits, const long* labels,
 extern "C" __global__ __launch_bounds__(256)
 void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
-                     uint64_t seed, uint64_t offset) {
+                     uint64_t seed, uint64_t offset,
+                     const float* lr_scale_dev, const long* offset_dev) {
+  if (lr_scale_dev) lr_scale = *lr_scale_dev;
+  if (offset_dev) offset = (uint64_t)*offset_dev;
   long i = (long)(blockIdx.x) * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (; i * 4 < n; i += stride) {
@@ -185,6 +188,21 @@ void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
         if (has_shadow) shadow[j] = f2bf(m);
       }
     }
+  }
+}
+
+// step/LR advance: device-side staircase LR so a captured graph needs no
+// per-step host argument updates (lr = lr0*decay^(step/decay_steps),
+// folded with 1/contributors; then step++)
+extern "C" __global__ void step_advance_kernel(long* step_dev,
+                                               float* lr_scale_dev, float lr0,
+                                               float decay, int decay_steps,
+                                               float inv_contrib) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    long s = *step_dev;
+    float lr = lr0 * __powf(decay, (float)(s / decay_steps));
+    *lr_scale_dev = lr * inv_contrib;
+    *step_dev = s + 1;
   }
 }
 
@@ -233,5 +251,26 @@ void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
   if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(sgd_step_kernel, dim3(blocks), dim3(256), 0, s, master,
                      grad, shadow, has_shadow, n, lr_scale, dc_keep, seed,
-                     offset);
+                     offset, (const float*)nullptr, (const long*)nullptr);
+}
+
+void launch_sgd_step_dev(float* master, const float* grad,
+                         unsigned short* shadow, int has_shadow, long n,
+                         const float* lr_scale_dev, float dc_keep,
+                         uint64_t seed, const long* offset_dev,
+                         hipStream_t s) {
+  long groups = (n + 3) / 4;
+  int blocks = (int)min((long)2048, (groups + 255) / 256);
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(sgd_step_kernel, dim3(blocks), dim3(256), 0, s, master,
+                     grad, shadow, has_shadow, n, 0.f, dc_keep, seed, 0,
+                     lr_scale_dev, offset_dev);
+}
+
+void launch_step_advance(long* step_dev, float* lr_scale_dev, float lr0,
+                         float decay, int decay_steps, float inv_contrib,
+                         hipStream_t s) {
+  hipLaunchKernelGGL(step_advance_kernel, dim3(1), dim3(64), 0, s, step_dev,
+                     lr_scale_dev, lr0, decay, decay_steps < 1 ? 1 : decay_steps,
+                     inv_contrib);
 }

@@ -1,0 +1,121 @@
+"""hipGraph-captured training step.
+
+The whole step — grad zero, fwd (HIP kernels), bwd, RCCL all-reduce, fused
+SGD — is captured once into a hipGraph and replayed per iteration: per-step
+host work collapses to two async copies (batch into the static buffers) +
+one graph launch.  Host kernel arguments are frozen under replay, so the
+pieces that change per step live in DEVICE memory:
+  - step_dev (int64): dropout / drop-connect philox offset
+  - lr_scale_dev (fp32): staircase LR x 1/contributors
+advanced INSIDE the graph by the step_advance kernel at the END of the body
+(so the values seen by the body equal the eager schedule's).
+
+Only the full_sync mode is graphable (K-of-N / interval / CDF need
+data-dependent host control flow); Trainer falls back to eager elsewhere.
+"""
+
+from __future__ import annotations
+
+import logging
+
+import torch
+import torch.distributed as dist
+
+from .. import _C
+
+log = logging.getLogger("dmnist.graph")
+
+
+class GraphedStep:
+    def __init__(self, trainer, batch_shape):
+        t = trainer
+        assert t.device.type == "cuda", "GraphedStep requires a GPU"
+        assert t.mode == "full_sync" and t.engine.timeout_s is None, \
+            "only the full_sync hot path is graph-captured"
+        self.t = t
+        ext = _C.ext()
+        dev = t.device
+        flags = t.flags
+        self.static_x = torch.zeros(batch_shape, dtype=t.compute_dtype, device=dev)
+        self.static_y = torch.zeros(batch_shape[0], dtype=torch.int64, device=dev)
+        self.step_dev = torch.zeros(1, dtype=torch.int64, device=dev)
+        self.lr_scale_dev = torch.zeros(1, dtype=torch.float32, device=dev)
+        num_batches = t._num_examples / flags.batch_size
+        self.decay_steps = max(1, int(num_batches * flags.num_epochs_per_decay
+                                      / max(1, t.engine.K)))
+        self.inv_contrib = 1.0 / max(1, t.world)
+        self.dc_keep = (flags.drop_connect_probability if flags.drop_connect
+                        else -1.0)
+        self._ext = ext
+
+        # snapshot state: the warmup iterations below really train
+        master0 = t.fp.flat_master.clone()
+        t.model.set_step_dev(self.step_dev)
+
+        self._prime(t.step)
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                self._body()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self._body()
+
+        # restore pre-warmup state
+        t.fp.flat_master.copy_(master0)
+        t.fp.sync_shadow()
+        self._prime(t.step)
+        torch.cuda.synchronize()
+
+    def _prime(self, step: int):
+        """Set device step + LR for the NEXT body execution."""
+        flags = self.t.flags
+        self.step_dev.fill_(step)
+        lr = (flags.initial_learning_rate *
+              flags.learning_rate_decay_factor ** (step // self.decay_steps))
+        self.lr_scale_dev.fill_(lr * self.inv_contrib)
+
+    def _body(self):
+        t = self.t
+        fp = t.fp
+        fp.flat_grad.zero_()
+        logits = t.model(self.static_x, train=True)
+        loss, acc = t.model.loss_and_accuracy(logits, self.static_y)
+        loss.backward()
+        if t.world > 1 and dist.is_initialized():
+            dist.all_reduce(fp.flat_grad, op=dist.ReduceOp.SUM)
+        self._ext.sgd_step_dev(fp.flat_master, fp.flat_grad,
+                               fp.flat_shadow if fp.flat_shadow is not None
+                               else fp.flat_master,
+                               fp.flat_shadow is not None,
+                               self.lr_scale_dev, self.dc_keep,
+                               t.flags.seed, self.step_dev)
+        # advance step + LR on-device for the next replay
+        self._ext.step_advance(self.step_dev, self.lr_scale_dev,
+                               t.flags.initial_learning_rate,
+                               t.flags.learning_rate_decay_factor,
+                               self.decay_steps, self.inv_contrib)
+        self.static_loss = loss.detach()
+        self.static_acc = acc.detach()
+
+    def run(self, images, labels):
+        """Replay one step. Returns (loss, acc) static device tensors."""
+        self.static_x.copy_(images, non_blocking=True)
+        self.static_y.copy_(labels, non_blocking=True)
+        self.graph.replay()
+        self.t.step += 1
+        return self.static_loss, self.static_acc
+
+
+def try_graph(trainer, batch_shape):
+    """Build a GraphedStep, or None if capture is unsupported here."""
+    try:
+        return GraphedStep(trainer, batch_shape)
+    except Exception as e:  # noqa: BLE001 — fall back to eager on any failure
+        log.warning("hipGraph capture unavailable (%s); running eager", e)
+        return None

@@ -107,6 +107,8 @@ class Trainer:
         self.step = 0
         self.num_contributors = world
         self._num_examples = 60000  # overwritten by train(); LR-schedule default
+        self._graph = None
+        self._graph_tried = False
         # parameter-init parity across ranks (broadcast once)
         if world > 1 and dist.is_initialized():
             dist.broadcast(self.fp.flat_master, src=0)
@@ -122,6 +124,32 @@ class Trainer:
                            non_blocking=True)
         labels = labels.to(device=self.device, non_blocking=True)
         return images, labels
+
+    def get_graph(self, images):
+        """hipGraph-captured step when eligible (GPU, full_sync, no
+        injection/timing instrumentation); None -> eager path."""
+        if self._graph is not None:
+            return self._graph
+        if self._graph_tried:
+            return None
+        self._graph_tried = True
+        flags = self.flags
+        eligible = (self.device.type == "cuda" and self.mode == "full_sync"
+                    and self.engine.timeout_s is None
+                    and flags.inject_slow_rank < 0
+                    and getattr(flags, "hip_graph", "auto") != "off")
+        if eligible:
+            from .graphstep import try_graph
+            self._graph = try_graph(self, tuple(images.shape))
+        return self._graph
+
+    def graph_or_eager_step(self, images, labels):
+        """Preferred step entry: graph replay when available."""
+        g = self.get_graph(images)
+        if g is not None:
+            loss, acc = g.run(images, labels)
+            return True, loss, acc, 0.0
+        return self.train_step(images, labels)
 
     @property
     def _needs_step_timing(self) -> bool:
@@ -208,7 +236,7 @@ class Trainer:
                     f"worker={self.rank}_timeline_iter={self.step}.json"))
                 finish_time = time.time()
             else:
-                applied, loss_v, acc_v, _ct = self.train_step(images, labels)
+                applied, loss_v, acc_v, _ct = self.graph_or_eager_step(images, labels)
                 finish_time = time.time()
             loss_v, acc_v = float(loss_v), float(acc_v)
             duration = finish_time - start_time

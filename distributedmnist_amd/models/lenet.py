@@ -55,6 +55,11 @@ class LeNet5(nn.Module):
         self.dropout_seed = seed
         self._step = 0
         self.training_direct_grads = True
+        self._step_dev = None  # device step counter (hipGraph capture)
+
+    def set_step_dev(self, t):
+        """Device-resident step counter: dropout offsets under hipGraph."""
+        self._step_dev = t
 
     def set_step(self, step: int):
         """Dropout offset — keeps masks deterministic per (seed, step)."""
@@ -88,7 +93,8 @@ class LeNet5(nn.Module):
                           relu=True, p_keep=p_keep,
                           seed=self.dropout_seed, offset=self._step,
                           dw_out=self._gout("fc1_w"),
-                          db_out=self._gout("fc1_b"))
+                          db_out=self._gout("fc1_b"),
+                          offset_dev=self._step_dev)
         logits = Fx.linear_act(h, self.fc2_w, self.fc2_b,
                                self._comp("fc2_w"), self.fc2_b, relu=False,
                                dw_out=self._gout("fc2_w"),

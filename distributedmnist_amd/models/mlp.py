@@ -32,6 +32,11 @@ class MLP(nn.Module):
         self.dropout_seed = seed
         self._step = 0
         self.training_direct_grads = True
+        self._step_dev = None  # device step counter (hipGraph capture)
+
+    def set_step_dev(self, t):
+        """Device-resident step counter: dropout offsets under hipGraph."""
+        self._step_dev = t
 
     def set_step(self, step: int):
         self._step = int(step)

@@ -82,10 +82,14 @@ class LinearActFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b, w_comp, b_comp, relu: bool, p_keep: float,
-                seed: int, offset: int, dw_out, db_out):
+                seed: int, offset: int, dw_out, db_out, offset_dev):
         if _use_hip(x):
-            y = _C.ext().linear_act_fwd(x, w_comp, b_comp, relu, p_keep,
-                                        seed, offset)
+            if offset_dev is not None and p_keep < 1.0:
+                y = _C.ext().linear_act_fwd_dev(x, w_comp, b_comp, relu,
+                                                p_keep, seed, offset_dev)
+            else:
+                y = _C.ext().linear_act_fwd(x, w_comp, b_comp, relu, p_keep,
+                                            seed, offset)
         else:
             y = cpu_ref.linear_fwd(x, w_comp, b_comp, relu)
             if p_keep < 1.0:
@@ -125,15 +129,16 @@ class LinearActFn(torch.autograd.Function):
             db = dyf.sum(dim=0)
         if not ctx.needs_input_grad[0]:
             dx = None
-        return (dx, dw, db, None, None, None, None, None, None, None, None)
+        return (dx, dw, db, None, None, None, None, None, None, None, None,
+                None)
 
 
 def linear_act(x, w, b, w_comp=None, b_comp=None, relu=False, p_keep=1.0,
-               seed=0, offset=0, dw_out=None, db_out=None):
+               seed=0, offset=0, dw_out=None, db_out=None, offset_dev=None):
     return LinearActFn.apply(x, w, b, w_comp if w_comp is not None else w,
                              b_comp if b_comp is not None else b, relu,
                              float(p_keep), int(seed), int(offset),
-                             dw_out, db_out)
+                             dw_out, db_out, offset_dev)
 
 
 class SoftmaxXentFn(torch.autograd.Function):
