@@ -289,7 +289,14 @@ def make_dataset(flags, rank: int, world: int, device, dtype):
     ds = load_mnist(flags.data_dir, fake_data=flags.fake_data,
                     worker_id=rank, n_workers=world,
                     shard=not flags.no_shard, seed=flags.seed + rank)
-    return ds.train
+    train = ds.train
+    if flags.subset and not flags.fake_data:
+        # reference --subset: train on the first N examples
+        from ..data import DataSet
+        train = DataSet(train.images[:flags.subset],
+                        train.labels[:flags.subset], shard=False,
+                        seed=flags.seed + rank)
+    return train
 
 
 def train_main(flags):
