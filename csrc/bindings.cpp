@@ -39,7 +39,8 @@ int pick_splitk(int mtiles, int ntiles, int ksteps) {
 torch::Tensor linear_act_fwd_impl(torch::Tensor x, torch::Tensor w,
                                   torch::Tensor b, bool relu, double p_keep,
                                   int64_t seed, int64_t offset,
-                                  const long* offset_dev) {
+                                  const long* offset_dev,
+                                  const c10::optional<torch::Tensor>& wT) {
   CHECK_CUDA(x); CHECK_BF16(x); CHECK_CONTIG(x);
   CHECK_BF16(w); CHECK_CONTIG(w);
   CHECK_F32(b); CHECK_CONTIG(b);
@@ -47,11 +48,11 @@ torch::Tensor linear_act_fwd_impl(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(w.size(0) == K, "w/x shape mismatch");
   auto y = torch::empty({M, N}, x.options());
   GemmParams p{};
-  p.A = bf16_ptr(x); p.B = bf16_ptr(w);
+  p.A = bf16_ptr(x);
   p.bias = b.data_ptr<float>();
   p.C = y.data_ptr(); p.amax = nullptr;
   p.M = M; p.N = N; p.K = K;
-  p.lda = K; p.ldb = N; p.ldc = N;
+  p.lda = K; p.ldc = N;
   p.splitk = 1;
   p.p_keep = (float)p_keep; p.seed = (uint64_t)seed; p.offset = (uint64_t)offset;
   p.offset_dev = offset_dev;
@@ -59,26 +60,40 @@ torch::Tensor linear_act_fwd_impl(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(!drop || relu, "dropout path requires relu epilogue");
   bool big = cdiv(M, 128) * cdiv(N, 128) >= 128;
   auto s = cur_stream();
-  if (drop) (big ? gemm_fwd_drop_128 : gemm_fwd_drop_64)(p, s);
-  else if (relu) (big ? gemm_fwd_relu_128 : gemm_fwd_relu_64)(p, s);
-  else (big ? gemm_fwd_bias_128 : gemm_fwd_bias_64)(p, s);
+  bool bt = wT.has_value() && wT->defined();
+  if (bt) {
+    // pre-transposed weight copy [N][K] -> vector B staging (B_NMAJ)
+    CHECK_BF16((*wT)); CHECK_CONTIG((*wT));
+    TORCH_CHECK(wT->size(0) == N && wT->size(1) == K, "wT shape mismatch");
+    p.B = bf16_ptr(*wT); p.ldb = K;
+    if (drop) (big ? gemm_fwd_drop_128_bt : gemm_fwd_drop_64_bt)(p, s);
+    else if (relu) (big ? gemm_fwd_relu_128_bt : gemm_fwd_relu_64_bt)(p, s);
+    else (big ? gemm_fwd_bias_128_bt : gemm_fwd_bias_64_bt)(p, s);
+  } else {
+    p.B = bf16_ptr(w); p.ldb = N;
+    if (drop) (big ? gemm_fwd_drop_128 : gemm_fwd_drop_64)(p, s);
+    else if (relu) (big ? gemm_fwd_relu_128 : gemm_fwd_relu_64)(p, s);
+    else (big ? gemm_fwd_bias_128 : gemm_fwd_bias_64)(p, s);
+  }
   return y;
 }
 
 torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                              bool relu, double p_keep, int64_t seed,
-                             int64_t offset) {
-  return linear_act_fwd_impl(x, w, b, relu, p_keep, seed, offset, nullptr);
+                             int64_t offset,
+                             c10::optional<torch::Tensor> wT) {
+  return linear_act_fwd_impl(x, w, b, relu, p_keep, seed, offset, nullptr, wT);
 }
 
 // hipGraph-capturable variant: dropout RNG offset read from device memory
 torch::Tensor linear_act_fwd_dev(torch::Tensor x, torch::Tensor w,
                                  torch::Tensor b, bool relu, double p_keep,
-                                 int64_t seed, torch::Tensor offset_dev) {
+                                 int64_t seed, torch::Tensor offset_dev,
+                                 c10::optional<torch::Tensor> wT) {
   TORCH_CHECK(offset_dev.scalar_type() == at::kLong && offset_dev.is_cuda(),
               "offset_dev must be a cuda int64 tensor");
   return linear_act_fwd_impl(x, w, b, relu, p_keep, seed, 0,
-                             offset_dev.data_ptr<long>());
+                             offset_dev.data_ptr<long>(), wT);
 }
 
 // core backward; when dw/db are passed they are accumulated into
@@ -160,7 +175,8 @@ torch::Tensor linear_act_bwd_into(torch::Tensor dy, torch::Tensor x,
 
 // --------------------------------------------------------------------------
 std::vector<torch::Tensor> conv_pool_fwd(torch::Tensor x, torch::Tensor w,
-                                         torch::Tensor b) {
+                                         torch::Tensor b,
+                                         c10::optional<torch::Tensor> wT) {
   CHECK_CUDA(x); CHECK_BF16(x); CHECK_CONTIG(x);
   CHECK_BF16(w); CHECK_CONTIG(w);
   CHECK_F32(b); CHECK_CONTIG(b);
@@ -181,10 +197,20 @@ std::vector<torch::Tensor> conv_pool_fwd(torch::Tensor x, torch::Tensor w,
   p.CB = NB; p.CH = H; p.CW = W; p.CHo = Ho; p.CWo = Wo;
   p.Cin = Cin; p.Cout = Cout;
   auto s = cur_stream();
-  if (Cin == 1) conv1_fwd_pool(p, s);
-  else {
+  if (Cin == 1) {
+    // direct VALU kernel: K=25 is too small for MFMA to win here
+    launch_conv1_direct_fwd(bf16_ptr(x), bf16_ptr(w), b.data_ptr<float>(),
+                            bf16_mut(y), amax.data_ptr<uint8_t>(), NB, H, W,
+                            Cout, s);
+  } else {
     TORCH_CHECK(Cin % 8 == 0, "conv requires Cin==1 or Cin%8==0");
-    conv_fwd_pool(p, s);
+    if (wT.has_value() && wT->defined()) {
+      CHECK_BF16((*wT)); CHECK_CONTIG((*wT));
+      p.B = bf16_ptr(*wT); p.ldb = 25 * Cin;
+      conv_fwd_pool_bt(p, s);
+    } else {
+      conv_fwd_pool(p, s);
+    }
   }
   return {y, amax};
 }
@@ -219,8 +245,13 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
     p.M = 25 * Cin; p.N = Cout; p.K = NB * H * W;
     p.ldb = Cout; p.ldc = Cout;
     p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
-    p.splitk = pick_splitk(cdiv(p.M, 64), cdiv(p.N, 64), cdiv(p.K, 32));
-    conv_dw_gemm(p, s);
+    if (Cin == 1) {
+      p.splitk = pick_splitk(cdiv(p.M, 32), cdiv(p.N, 32), cdiv(p.K, 64));
+      conv1_dw_gemm(p, s);
+    } else {
+      p.splitk = pick_splitk(cdiv(p.M, 128), cdiv(p.N, 64), cdiv(p.K, 64));
+      conv_dw_gemm(p, s);
+    }
   }
   torch::Tensor dx;
   if (need_dx) {
@@ -301,6 +332,15 @@ void sgd_step_dev(torch::Tensor master, torch::Tensor grad,
                       cur_stream());
 }
 
+void transpose_bf16(torch::Tensor src, torch::Tensor dst) {
+  CHECK_CUDA(src); CHECK_BF16(src); CHECK_CONTIG(src);
+  CHECK_BF16(dst); CHECK_CONTIG(dst);
+  TORCH_CHECK(src.dim() == 2 && dst.size(0) == src.size(1) &&
+              dst.size(1) == src.size(0), "transpose shape mismatch");
+  launch_transpose_bf16(bf16_ptr(src), bf16_mut(dst), src.size(0),
+                        src.size(1), cur_stream());
+}
+
 void step_advance(torch::Tensor step_dev, torch::Tensor lr_scale_dev,
                   double lr0, double decay, int64_t decay_steps,
                   double inv_contrib) {
@@ -313,11 +353,16 @@ void step_advance(torch::Tensor step_dev, torch::Tensor lr_scale_dev,
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("linear_act_fwd", &linear_act_fwd, "fused linear+bias+relu+dropout");
+  m.def("linear_act_fwd", &linear_act_fwd, "fused linear+bias+relu+dropout",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("relu"),
+        py::arg("p_keep"), py::arg("seed"), py::arg("offset"),
+        py::arg("wT") = c10::nullopt);
+  m.def("transpose_bf16", &transpose_bf16, "bf16 2-D transpose (wT refresh)");
   m.def("linear_act_bwd", &linear_act_bwd, "linear backward (dx, dw, db)");
   m.def("linear_act_bwd_into", &linear_act_bwd_into,
         "linear backward accumulating dw/db into bucket views");
-  m.def("conv_pool_fwd", &conv_pool_fwd, "fused conv5x5+bias+relu+maxpool");
+  m.def("conv_pool_fwd", &conv_pool_fwd, "fused conv5x5+bias+relu+maxpool",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("wT") = c10::nullopt);
   m.def("conv_pool_bwd", &conv_pool_bwd, "conv+pool backward (dx, dw, db)");
   m.def("conv_pool_bwd_into", &conv_pool_bwd_into,
         "conv backward accumulating dw/db into bucket views");
@@ -328,5 +373,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("step_advance", &step_advance,
         "device-side staircase LR + step increment (inside the graph)");
   m.def("linear_act_fwd_dev", &linear_act_fwd_dev,
-        "linear fwd with device-side dropout offset");
+        "linear fwd with device-side dropout offset",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("relu"),
+        py::arg("p_keep"), py::arg("seed"), py::arg("offset_dev"),
+        py::arg("wT") = c10::nullopt);
 }

@@ -404,8 +404,20 @@ GEMM_ENTRY(gemm_dx_128, 128, 128, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dx_64, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
 GEMM_ENTRY(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+// B-transposed (pre-transposed weight) fwd variants: ldb = K, vector staging
+GEMM_ENTRY(gemm_fwd_bias_128_bt, 128, 128, A_N, B_NMAJ, EPI_BIAS, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_bias_64_bt, 64, 64, A_N, B_NMAJ, EPI_BIAS, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_relu_128_bt, 128, 128, A_N, B_NMAJ, EPI_BIAS_RELU, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_relu_64_bt, 64, 64, A_N, B_NMAJ, EPI_BIAS_RELU, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_drop_128_bt, 128, 128, A_N, B_NMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
+GEMM_ENTRY(gemm_fwd_drop_64_bt, 64, 64, A_N, B_NMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
 // conv gathers: thousands of WGs -> single-buffer, TLP hides latency
 GEMM_ENTRY(conv_fwd_pool, 128, 64, A_CONV_FWD, B_KMAJ, EPI_POOL, OUT_BF16, 0)
+GEMM_ENTRY(conv_fwd_pool_bt, 128, 64, A_CONV_FWD, B_NMAJ, EPI_POOL, OUT_BF16, 0)
 GEMM_ENTRY(conv1_fwd_pool, 128, 64, A_CONV1_FWD, B_KMAJ, EPI_POOL, OUT_BF16, 0)
 GEMM_ENTRY(conv_dx_gemm, 128, 32, A_CONV_DX, B_CONV_DX_W, EPI_NONE, OUT_BF16, 0)
-GEMM_ENTRY(conv_dw_gemm, 64, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)
+// conv dW: moderate grids (re-reads dact per M-tile: BM=128 halves traffic
+// vs 64) -> pipelined
+GEMM_ENTRY(conv_dw_gemm, 128, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+// conv1 dW: M=25 -> 32x32 tile (78% M-utilization vs 39% at BM=64)
+GEMM_ENTRY(conv1_dw_gemm, 32, 32, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)

@@ -57,3 +57,18 @@ void launch_sgd_step_dev(float* master, const float* grad,
 void launch_step_advance(long* step_dev, float* lr_scale_dev, float lr0,
                          float decay, int decay_steps, float inv_contrib,
                          hipStream_t);
+void launch_transpose_bf16(const unsigned short* src, unsigned short* dst,
+                           int R, int C, hipStream_t);
+void launch_conv1_direct_fwd(const unsigned short* x, const unsigned short* w,
+                             const float* bias, unsigned short* y,
+                             uint8_t* amax, int NB, int H, int W, int Cout,
+                             hipStream_t);
+// B-transposed (pre-transposed weight) forward GEMM entries
+void gemm_fwd_bias_128_bt(const GemmParams&, hipStream_t);
+void gemm_fwd_bias_64_bt(const GemmParams&, hipStream_t);
+void gemm_fwd_relu_128_bt(const GemmParams&, hipStream_t);
+void gemm_fwd_relu_64_bt(const GemmParams&, hipStream_t);
+void gemm_fwd_drop_128_bt(const GemmParams&, hipStream_t);
+void gemm_fwd_drop_64_bt(const GemmParams&, hipStream_t);
+void conv_fwd_pool_bt(const GemmParams&, hipStream_t);
+void conv1_dw_gemm(const GemmParams&, hipStream_t);

@@ -95,6 +95,7 @@ class GraphedStep:
                                fp.flat_shadow is not None,
                                self.lr_scale_dev, self.dc_keep,
                                t.flags.seed, self.step_dev)
+        fp.refresh_transposes()
         # advance step + LR on-device for the next replay
         self._ext.step_advance(self.step_dev, self.lr_scale_dev,
                                t.flags.initial_learning_rate,

@@ -188,6 +188,7 @@ class Trainer:
                                            if flags.drop_connect else None),
                         seed=flags.seed, offset=self.step,
                         shadow=self.fp.flat_shadow)
+            self.fp.refresh_transposes()
             self.num_contributors = contributors
         self.step += 1
         return applied, loss.detach(), acc.detach(), compute_time

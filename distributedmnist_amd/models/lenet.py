@@ -35,6 +35,9 @@ def trunc_normal_(t: torch.Tensor, std: float, gen: torch.Generator):
 class LeNet5(nn.Module):
     PARAM_ORDER = ["conv1_w", "conv1_b", "conv2_w", "conv2_b",
                    "fc1_w", "fc1_b", "fc2_w", "fc2_b"]
+    # forward-GEMM B operands kept as pre-transposed bf16 copies
+    # (conv1 uses the direct VALU kernel -> no transpose needed)
+    TRANSPOSED_WEIGHTS = ["conv2_w", "fc1_w", "fc2_w"]
 
     def __init__(self, seed: int = SEED, compute_dtype: torch.dtype = torch.float32):
         super().__init__()
@@ -52,6 +55,7 @@ class LeNet5(nn.Module):
         # bf16 shadow weights (set up by parallel.flatten.FlatParams on GPU);
         # on CPU/fp32 they alias the masters.
         self.shadows: dict[str, torch.Tensor] = {}
+        self.shadows_T: dict[str, torch.Tensor] = {}
         self.dropout_seed = seed
         self._step = 0
         self.training_direct_grads = True
@@ -67,6 +71,9 @@ class LeNet5(nn.Module):
 
     def _comp(self, name: str) -> torch.Tensor:
         return self.shadows.get(name, getattr(self, name))
+
+    def _compT(self, name: str):
+        return self.shadows_T.get(name)
 
     def _gout(self, name: str):
         """Direct-grad bucket view (GPU bf16 path): backward kernels
@@ -85,7 +92,8 @@ class LeNet5(nn.Module):
         h = Fx.conv_pool(h, self.conv2_w, self.conv2_b,
                          self._comp("conv2_w"), self.conv2_b, need_dx=True,
                          dw_out=self._gout("conv2_w"),
-                         db_out=self._gout("conv2_b"))
+                         db_out=self._gout("conv2_b"),
+                         w_t=self._compT("conv2_w"))
         h = h.reshape(h.shape[0], 7 * 7 * 64)
         p_keep = 0.5 if train else 1.0
         h = Fx.linear_act(h, self.fc1_w, self.fc1_b,
@@ -94,11 +102,13 @@ class LeNet5(nn.Module):
                           seed=self.dropout_seed, offset=self._step,
                           dw_out=self._gout("fc1_w"),
                           db_out=self._gout("fc1_b"),
-                          offset_dev=self._step_dev)
+                          offset_dev=self._step_dev,
+                          w_t=self._compT("fc1_w"))
         logits = Fx.linear_act(h, self.fc2_w, self.fc2_b,
                                self._comp("fc2_w"), self.fc2_b, relu=False,
                                dw_out=self._gout("fc2_w"),
-                               db_out=self._gout("fc2_b"))
+                               db_out=self._gout("fc2_b"),
+                               w_t=self._compT("fc2_w"))
         return logits
 
     def loss_and_accuracy(self, logits, labels):

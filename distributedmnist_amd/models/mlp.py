@@ -17,6 +17,7 @@ NUM_LABELS = 10
 
 class MLP(nn.Module):
     PARAM_ORDER = ["fc1_w", "fc1_b", "fc2_w", "fc2_b"]
+    TRANSPOSED_WEIGHTS = ["fc1_w", "fc2_w"]
 
     def __init__(self, seed: int = 66478, hidden: int = 512,
                  compute_dtype: torch.dtype = torch.float32):
@@ -29,6 +30,7 @@ class MLP(nn.Module):
         self.fc2_w = nn.Parameter(trunc_normal_(torch.empty(hidden, NUM_LABELS), 0.1, gen))
         self.fc2_b = nn.Parameter(torch.full((NUM_LABELS,), 0.1))
         self.shadows: dict[str, torch.Tensor] = {}
+        self.shadows_T: dict[str, torch.Tensor] = {}
         self.dropout_seed = seed
         self._step = 0
         self.training_direct_grads = True
@@ -44,6 +46,9 @@ class MLP(nn.Module):
     def _comp(self, name: str) -> torch.Tensor:
         return self.shadows.get(name, getattr(self, name))
 
+    def _compT(self, name: str):
+        return self.shadows_T.get(name)
+
     def _gout(self, name: str):
         if not self.shadows or not self.training_direct_grads:
             return None
@@ -55,11 +60,13 @@ class MLP(nn.Module):
         h = Fx.linear_act(h, self.fc1_w, self.fc1_b,
                           self._comp("fc1_w"), self.fc1_b, relu=True,
                           dw_out=self._gout("fc1_w"),
-                          db_out=self._gout("fc1_b"))
+                          db_out=self._gout("fc1_b"),
+                          w_t=self._compT("fc1_w"))
         return Fx.linear_act(h, self.fc2_w, self.fc2_b,
                              self._comp("fc2_w"), self.fc2_b, relu=False,
                              dw_out=self._gout("fc2_w"),
-                             db_out=self._gout("fc2_b"))
+                             db_out=self._gout("fc2_b"),
+                             w_t=self._compT("fc2_w"))
 
     def loss_and_accuracy(self, logits, labels):
         loss, correct = Fx.softmax_xent(logits, labels)

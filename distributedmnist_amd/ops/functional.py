@@ -35,9 +35,9 @@ class ConvPoolFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b, w_comp, b_comp, need_dx: bool,
-                dw_out, db_out):
+                dw_out, db_out, w_t):
         if _use_hip(x):
-            y, amax = _C.ext().conv_pool_fwd(x, w_comp, b_comp)
+            y, amax = _C.ext().conv_pool_fwd(x, w_comp, b_comp, w_t)
         else:
             y, amax = cpu_ref.conv_pool_fwd(x, w_comp, b_comp)
         ctx.save_for_backward(x, w_comp, y, amax)
@@ -62,14 +62,14 @@ class ConvPoolFn(torch.autograd.Function):
             dx, dw, db = cpu_ref.conv_pool_bwd(dy, x, w_comp, y, amax)
         if not ctx.need_dx:
             dx = None
-        return dx, dw, db, None, None, None, None, None
+        return dx, dw, db, None, None, None, None, None, None
 
 
 def conv_pool(x, w, b, w_comp=None, b_comp=None, need_dx=True,
-              dw_out=None, db_out=None):
+              dw_out=None, db_out=None, w_t=None):
     return ConvPoolFn.apply(x, w, b, w_comp if w_comp is not None else w,
                             b_comp if b_comp is not None else b, need_dx,
-                            dw_out, db_out)
+                            dw_out, db_out, w_t)
 
 
 class LinearActFn(torch.autograd.Function):
@@ -82,14 +82,15 @@ class LinearActFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b, w_comp, b_comp, relu: bool, p_keep: float,
-                seed: int, offset: int, dw_out, db_out, offset_dev):
+                seed: int, offset: int, dw_out, db_out, offset_dev, w_t):
         if _use_hip(x):
             if offset_dev is not None and p_keep < 1.0:
                 y = _C.ext().linear_act_fwd_dev(x, w_comp, b_comp, relu,
-                                                p_keep, seed, offset_dev)
+                                                p_keep, seed, offset_dev,
+                                                wT=w_t)
             else:
                 y = _C.ext().linear_act_fwd(x, w_comp, b_comp, relu, p_keep,
-                                            seed, offset)
+                                            seed, offset, wT=w_t)
         else:
             y = cpu_ref.linear_fwd(x, w_comp, b_comp, relu)
             if p_keep < 1.0:
@@ -130,15 +131,16 @@ class LinearActFn(torch.autograd.Function):
         if not ctx.needs_input_grad[0]:
             dx = None
         return (dx, dw, db, None, None, None, None, None, None, None, None,
-                None)
+                None, None)
 
 
 def linear_act(x, w, b, w_comp=None, b_comp=None, relu=False, p_keep=1.0,
-               seed=0, offset=0, dw_out=None, db_out=None, offset_dev=None):
+               seed=0, offset=0, dw_out=None, db_out=None, offset_dev=None,
+               w_t=None):
     return LinearActFn.apply(x, w, b, w_comp if w_comp is not None else w,
                              b_comp if b_comp is not None else b, relu,
                              float(p_keep), int(seed), int(offset),
-                             dw_out, db_out, offset_dev)
+                             dw_out, db_out, offset_dev, w_t)
 
 
 class SoftmaxXentFn(torch.autograd.Function):

@@ -52,8 +52,24 @@ class FlatParams:
                 n: self.flat_shadow[o:o + sz].view(shape)
                 for n, o, sz, shape in zip(names, self.offsets, self.numels, self.shapes)
             }
+            # pre-transposed copies of the forward-GEMM B operands: vector
+            # LDS staging instead of scatter transposes every K-step
+            self.shadow_T = {}
+            self._t_pairs = []
+            for n in getattr(model, "TRANSPOSED_WEIGHTS", []):
+                v = model.shadows[n]
+                v2 = v.reshape(-1, v.shape[-1])        # [K][N] k-major
+                t = torch.empty(v2.shape[1], v2.shape[0],
+                                dtype=compute_dtype, device=device)
+                self.shadow_T[n] = t
+                self._t_pairs.append((v2, t))
+            model.shadows_T = self.shadow_T
+            self.refresh_transposes()
         else:
             model.shadows = {}
+            model.shadows_T = {}
+            self.shadow_T = {}
+            self._t_pairs = []
 
     def zero_grad(self):
         self.flat_grad.zero_()
@@ -78,6 +94,19 @@ class FlatParams:
     def sync_shadow(self):
         if self.flat_shadow is not None:
             self.flat_shadow.copy_(self.flat_master.to(self.flat_shadow.dtype))
+            self.refresh_transposes()
+
+    def refresh_transposes(self):
+        if not self._t_pairs:
+            return
+        if self.flat_shadow.is_cuda:
+            from .. import _C
+            ext = _C.ext()
+            for v2, t in self._t_pairs:
+                ext.transpose_bf16(v2, t)
+        else:
+            for v2, t in self._t_pairs:
+                t.copy_(v2.t())
 
     def state_dict_params(self):
         return {n: self.flat_master[o:o + sz].view(shape).clone()
