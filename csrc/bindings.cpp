@@ -246,10 +246,13 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
     p.ldb = Cout; p.ldc = Cout;
     p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
     if (Cin == 1) {
-      p.splitk = pick_splitk(cdiv(p.M, 32), cdiv(p.N, 32), cdiv(p.K, 64));
+      // single 32x32 tile: target ~2048 blocks so the latency-bound gather
+      // k-chain is short and the chip stays full
+      p.splitk = std::min(cdiv(p.K, 64), 2048);
       conv1_dw_gemm(p, s);
     } else {
-      p.splitk = pick_splitk(cdiv(p.M, 128), cdiv(p.N, 64), cdiv(p.K, 64));
+      int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
+      p.splitk = std::min(cdiv(p.K, 64), std::max(1, 2048 / tiles));
       conv_dw_gemm(p, s);
     }
   }

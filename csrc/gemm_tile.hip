@@ -416,8 +416,9 @@ GEMM_ENTRY(conv_fwd_pool, 128, 64, A_CONV_FWD, B_KMAJ, EPI_POOL, OUT_BF16, 0)
 GEMM_ENTRY(conv_fwd_pool_bt, 128, 64, A_CONV_FWD, B_NMAJ, EPI_POOL, OUT_BF16, 0)
 GEMM_ENTRY(conv1_fwd_pool, 128, 64, A_CONV1_FWD, B_KMAJ, EPI_POOL, OUT_BF16, 0)
 GEMM_ENTRY(conv_dx_gemm, 128, 32, A_CONV_DX, B_CONV_DX_W, EPI_NONE, OUT_BF16, 0)
-// conv dW: moderate grids (re-reads dact per M-tile: BM=128 halves traffic
-// vs 64) -> pipelined
-GEMM_ENTRY(conv_dw_gemm, 128, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+// conv dW: gather-staged -> occupancy (PIPE=0) beats the in-wave pipeline
+// (measured 2069us PIPE=1 vs 1309 at BM=64 PIPE=0); BM=128 halves the
+// per-M-tile dact re-reads
+GEMM_ENTRY(conv_dw_gemm, 128, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)
 // conv1 dW: M=25 -> 32x32 tile (78% M-utilization vs 39% at BM=64)
-GEMM_ENTRY(conv1_dw_gemm, 32, 32, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+GEMM_ENTRY(conv1_dw_gemm, 32, 32, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)

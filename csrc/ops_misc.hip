@@ -333,16 +333,17 @@ extern "C" __global__ __launch_bounds__(256)
 void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
                              const float* bias, ushort_t* y, uint8_t* amax,
                              int NB, int H, int W, int Cout, int Mpool) {
-  __shared__ float win[8][6][6];
+  // 64 pooled pixels per block (8 per thread-slot); the 25 filter taps for
+  // this thread's Cout live in REGISTERS, windows in LDS (broadcast reads
+  // across the 32 co lanes); 100 fp32 FMA per pixel.
+  __shared__ float win[64][6][6];
   __shared__ float wl[25][32];
   const int Ho = H / 2, Wo = W / 2;
   const int tid = threadIdx.x;
-  const long q0 = (long)blockIdx.x * 8;
-  // stage weights: 25*Cout elements (Cout<=32)
+  const long q0 = (long)blockIdx.x * 64;
   for (int i = tid; i < 25 * Cout; i += 256)
     wl[i / Cout][i % Cout] = bf2f(w[i]);
-  // stage the 8 windows (6x6 each, zero-padded at image borders)
-  for (int i = tid; i < 8 * 36; i += 256) {
+  for (int i = tid; i < 64 * 36; i += 256) {
     int p = i / 36, e = i % 36;
     int wr = e / 6, wc = e % 6;
     long q = q0 + p;
@@ -359,29 +360,42 @@ void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
     win[p][wr][wc] = v;
   }
   __syncthreads();
-  const int co = tid % 32, p = tid / 32;
+  const int co = tid & 31, slot = tid >> 5;
   if (co >= Cout) return;
-  long q = q0 + p;
-  if (q >= Mpool) return;
-  float best = -1.0f / 0.0f;
-  int barg = 0;
-  float b = bias[co];
+  float wreg[25];
 #pragma unroll
-  for (int r = 0; r < 2; ++r)
+  for (int t = 0; t < 25; ++t) wreg[t] = wl[t][co];
+  const float b = bias[co];
 #pragma unroll
-    for (int cx = 0; cx < 2; ++cx) {
-      float acc = b;
+  for (int j = 0; j < 8; ++j) {
+    int p = slot * 8 + j;
+    long q = q0 + p;
+    if (q >= Mpool) break;
+    float wv[36];
 #pragma unroll
-      for (int kh = 0; kh < 5; ++kh)
+    for (int e = 0; e < 36; ++e) wv[e] = win[p][e / 6][e % 6];
+    float acc0 = b, acc1 = b, acc2 = b, acc3 = b;
 #pragma unroll
-        for (int kw = 0; kw < 5; ++kw)
-          acc += win[p][r + kh][cx + kw] * wl[kh * 5 + kw][co];
-      acc = acc > 0.f ? acc : 0.f;
-      int pos = r * 2 + cx;
-      if (acc > best) { best = acc; barg = pos; }
+    for (int kh = 0; kh < 5; ++kh)
+#pragma unroll
+      for (int kw = 0; kw < 5; ++kw) {
+        float wc_ = wreg[kh * 5 + kw];
+        acc0 += wv[kh * 6 + kw] * wc_;
+        acc1 += wv[kh * 6 + kw + 1] * wc_;
+        acc2 += wv[(kh + 1) * 6 + kw] * wc_;
+        acc3 += wv[(kh + 1) * 6 + kw + 1] * wc_;
+      }
+    float vals[4] = {acc0, acc1, acc2, acc3};
+    float best = -1.0f / 0.0f;
+    int barg = 0;
+#pragma unroll
+    for (int pz = 0; pz < 4; ++pz) {
+      float v = vals[pz] > 0.f ? vals[pz] : 0.f;
+      if (v > best) { best = v; barg = pz; }
     }
-  y[(size_t)q * Cout + co] = f2bf(best);
-  amax[(size_t)q * Cout + co] = (uint8_t)barg;
+    y[(size_t)q * Cout + co] = f2bf(best);
+    amax[(size_t)q * Cout + co] = (uint8_t)barg;
+  }
 }
 
 void launch_transpose_bf16(const unsigned short* src, unsigned short* dst,
@@ -396,7 +410,7 @@ void launch_conv1_direct_fwd(const unsigned short* x, const unsigned short* w,
                              uint8_t* amax, int NB, int H, int W, int Cout,
                              hipStream_t s) {
   long mpool = (long)NB * (H / 2) * (W / 2);
-  dim3 grid((mpool + 7) / 8);
+  dim3 grid((mpool + 63) / 64);
   hipLaunchKernelGGL(conv1_direct_fwd_kernel, grid, dim3(256), 0, s, x, w,
                      bias, y, amax, NB, H, W, Cout, (int)mpool);
 }

@@ -1,0 +1,51 @@
+"""bench.py driver-contract tests (CPU): single-proc and the torchrun
+launch pattern the driver uses for N>1."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _last_json_line(stdout):
+    for line in reversed(stdout.strip().splitlines()):
+        if line.startswith("{"):
+            return json.loads(line)
+    raise AssertionError(f"no JSON line in output:\n{stdout}")
+
+
+@pytest.mark.timeout(300)
+def test_bench_single_process():
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"), "--steps", "3",
+         "--warmup", "1", "--batch_size", "64"],
+        capture_output=True, text=True, timeout=240, cwd=ROOT)
+    assert out.returncode == 0, out.stderr
+    j = _last_json_line(out.stdout)
+    assert j["n_gpus"] == 1 and j["steps"] == 3 and j["warmup"] == 1
+    assert j["unit"] == "images/sec" and j["value"] > 0
+    assert j["scaling"] == "weak" and j["higher_is_better"] is True
+    assert j["config"]["global_batch"] == 64
+    assert j["config"]["parallelism"] == "dp1"
+    assert j["data"] == "synthetic"
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_2proc():
+    env = dict(os.environ)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29417",
+         os.path.join(ROOT, "bench.py"), "--gpus", "2", "--steps", "3",
+         "--warmup", "1", "--batch_size", "32"],
+        capture_output=True, text=True, timeout=540, cwd=ROOT, env=env)
+    assert out.returncode == 0, out.stderr[-3000:]
+    j = _last_json_line(out.stdout)
+    assert j["n_gpus"] == 2
+    assert j["config"]["global_batch"] == 64
+    assert j["config"]["parallelism"] == "dp2"
