@@ -202,6 +202,11 @@ std::vector<torch::Tensor> conv_pool_fwd(torch::Tensor x, torch::Tensor w,
     launch_conv1_direct_fwd(bf16_ptr(x), bf16_ptr(w), b.data_ptr<float>(),
                             bf16_mut(y), amax.data_ptr<uint8_t>(), NB, H, W,
                             Cout, s);
+  } else if (conv_slab_supported(H, W, Cin, Cout)) {
+    // per-image LDS slab: activations cross the fabric once (not 25x)
+    launch_conv_fwd_slab(bf16_ptr(x), bf16_ptr(w), b.data_ptr<float>(),
+                         bf16_mut(y), amax.data_ptr<uint8_t>(), NB, H, W,
+                         Cin, Cout, s);
   } else {
     TORCH_CHECK(Cin % 8 == 0, "conv requires Cin==1 or Cin%8==0");
     if (wT.has_value() && wT->defined()) {
@@ -257,7 +262,11 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
     }
   }
   torch::Tensor dx;
-  if (need_dx) {
+  if (need_dx && conv_slab_supported(H, W, Cin, Cout)) {
+    dx = torch::empty({NB, H, W, Cin}, x.options());
+    launch_conv_dx_slab(bf16_ptr(dact), bf16_ptr(w), bf16_mut(dx), NB, H, W,
+                        Cin, Cout, s);
+  } else if (need_dx) {
     TORCH_CHECK(Cout % 32 == 0, "conv_dx requires Cout%32==0");
     dx = torch::empty({NB, H, W, Cin}, x.options());
     GemmParams p{};

@@ -1,0 +1,288 @@
+// Per-image LDS-slab conv kernels (gfx950).
+//
+// The implicit-GEMM conv kernels in gemm_tile.hip gather their A operand
+// from global memory once per (m, k) — the 5x5 im2col expansion re-reads
+// every activation element 25x (L3-served, latency-bound: conv_dx measured
+// 658us at B=8192).  These kernels instead stage ONE image (+2-pixel halo,
+// zero-padded) in LDS and feed the MFMA A-fragments DIRECTLY from the slab
+// at shifted offsets — each activation byte crosses the fabric exactly
+// once, and the 25 filter taps become LDS address arithmetic.
+//
+// Geometry (LeNet conv2, H=W=14): BM=224 rows = one image's 196 outputs
+// (pool-grouped for fwd / input pixels for dX) + pad; 4 waves as 2x2,
+// wave tile 112x(BN/2); one block per image.
+
+#include "common.h"
+#include "kernels.h"
+
+#define NTHREADS 256
+#define BK 64
+#define LDK (BK + 8)
+
+typedef ushort_t u16;
+
+// ---------------------------------------------------------------------------
+// conv2 fwd: y = maxpool(relu(conv5x5(x) + b)), one image per block.
+// x slab: [H+4][W+4][Cin] bf16, zero halo. K = 25*Cin, BK = 64.
+// ---------------------------------------------------------------------------
+template <int H, int W, int CIN, int COUT>
+__global__ __launch_bounds__(NTHREADS)
+void conv_fwd_slab_kernel(const u16* __restrict__ x,
+                          const u16* __restrict__ w,  // [25*CIN][COUT] k-major
+                          const float* __restrict__ bias, u16* __restrict__ y,
+                          uint8_t* __restrict__ amax, int NB) {
+  constexpr int HP = H + 4, WP = W + 4;
+  constexpr int HO = H / 2, WO = W / 2;
+  constexpr int M = HO * WO * 4;       // pool-grouped outputs of one image
+  constexpr int BM = 224;
+  constexpr int BN = COUT;             // 64
+  constexpr int WM = 112, WN = BN / 2;
+  constexpr int MI = WM / 16, NI = WN / 16;
+  constexpr int K = 25 * CIN;
+  __shared__ __align__(16) u16 slab[HP * WP * CIN];
+  __shared__ __align__(16) u16 Bs[BN][LDK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int img = blockIdx.x;
+  const u16* xi = x + (size_t)img * H * W * CIN;
+
+  // stage padded slab (zero halo): chunks of 8 ci
+  for (int c = tid; c < HP * WP * (CIN / 8); c += NTHREADS) {
+    int ci = (c % (CIN / 8)) * 8;
+    int pix = c / (CIN / 8);
+    int xx = pix % WP, yy = pix / WP;
+    short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+    int sy = yy - 2, sx = xx - 2;
+    if (sy >= 0 && sy < H && sx >= 0 && sx < W)
+      v = *reinterpret_cast<const short8*>(xi + ((size_t)sy * W + sx) * CIN + ci);
+    *reinterpret_cast<short8*>(&slab[(yy * WP + xx) * CIN + ci]) = v;
+  }
+
+  // per-lane A-fragment slab offsets for this wave's MI row-fragments:
+  // row lm -> conv-output pixel, pool-grouped (lm = q*4 + pos)
+  int arow_off[MI];  // slab element offset of (oy, ox) for each frag row
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+    int lm = wr * WM + mi * 16 + (lane & 15);
+    int q = lm >> 2, pos = lm & 3;
+    int wo = q % WO, ho = (q / WO) % HO;
+    if (lm >= M) { wo = 0; ho = 0; pos = 0; }
+    int oy = ho * 2 + (pos >> 1) + 2;   // +2: padded coords
+    int ox = wo * 2 + (pos & 1) + 2;
+    arow_off[mi] = (oy * WP + ox) * CIN;
+  }
+
+  f32x4 acc[MI][NI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int kq = (lane >> 4) * 8;
+  __syncthreads();
+
+  for (int kt = 0; kt < K; kt += BK) {
+    // stage B tile [BK][BN] -> Bs[n][kk] (scatter-transpose, w is k-major)
+    for (int c = tid; c < BK * (BN / 8); c += NTHREADS) {
+      int kk = c / (BN / 8);
+      int j0 = (c % (BN / 8)) * 8;
+      int k = kt + kk;
+      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (k < K)
+        v = *reinterpret_cast<const short8*>(w + (size_t)k * COUT + j0);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) Bs[j0 + e][kk] = v[e];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kh2 = 0; kh2 < 2; ++kh2) {
+      int kbase = kt + kh2 * 32 + kq;       // k = khkw*CIN + ci
+      int khkw = kbase / CIN, ci = kbase % CIN;
+      int kh = khkw / 5, kw = khkw % 5;
+      int shift = ((kh - 2) * WP + (kw - 2)) * CIN + ci;
+      short8 af[MI], bf[NI];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+        af[mi] = *reinterpret_cast<const short8*>(
+            &slab[arow_off[mi] + shift]);
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni)
+        bf[ni] = *reinterpret_cast<const short8*>(
+            &Bs[wc * WN + ni * 16 + (lane & 15)][kh2 * 32 + kq]);
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // pool epilogue (per-lane 4-register max, frow multiple of 4)
+  const int frow = (lane >> 4) * 4;
+  const int fcol = lane & 15;
+  u16* yi = y + (size_t)img * (M / 4) * COUT;
+  uint8_t* ai = amax + (size_t)img * (M / 4) * COUT;
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+    int lm = wr * WM + mi * 16 + frow;
+    if (lm >= M) continue;
+    int gq = lm >> 2;
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) {
+      int gc = wc * WN + ni * 16 + fcol;
+      float bias_v = bias[gc];
+      float best = -1.0f / 0.0f;
+      int barg = 0;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[mi][ni][r] + bias_v;
+        v = v > 0.f ? v : 0.f;
+        if (v > best) { best = v; barg = r; }
+      }
+      yi[(size_t)gq * COUT + gc] = f2bf(best);
+      ai[(size_t)gq * COUT + gc] = (uint8_t)barg;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// conv dX: dx[hi,wi,ci] = sum_{kh,kw,co} dact[hi-kh+2, wi-kw+2, co] *
+//          w[kh,kw,ci,co] — dact slab per image, W read from global per tile.
+// ---------------------------------------------------------------------------
+template <int H, int W, int CIN, int COUT>
+__global__ __launch_bounds__(NTHREADS)
+void conv_dx_slab_kernel(const u16* __restrict__ dact,
+                         const u16* __restrict__ w,  // [25*CIN][COUT]
+                         u16* __restrict__ dx, int NB) {
+  constexpr int HP = H + 4, WP = W + 4;
+  constexpr int M = H * W;
+  constexpr int BM = 224;
+  constexpr int BN = CIN;              // 32
+  constexpr int WM = 112, WN = BN / 2; // 16
+  constexpr int MI = WM / 16, NI = WN / 16;  // 7, 1
+  constexpr int K = 25 * COUT;
+  __shared__ __align__(16) u16 slab[HP * WP * COUT];
+  __shared__ __align__(16) u16 Bs[BN][LDK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int img = blockIdx.x;
+  const u16* di = dact + (size_t)img * H * W * COUT;
+
+  for (int c = tid; c < HP * WP * (COUT / 8); c += NTHREADS) {
+    int co = (c % (COUT / 8)) * 8;
+    int pix = c / (COUT / 8);
+    int xx = pix % WP, yy = pix / WP;
+    short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+    int sy = yy - 2, sx = xx - 2;
+    if (sy >= 0 && sy < H && sx >= 0 && sx < W)
+      v = *reinterpret_cast<const short8*>(di + ((size_t)sy * W + sx) * COUT + co);
+    *reinterpret_cast<short8*>(&slab[(yy * WP + xx) * COUT + co]) = v;
+  }
+
+  int arow_off[MI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+    int lm = wr * WM + mi * 16 + (lane & 15);
+    int wi = lm % W, hi = (lm / W) % H;
+    if (lm >= M) { wi = 0; hi = 0; }
+    arow_off[mi] = ((hi + 2) * WP + (wi + 2)) * COUT;  // padded (hi,wi)
+  }
+
+  f32x4 acc[MI][NI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int kq = (lane >> 4) * 8;
+  __syncthreads();
+
+  for (int kt = 0; kt < K; kt += BK) {
+    // Bs[ci][kk] = w[((khkw)*CIN + ci)*COUT + co(k)] — rows16 from w slices
+    for (int c = tid; c < BN * (BK / 8); c += NTHREADS) {
+      int i = c / (BK / 8);
+      int kc = (c % (BK / 8)) * 8;
+      int k = kt + kc;
+      int khkw = k / COUT, co = k % COUT;
+      short8 v = *reinterpret_cast<const short8*>(
+          w + ((size_t)khkw * CIN + i) * COUT + co);
+      *reinterpret_cast<short8*>(&Bs[i][kc]) = v;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kh2 = 0; kh2 < 2; ++kh2) {
+      int kbase = kt + kh2 * 32 + kq;       // k = khkw*COUT + co
+      int khkw = kbase / COUT, co = kbase % COUT;
+      int kh = khkw / 5, kw = khkw % 5;
+      // dact[hi - kh + 2, wi - kw + 2] -> padded offset shift
+      int shift = (-(kh - 2) * WP - (kw - 2)) * COUT + co;
+      short8 af[MI], bf[NI];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+        af[mi] = *reinterpret_cast<const short8*>(
+            &slab[arow_off[mi] + shift]);
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni)
+        bf[ni] = *reinterpret_cast<const short8*>(
+            &Bs[wc * WN + ni * 16 + (lane & 15)][kh2 * 32 + kq]);
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int frow = (lane >> 4) * 4;
+  const int fcol = lane & 15;
+  u16* xo = dx + (size_t)img * M * CIN;
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int lm = wr * WM + mi * 16 + frow + r;
+      if (lm >= M) continue;
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni) {
+        int gc = wc * WN + ni * 16 + fcol;
+        xo[(size_t)lm * CIN + gc] = f2bf(acc[mi][ni][r]);
+      }
+    }
+  }
+}
+
+// ---- host wrappers --------------------------------------------------------
+void launch_conv_fwd_slab(const unsigned short* x, const unsigned short* w,
+                          const float* bias, unsigned short* y, uint8_t* amax,
+                          int NB, int H, int W, int Cin, int Cout,
+                          hipStream_t s) {
+  if (H == 14 && W == 14 && Cin == 32 && Cout == 64) {
+    hipLaunchKernelGGL((conv_fwd_slab_kernel<14, 14, 32, 64>), dim3(NB),
+                       dim3(NTHREADS), 0, s, x, w, bias, y, amax, NB);
+  } else {
+    // unsupported geometry falls back at the binding level
+  }
+}
+
+bool conv_slab_supported(int H, int W, int Cin, int Cout) {
+  return H == 14 && W == 14 && Cin == 32 && Cout == 64;
+}
+
+void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
+                         unsigned short* dx, int NB, int H, int W, int Cin,
+                         int Cout, hipStream_t s) {
+  if (H == 14 && W == 14 && Cin == 32 && Cout == 64) {
+    hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64>), dim3(NB),
+                       dim3(NTHREADS), 0, s, dact, w, dx, NB);
+  }
+}

@@ -72,3 +72,11 @@ void gemm_fwd_drop_128_bt(const GemmParams&, hipStream_t);
 void gemm_fwd_drop_64_bt(const GemmParams&, hipStream_t);
 void conv_fwd_pool_bt(const GemmParams&, hipStream_t);
 void conv1_dw_gemm(const GemmParams&, hipStream_t);
+// conv_slab.hip — per-image LDS-slab conv kernels (H=W=14, Cin=32, Cout=64)
+bool conv_slab_supported(int H, int W, int Cin, int Cout);
+void launch_conv_fwd_slab(const unsigned short* x, const unsigned short* w,
+                          const float* bias, unsigned short* y, uint8_t* amax,
+                          int NB, int H, int W, int Cin, int Cout, hipStream_t);
+void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
+                         unsigned short* dx, int NB, int H, int W, int Cin,
+                         int Cout, hipStream_t);

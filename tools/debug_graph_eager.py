@@ -1,4 +1,7 @@
 """Lockstep graph-vs-eager divergence probe (run on a GPU box)."""
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from distributedmnist_amd.engine.train import Trainer, make_dataset
 from distributedmnist_amd.utils.flags import build_train_parser
