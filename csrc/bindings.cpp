@@ -255,6 +255,9 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
       // k-chain is short and the chip stays full
       p.splitk = std::min(cdiv(p.K, 64), 2048);
       conv1_dw_gemm(p, s);
+    } else if (conv_slab_supported(H, W, Cin, Cout)) {
+      launch_conv_dw_slab(bf16_ptr(x), bf16_ptr(dact), dw.data_ptr<float>(),
+                          NB, H, W, Cin, Cout, s);
     } else {
       int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
       p.splitk = std::min(cdiv(p.K, 64), std::max(1, 2048 / tiles));

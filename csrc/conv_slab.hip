@@ -286,3 +286,157 @@ void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
                        dim3(NTHREADS), 0, s, dact, w, dx, NB);
   }
 }
+
+// ---------------------------------------------------------------------------
+// conv dW (Cin=32, Cout=64): dW[kh,kw,ci,co] = sum_{n,h,w} x[n,h+kh-2,w+kw-2,ci]
+//   * dact[n,h,w,co], computed TRANSPOSED (M'=co=64, N'=(khkw,ci)=800) so the
+//   A operand (dact^T slab, pixel-contiguous per co row) reads as aligned
+//   ds_read_b128 fragments.
+//
+// Padded-linear trick: the reduction k enumerates the PADDED 18x18 pixel
+// space (halo rows/cols of dact^T are zero, so halo k contribute nothing).
+// The x operand address is then LINEAR in k: a = (kh*18 + kw) + k, over an
+// x slab stored per-ci with implicit row stride 18 covering rows -4..17 and
+// cols -4..13 (entry (r,c) = x[r-4][c-4], zero outside).  A column overflow
+// wp+kw >= 18 wraps to (row+1, col-18) whose col-4 < 0 is halo-ZERO — which
+// equals the true out-of-range-zero value, so the wrap is CORRECT by
+// construction (no per-element bounds checks on the hot path).
+//
+// Block: G=16 images accumulated into the same registers (flush atomics
+// /16); 2 N'-tiles of 448 (waves 1x4, wave tile 64x112, acc 4x7).
+// ---------------------------------------------------------------------------
+template <int H, int W, int CIN, int COUT, int G>
+__global__ __launch_bounds__(NTHREADS)
+void conv_dw_slab_kernel(const u16* __restrict__ x,
+                         const u16* __restrict__ dact,
+                         float* __restrict__ dw, int NB) {
+  constexpr int HP = H + 4, WP = W + 4;           // 18 x 18
+  constexpr int KPAD = 384;                       // 324 padded to 6 BK-steps
+  constexpr int XROW = 472;                       // max addr 466 + margin
+  constexpr int NP = 25 * CIN;                    // 800
+  constexpr int BN = 448;                         // 2 tiles
+  constexpr int WN = 112, NI = WN / 16;           // 7
+  constexpr int MI = COUT / 16;                   // 4 (WM = COUT = 64)
+  __shared__ __align__(16) u16 dslab[COUT][KPAD];
+  __shared__ __align__(16) u16 xslab[CIN][XROW];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wc = tid >> 6;  // wave = N position (waves 1x4)
+  const int img0 = blockIdx.x * G;
+
+  const int lcol = lane & 15;
+  const int kq = (lane >> 4) * 8;
+
+  for (int nt = 0; nt < 2; ++nt) {
+    f32x4 acc[MI][NI];
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    // per-lane B (x slab) base offsets for this tile's columns
+    int bshift[NI];
+    int bci[NI];
+    bool bvalid[NI];
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) {
+      int np = nt * BN + wc * WN + ni * 16 + lcol;
+      int khkw = np / CIN, ci = np % CIN;
+      bvalid[ni] = np < NP;
+      if (!bvalid[ni]) { khkw = 0; ci = 0; }
+      bci[ni] = ci;
+      bshift[ni] = (khkw / 5) * WP + (khkw % 5);
+    }
+
+    for (int g = 0; g < G; ++g) {
+      int img = img0 + g;
+      if (img >= NB) break;
+      // ---- stage slabs (zero-fill + transpose scatter) ----
+      __syncthreads();  // previous compute done before overwrite
+      for (int c = tid; c < COUT * (KPAD / 8); c += NTHREADS) {
+        *reinterpret_cast<short8*>(&dslab[c / (KPAD / 8)][(c % (KPAD / 8)) * 8]) =
+            short8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+      for (int c = tid; c < CIN * (XROW / 8); c += NTHREADS) {
+        *reinterpret_cast<short8*>(&xslab[c / (XROW / 8)][(c % (XROW / 8)) * 8]) =
+            short8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+      __syncthreads();
+      const u16* di = dact + (size_t)img * H * W * COUT;
+      const u16* xi = x + (size_t)img * H * W * CIN;
+      for (int c = tid; c < H * W * (COUT / 8); c += NTHREADS) {
+        int co0 = (c % (COUT / 8)) * 8;
+        int pix = c / (COUT / 8);
+        int w_ = pix % W, h_ = pix / W;
+        short8 v = *reinterpret_cast<const short8*>(di + (size_t)pix * COUT + co0);
+        int pk = (h_ + 2) * WP + (w_ + 2);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) dslab[co0 + e][pk] = v[e];
+      }
+      for (int c = tid; c < H * W * (CIN / 8); c += NTHREADS) {
+        int ci0 = (c % (CIN / 8)) * 8;
+        int pix = c / (CIN / 8);
+        int w_ = pix % W, h_ = pix / W;
+        short8 v = *reinterpret_cast<const short8*>(xi + (size_t)pix * CIN + ci0);
+        int pk = (h_ + 4) * WP + (w_ + 4);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) xslab[ci0 + e][pk] = v[e];
+      }
+      __syncthreads();
+      // ---- 6 K-steps over the padded pixel space ----
+      for (int kt = 0; kt < KPAD; kt += BK) {
+#pragma unroll
+        for (int kh2 = 0; kh2 < 2; ++kh2) {
+          int k0 = kt + kh2 * 32 + kq;
+          short8 af[MI], bf[NI];
+#pragma unroll
+          for (int mi = 0; mi < MI; ++mi)
+            af[mi] = *reinterpret_cast<const short8*>(
+                &dslab[mi * 16 + lcol][k0]);
+#pragma unroll
+          for (int ni = 0; ni < NI; ++ni) {
+            const u16* base = &xslab[bci[ni]][bshift[ni] + k0];
+#pragma unroll
+            for (int e = 0; e < 8; ++e) bf[ni][e] = (short)base[e];
+          }
+#pragma unroll
+          for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < NI; ++ni)
+              acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+    // ---- flush: dW'[co][n'] -> atomicAdd dW[(khkw*CIN+ci)*COUT + co] ----
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int co = mi * 16 + (lane >> 4) * 4 + r;   // C/D row map
+          int np = nt * BN + wc * WN + ni * 16 + lcol;
+          if (np >= NP) continue;
+          int khkw = np / CIN, ci = np % CIN;
+          float v = acc[mi][ni][r];
+          if (v != 0.f)
+            atomicAdd(dw + ((size_t)khkw * CIN + ci) * COUT + co, v);
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
+                         float* dw, int NB, int H, int W, int Cin, int Cout,
+                         hipStream_t s) {
+  if (H == 14 && W == 14 && Cin == 32 && Cout == 64) {
+    constexpr int G = 16;
+    hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, G>),
+                       dim3((NB + G - 1) / G), dim3(NTHREADS), 0, s, x, dact,
+                       dw, NB);
+  }
+}

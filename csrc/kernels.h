@@ -80,3 +80,6 @@ void launch_conv_fwd_slab(const unsigned short* x, const unsigned short* w,
 void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
                          unsigned short* dx, int NB, int H, int W, int Cin,
                          int Cout, hipStream_t);
+void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
+                         float* dw, int NB, int H, int W, int Cin, int Cout,
+                         hipStream_t);
