@@ -434,9 +434,19 @@ void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
                          float* dw, int NB, int H, int W, int Cin, int Cout,
                          hipStream_t s) {
   if (H == 14 && W == 14 && Cin == 32 && Cout == 64) {
-    constexpr int G = 16;
-    hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, G>),
-                       dim3((NB + G - 1) / G), dim3(NTHREADS), 0, s, x, dact,
-                       dw, NB);
+    // pick the group size so the grid has >= ~512 blocks (256 CUs need
+    // >>256 workgroups); bigger G amortizes the flush atomics
+    if (NB >= 8192) {
+      hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16>),
+                         dim3((NB + 15) / 16), dim3(NTHREADS), 0, s, x, dact,
+                         dw, NB);
+    } else if (NB >= 2048) {
+      hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 4>),
+                         dim3((NB + 3) / 4), dim3(NTHREADS), 0, s, x, dact,
+                         dw, NB);
+    } else {
+      hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 1>),
+                         dim3(NB), dim3(NTHREADS), 0, s, x, dact, dw, NB);
+    }
   }
 }
