@@ -389,23 +389,25 @@ void conv_dw_slab_kernel(const u16* __restrict__ x,
 #pragma unroll
         for (int kh2 = 0; kh2 < 2; ++kh2) {
           int k0 = kt + kh2 * 32 + kq;
-          short8 af[MI], bf[NI];
+          short8 af[MI];
 #pragma unroll
           for (int mi = 0; mi < MI; ++mi)
             af[mi] = *reinterpret_cast<const short8*>(
                 &dslab[mi * 16 + lcol][k0]);
+          // B fragments just-in-time (one live at a time — the 8 unaligned
+          // u16 reads per fragment otherwise blow register liveness to
+          // 1 wave/SIMD)
 #pragma unroll
           for (int ni = 0; ni < NI; ++ni) {
             const u16* base = &xslab[bci[ni]][bshift[ni] + k0];
+            short8 bf;
 #pragma unroll
-            for (int e = 0; e < 8; ++e) bf[ni][e] = (short)base[e];
-          }
+            for (int e = 0; e < 8; ++e) bf[e] = (short)base[e];
 #pragma unroll
-          for (int mi = 0; mi < MI; ++mi)
-#pragma unroll
-            for (int ni = 0; ni < NI; ++ni)
+            for (int mi = 0; mi < MI; ++mi)
               acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+                  af[mi], bf, acc[mi][ni], 0, 0, 0);
+          }
         }
       }
     }
