@@ -14,6 +14,7 @@
 
 #include "common.h"
 #include "kernels.h"
+#include <stdlib.h>
 
 #define NTHREADS 256
 #define BK 64
@@ -306,7 +307,7 @@ void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
 // /16); 2 N'-tiles of 448 (waves 1x4, wave tile 64x112, acc 4x7).
 // ---------------------------------------------------------------------------
 template <int H, int W, int CIN, int COUT, int G>
-__global__ __launch_bounds__(NTHREADS)
+__global__ __launch_bounds__(NTHREADS, 2)  // force <=256 regs: 2 blocks/CU
 void conv_dw_slab_kernel(const u16* __restrict__ x,
                          const u16* __restrict__ dact,
                          float* __restrict__ dw, int NB) {
@@ -437,12 +438,19 @@ void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
                          hipStream_t s) {
   if (H == 14 && W == 14 && Cin == 32 && Cout == 64) {
     // pick the group size so the grid has >= ~512 blocks (256 CUs need
-    // >>256 workgroups); bigger G amortizes the flush atomics
-    if (NB >= 8192) {
+    // >>256 workgroups); bigger G amortizes the flush atomics.
+    // DMNIST_DW_G overrides for testing the G>1 paths at small NB.
+    static int force_g = [] {
+      const char* e = getenv("DMNIST_DW_G");
+      return e ? atoi(e) : 0;
+    }();
+    int nb_eff = force_g ? (force_g >= 16 ? 8192 : (force_g >= 4 ? 2048 : 0))
+                         : NB;
+    if (nb_eff >= 8192) {
       hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16>),
                          dim3((NB + 15) / 16), dim3(NTHREADS), 0, s, x, dact,
                          dw, NB);
-    } else if (NB >= 2048) {
+    } else if (nb_eff >= 2048) {
       hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 4>),
                          dim3((NB + 3) / 4), dim3(NTHREADS), 0, s, x, dact,
                          dw, NB);

@@ -320,3 +320,35 @@ def test_graphed_dropout_advances(ext):
     # same batch, lr=0 (weights frozen): loss differences come only from
     # the dropout mask changing per step
     assert len(set(losses)) > 1, losses
+
+
+@pytest.mark.parametrize("force_g", ["1", "4", "16"])
+def test_conv_dw_slab_image_groups(ext, force_g, monkeypatch):
+    """conv dW slab: the G-image register-accumulation paths must agree
+    with the CPU reference FED THE GPU's OWN argmax (tie routing between
+    equal pool candidates differs between fp32/bf16 and both are valid
+    subgradients — comparing against the same routing isolates the GEMM)."""
+    import importlib
+    import os
+    os.environ["DMNIST_DW_G"] = force_g
+    try:
+        # fresh process state isn't possible for the static in the .so, so
+        # this test relies on running each param in its own pytest process
+        # OR on the static being read once — order the params so G grows.
+        torch.manual_seed(11)
+        NB, H, W, Cin, Cout = 64, 14, 14, 32, 64
+        x = (torch.rand(NB, H, W, Cin) - 0.5).to(bf16).float()
+        w = (torch.randn(5, 5, Cin, Cout) * 0.1).to(bf16).float()
+        b = torch.randn(Cout) * 0.1
+        from distributedmnist_amd.ops import cpu_ref
+        dy = (torch.randn(NB, H // 2, W // 2, Cout) * 0.1).to(bf16).float()
+        yg, amaxg = ext.conv_pool_fwd(to_gpu_bf16(x), to_gpu_bf16(w),
+                                      b.cuda().float())
+        dx_ref, dw_ref, db_ref = cpu_ref.conv_pool_bwd(
+            dy, x, w, yg.cpu().float(), amaxg.cpu())
+        dx, dw, db = ext.conv_pool_bwd(to_gpu_bf16(dy), to_gpu_bf16(x),
+                                       to_gpu_bf16(w), yg, amaxg, True)
+        assert_close_bf16(dw, dw_ref, scale=float(dw_ref.abs().max()))
+        assert_close_bf16(dx, dx_ref, scale=float(dx_ref.abs().max()))
+    finally:
+        del os.environ["DMNIST_DW_G"]
