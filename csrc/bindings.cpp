@@ -250,13 +250,16 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
     p.M = 25 * Cin; p.N = Cout; p.K = NB * H * W;
     p.ldb = Cout; p.ldc = Cout;
     p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
-    if (Cin == 1) {
+    if (Cin == 1 && conv1_slab_supported(H, W, Cin, Cout)) {
+      launch_conv1_dw_slab(bf16_ptr(x), bf16_ptr(dact), dw.data_ptr<float>(),
+                           NB, H, W, Cout, s);
+    } else if (Cin == 1) {
       // single 32x32 tile: target ~2048 blocks so the latency-bound gather
       // k-chain is short and the chip stays full
       p.splitk = std::min(cdiv(p.K, 64), 2048);
       conv1_dw_gemm(p, s);
     } else if (conv_slab_supported(H, W, Cin, Cout) &&
-               (NB >= 2048 || getenv("DMNIST_DW_G"))) {
+               (NB >= 1024 || getenv("DMNIST_DW_G"))) {
       // per-image-group slab dW: wins when the flush atomics amortize over
       // >=4 images/block; below that the implicit-GEMM form is faster
       launch_conv_dw_slab(bf16_ptr(x), bf16_ptr(dact), dw.data_ptr<float>(),
@@ -303,6 +306,121 @@ torch::Tensor conv_pool_bwd_into(torch::Tensor dy, torch::Tensor x,
   auto r = conv_pool_bwd_impl(dy, x, w, y, amax, need_dx,
                               dw_out.view_as(w), db_out);
   return r[0];
+}
+
+// --------------------------------------------------------------------------
+// Fine-grained backward pieces for the hand-scheduled fused step
+// (engine/fused_step.py): the composite linear_act_bwd/conv_pool_bwd above
+// stay for the autograd path; these let dW run on a side stream while the
+// dX chain proceeds.
+// --------------------------------------------------------------------------
+torch::Tensor mask_db(torch::Tensor dy, torch::Tensor y, bool relu,
+                      double p_keep, torch::Tensor db_out) {
+  CHECK_CUDA(dy); CHECK_BF16(dy); CHECK_CONTIG(dy);
+  CHECK_F32(db_out); CHECK_CONTIG(db_out);
+  int B = dy.size(0), N = dy.size(1);
+  auto s = cur_stream();
+  bool mask = relu || p_keep < 1.0;
+  torch::Tensor dyeff = mask ? torch::empty_like(dy) : dy;
+  launch_relu_drop_bwd(bf16_ptr(dy), mask ? bf16_ptr(y) : bf16_ptr(dy),
+                       bf16_mut(dyeff), db_out.data_ptr<float>(), B, N,
+                       (float)(1.0 / p_keep), mask ? 1 : 0, s);
+  return dyeff;
+}
+
+void linear_dw_into(torch::Tensor x, torch::Tensor dyeff,
+                    torch::Tensor dw_out) {
+  CHECK_BF16(x); CHECK_CONTIG(x); CHECK_BF16(dyeff); CHECK_CONTIG(dyeff);
+  CHECK_F32(dw_out);
+  int B = x.size(0), K = x.size(1), N = dyeff.size(1);
+  GemmParams p{};
+  p.A = bf16_ptr(x); p.B = bf16_ptr(dyeff);
+  p.C = dw_out.data_ptr();
+  p.M = K; p.N = N; p.K = B;
+  p.lda = K; p.ldb = N; p.ldc = N;
+  bool big = cdiv(K, 128) * cdiv(N, 128) >= 128;
+  int bm = big ? 128 : 64;
+  p.splitk = pick_splitk(cdiv(K, bm), cdiv(N, bm), cdiv(B, 32));
+  (big ? gemm_dw_128 : gemm_dw_64)(p, cur_stream());
+}
+
+torch::Tensor linear_dx(torch::Tensor dyeff, torch::Tensor w) {
+  CHECK_BF16(dyeff); CHECK_CONTIG(dyeff); CHECK_BF16(w); CHECK_CONTIG(w);
+  int B = dyeff.size(0), N = dyeff.size(1), K = w.size(0);
+  auto dx = torch::empty({B, K}, dyeff.options());
+  GemmParams p{};
+  p.A = bf16_ptr(dyeff); p.B = bf16_ptr(w);
+  p.C = dx.data_ptr();
+  p.M = B; p.N = K; p.K = N;
+  p.lda = N; p.ldb = N; p.ldc = K;
+  p.splitk = 1;
+  bool big = cdiv(B, 128) * cdiv(K, 128) >= 128;
+  (big ? gemm_dx_128 : gemm_dx_64)(p, cur_stream());
+  return dx;
+}
+
+torch::Tensor pool_scatter(torch::Tensor dy, torch::Tensor y,
+                           torch::Tensor amax, torch::Tensor db_out,
+                           int64_t H, int64_t W) {
+  CHECK_BF16(dy); CHECK_CONTIG(dy);
+  int NB = dy.size(0), Ho = dy.size(1), Wo = dy.size(2), C = dy.size(3);
+  auto dact = torch::empty({NB, (int)H, (int)W, C}, dy.options());
+  launch_pool_bwd_scatter(bf16_ptr(dy), bf16_ptr(y),
+                          amax.data_ptr<uint8_t>(), bf16_mut(dact),
+                          db_out.data_ptr<float>(), NB * Ho * Wo, C, H, W,
+                          Wo, cur_stream());
+  return dact;
+}
+
+void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
+  CHECK_BF16(x); CHECK_CONTIG(x); CHECK_BF16(dact); CHECK_CONTIG(dact);
+  int NB = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int Cout = dact.size(3);
+  auto s = cur_stream();
+  GemmParams p{};
+  p.A = bf16_ptr(x); p.B = bf16_ptr(dact);
+  p.C = dw_out.data_ptr();
+  p.M = 25 * Cin; p.N = Cout; p.K = NB * H * W;
+  p.ldb = Cout; p.ldc = Cout;
+  p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
+  if (Cin == 1 && conv1_slab_supported(H, W, Cin, Cout)) {
+    launch_conv1_dw_slab(bf16_ptr(x), bf16_ptr(dact),
+                         dw_out.data_ptr<float>(), NB, H, W, Cout, s);
+  } else if (Cin == 1) {
+    p.splitk = std::min(cdiv(p.K, 64), 2048);
+    conv1_dw_gemm(p, s);
+  } else if (conv_slab_supported(H, W, Cin, Cout) &&
+             (NB >= 1024 || getenv("DMNIST_DW_G"))) {
+    launch_conv_dw_slab(bf16_ptr(x), bf16_ptr(dact), dw_out.data_ptr<float>(),
+                        NB, H, W, Cin, Cout, s);
+  } else {
+    int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
+    p.splitk = std::min(cdiv(p.K, 64), std::max(1, 2048 / tiles));
+    conv_dw_gemm(p, s);
+  }
+}
+
+torch::Tensor conv_dx(torch::Tensor dact, torch::Tensor w, int64_t Cin) {
+  CHECK_BF16(dact); CHECK_CONTIG(dact); CHECK_BF16(w); CHECK_CONTIG(w);
+  int NB = dact.size(0), H = dact.size(1), W = dact.size(2),
+      Cout = dact.size(3);
+  auto s = cur_stream();
+  auto dx = torch::empty({NB, H, W, (int)Cin}, dact.options());
+  if (conv_slab_supported(H, W, Cin, Cout)) {
+    launch_conv_dx_slab(bf16_ptr(dact), bf16_ptr(w), bf16_mut(dx), NB, H, W,
+                        Cin, Cout, s);
+  } else {
+    TORCH_CHECK(Cout % 32 == 0, "conv_dx requires Cout%32==0");
+    GemmParams p{};
+    p.A = bf16_ptr(dact); p.B = bf16_ptr(w);
+    p.C = dx.data_ptr();
+    p.M = NB * H * W; p.N = Cin; p.K = 25 * Cout;
+    p.ldc = Cin;
+    p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
+    p.splitk = 1;
+    conv_dx_gemm(p, s);
+  }
+  return dx;
 }
 
 // --------------------------------------------------------------------------
@@ -376,6 +494,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("p_keep"), py::arg("seed"), py::arg("offset"),
         py::arg("wT") = c10::nullopt);
   m.def("transpose_bf16", &transpose_bf16, "bf16 2-D transpose (wT refresh)");
+  m.def("mask_db", &mask_db, "relu/dropout grad mask + bias-grad column sums");
+  m.def("linear_dw_into", &linear_dw_into, "dW = x^T dyeff into bucket view");
+  m.def("linear_dx", &linear_dx, "dx = dyeff @ W^T");
+  m.def("pool_scatter", &pool_scatter, "maxpool bwd scatter + conv db");
+  m.def("conv_dw_into", &conv_dw_into, "conv dW into bucket view");
+  m.def("conv_dx", &conv_dx, "conv dX");
   m.def("linear_act_bwd", &linear_act_bwd, "linear backward (dx, dw, db)");
   m.def("linear_act_bwd_into", &linear_act_bwd_into,
         "linear backward accumulating dw/db into bucket views");

@@ -450,7 +450,7 @@ void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
       hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16>),
                          dim3((NB + 15) / 16), dim3(NTHREADS), 0, s, x, dact,
                          dw, NB);
-    } else if (nb_eff >= 2048) {
+    } else if (nb_eff >= 1024) {
       hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 4>),
                          dim3((NB + 3) / 4), dim3(NTHREADS), 0, s, x, dact,
                          dw, NB);
@@ -459,4 +459,105 @@ void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
                          dim3(NB), dim3(NTHREADS), 0, s, x, dact, dw, NB);
     }
   }
+}
+
+// ---------------------------------------------------------------------------
+// conv1 dW (Cin=1): same padded-linear transposed formulation as above but
+// with the trivial N' = 25 taps (no ci): waves 2x2, one 16x16 fragment per
+// wave covers the whole 32(co) x 32(khkw-padded) output.
+// ---------------------------------------------------------------------------
+template <int H, int W, int COUT, int G>
+__global__ __launch_bounds__(NTHREADS, 2)
+void conv1_dw_slab_kernel(const u16* __restrict__ x,
+                          const u16* __restrict__ dact,
+                          float* __restrict__ dw, int NB) {
+  constexpr int HP = H + 4, WP = W + 4;            // 32 x 32
+  constexpr int KPAD = HP * WP;                    // 1024 = 16 BK-steps
+  constexpr int XROW = 1168;  // max addr (4*32+4) + 1023 + 8 margin
+  __shared__ __align__(16) u16 dslab[COUT][KPAD];
+  __shared__ __align__(16) u16 xslab[XROW];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;         // 2x2 over [32co][32khkw]
+  const int img0 = blockIdx.x * G;
+  const int lcol = lane & 15;
+  const int kq = (lane >> 4) * 8;
+
+  const int khkw = wc * 16 + lcol;
+  const int kh = khkw / 5, kw = khkw % 5;
+  const int bshift = (khkw < 25) ? kh * WP + kw : 0;
+
+  f32x4 acc = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int g = 0; g < G; ++g) {
+    int img = img0 + g;
+    if (img >= NB) break;
+    __syncthreads();
+    for (int c = tid; c < COUT * (KPAD / 8); c += NTHREADS)
+      *reinterpret_cast<short8*>(&dslab[c / (KPAD / 8)][(c % (KPAD / 8)) * 8]) =
+          short8{0, 0, 0, 0, 0, 0, 0, 0};
+    for (int c = tid; c < XROW / 8; c += NTHREADS)
+      *reinterpret_cast<short8*>(&xslab[c * 8]) = short8{0, 0, 0, 0, 0, 0, 0, 0};
+    __syncthreads();
+    const u16* di = dact + (size_t)img * H * W * COUT;
+    const u16* xi = x + (size_t)img * H * W;
+    for (int c = tid; c < H * W * (COUT / 8); c += NTHREADS) {
+      int co0 = (c % (COUT / 8)) * 8;
+      int pix = c / (COUT / 8);
+      int w_ = pix % W, h_ = pix / W;
+      short8 v = *reinterpret_cast<const short8*>(di + (size_t)pix * COUT + co0);
+      int pk = (h_ + 2) * WP + (w_ + 2);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) dslab[co0 + e][pk] = v[e];
+    }
+    for (int c = tid; c < H * W; c += NTHREADS) {
+      int w_ = c % W, h_ = c / W;
+      xslab[(h_ + 4) * WP + (w_ + 4)] = xi[c];
+    }
+    __syncthreads();
+    for (int kt = 0; kt < KPAD; kt += BK) {
+#pragma unroll
+      for (int kh2 = 0; kh2 < 2; ++kh2) {
+        int k0 = kt + kh2 * 32 + kq;
+        short8 af = *reinterpret_cast<const short8*>(
+            &dslab[wr * 16 + lcol][k0]);
+        const u16* base = &xslab[bshift + k0];
+        short8 bf;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) bf[e] = (short)base[e];
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+      }
+    }
+  }
+  // flush: D[co][khkw]
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int co = wr * 16 + (lane >> 4) * 4 + r;
+    if (khkw < 25) {
+      float v = acc[r];
+      if (v != 0.f) atomicAdd(dw + (size_t)khkw * COUT + co, v);
+    }
+  }
+}
+
+void launch_conv1_dw_slab(const unsigned short* x, const unsigned short* dact,
+                          float* dw, int NB, int H, int W, int Cout,
+                          hipStream_t s) {
+  if (H == 28 && W == 28 && Cout == 32) {
+    if (NB >= 4096) {
+      hipLaunchKernelGGL((conv1_dw_slab_kernel<28, 28, 32, 8>),
+                         dim3((NB + 7) / 8), dim3(NTHREADS), 0, s, x, dact,
+                         dw, NB);
+    } else {
+      hipLaunchKernelGGL((conv1_dw_slab_kernel<28, 28, 32, 2>),
+                         dim3((NB + 1) / 2), dim3(NTHREADS), 0, s, x, dact,
+                         dw, NB);
+    }
+  }
+}
+
+bool conv1_slab_supported(int H, int W, int Cin, int Cout) {
+  return H == 28 && W == 28 && Cin == 1 && Cout == 32;
 }

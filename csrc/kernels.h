@@ -83,3 +83,7 @@ void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
 void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
                          float* dw, int NB, int H, int W, int Cin, int Cout,
                          hipStream_t);
+bool conv1_slab_supported(int H, int W, int Cin, int Cout);
+void launch_conv1_dw_slab(const unsigned short* x, const unsigned short* dact,
+                          float* dw, int NB, int H, int W, int Cout,
+                          hipStream_t);

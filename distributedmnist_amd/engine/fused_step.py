@@ -1,0 +1,91 @@
+"""Hand-scheduled LeNet training step (GPU bf16 path).
+
+Replaces autograd's serial backward with an explicit schedule on two HIP
+streams: the dX chain stays on the main stream while every dW GEMM (which
+only the optimizer needs) runs concurrently on a side stream — ~30% of the
+backward is off the critical path.  Used inside the hipGraph capture
+(graphstep.py); stream forks/joins are recorded as graph dependencies.
+
+Backward dataflow (s0 = main, s1 = side):
+
+  dl ->[mask_db: db2]-> dyeff2 --s1--> dW2
+        |--s0--> dx2 ->[mask_db: db1]-> dyeff1 --s1--> dW1
+                  |--s0--> dx1 ->[pool_scatter: dbc2]-> dact2 --s1--> conv2 dW
+                            |--s0--> conv_dx -> [pool_scatter: dbc1]-> dact1
+                                      |--s0--> conv1 dW
+All dW/db land directly in the flat fp32 all-reduce bucket (pre-zeroed).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .. import _C
+
+
+class FusedLeNetStep:
+    def __init__(self, trainer):
+        t = trainer
+        m = t.model
+        from ..models import LeNet5
+        assert isinstance(m, LeNet5), "fused step is LeNet-specific"
+        assert t.compute_dtype == torch.bfloat16 and t.device.type == "cuda"
+        assert m.shadows and m.shadows_T, "fused step needs bf16 shadows"
+        self.t = t
+        self.ext = _C.ext()
+        self.side = torch.cuda.Stream()
+        self.p_keep = 0.5
+        self.seed = t.flags.seed
+
+    def __call__(self, x, labels, step_dev):
+        """One fwd+bwd; gradients land in the flat bucket.
+        Returns (loss, correct_count) device scalars."""
+        ext = self.ext
+        m = self.t.model
+        sh, shT = m.shadows, m.shadows_T
+        B = x.shape[0]
+
+        def gv(name):
+            return getattr(m, name).grad
+
+        s0 = torch.cuda.current_stream()
+        s1 = self.side
+
+        # ---- forward (s0) ----
+        y1, am1 = ext.conv_pool_fwd(x, sh["conv1_w"], m.conv1_b, None)
+        y2, am2 = ext.conv_pool_fwd(y1, sh["conv2_w"], m.conv2_b,
+                                    shT["conv2_w"])
+        h2 = y2.view(B, 7 * 7 * 64)
+        a1 = ext.linear_act_fwd_dev(h2, sh["fc1_w"], m.fc1_b, True,
+                                    self.p_keep, self.seed, step_dev,
+                                    shT["fc1_w"])
+        logits = ext.linear_act_fwd(a1, sh["fc2_w"], m.fc2_b, False, 1.0,
+                                    0, 0, shT["fc2_w"])
+        loss, correct, dl = ext.softmax_xent_fwd(logits, labels)
+
+        # ---- backward: dX chain on s0, dW GEMMs on s1 ----
+        dyeff2 = ext.mask_db(dl, dl, False, 1.0, gv("fc2_b"))
+        s1.wait_stream(s0)
+        with torch.cuda.stream(s1):
+            ext.linear_dw_into(a1, dyeff2, gv("fc2_w"))
+        dx2 = ext.linear_dx(dyeff2, sh["fc2_w"])
+
+        dyeff1 = ext.mask_db(dx2, a1, True, self.p_keep, gv("fc1_b"))
+        s1.wait_stream(s0)
+        with torch.cuda.stream(s1):
+            ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
+        dx1 = ext.linear_dx(dyeff1, sh["fc1_w"]).view(B, 7, 7, 64)
+
+        dact2 = ext.pool_scatter(dx1, y2, am2, gv("conv2_b"), 14, 14)
+        s1.wait_stream(s0)
+        with torch.cuda.stream(s1):
+            ext.conv_dw_into(y1, dact2, gv("conv2_w"))
+        dxc = ext.conv_dx(dact2, sh["conv2_w"], 32)
+
+        dact1 = ext.pool_scatter(dxc, y1, am1, gv("conv1_b"), 28, 28)
+        ext.conv_dw_into(x, dact1, gv("conv1_w"))
+
+        s0.wait_stream(s1)
+        # keep the side-stream consumers alive until the join (capture-safe)
+        self._keep = (a1, h2, dyeff1, dyeff2, y1, dact2)
+        return loss, correct

@@ -47,6 +47,15 @@ class GraphedStep:
         self.dc_keep = (flags.drop_connect_probability if flags.drop_connect
                         else -1.0)
         self._ext = ext
+        # hand-scheduled two-stream step for LeNet (dW off the dX chain);
+        # autograd body otherwise
+        self._fused = None
+        from ..models import LeNet5
+        if (isinstance(t.model, LeNet5) and t.model.shadows
+                and t.model.shadows_T
+                and getattr(flags, "fused_step", "auto") != "off"):
+            from .fused_step import FusedLeNetStep
+            self._fused = FusedLeNetStep(t)
 
         # snapshot state: the warmup iterations below really train
         master0 = t.fp.flat_master.clone()
@@ -84,9 +93,14 @@ class GraphedStep:
         t = self.t
         fp = t.fp
         fp.flat_grad.zero_()
-        logits = t.model(self.static_x, train=True)
-        loss, acc = t.model.loss_and_accuracy(logits, self.static_y)
-        loss.backward()
+        if self._fused is not None:
+            loss, correct = self._fused(self.static_x, self.static_y,
+                                        self.step_dev)
+            acc = correct / self.static_x.shape[0]
+        else:
+            logits = t.model(self.static_x, train=True)
+            loss, acc = t.model.loss_and_accuracy(logits, self.static_y)
+            loss.backward()
         if t.world > 1 and dist.is_initialized():
             dist.all_reduce(fp.flat_grad, op=dist.ReduceOp.SUM)
         self._ext.sgd_step_dev(fp.flat_master, fp.flat_grad,

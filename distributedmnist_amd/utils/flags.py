@@ -72,6 +72,8 @@ def build_train_parser() -> argparse.ArgumentParser:
     p.add_argument("--device", default="auto", help="auto|cpu|cuda")
     p.add_argument("--hip_graph", default="auto", choices=["auto", "off"],
                    help="capture the training step in a hipGraph (GPU full-sync)")
+    p.add_argument("--fused_step", default="auto", choices=["auto", "off"],
+                   help="hand-scheduled two-stream LeNet step inside the graph")
     p.add_argument("--backend", default="auto", help="auto|nccl|gloo")
     return p
 
