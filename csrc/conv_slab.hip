@@ -450,7 +450,7 @@ void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
       hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16>),
                          dim3((NB + 15) / 16), dim3(NTHREADS), 0, s, x, dact,
                          dw, NB);
-    } else if (nb_eff >= 1024) {
+    } else if (nb_eff >= 2048) {
       hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 4>),
                          dim3((NB + 3) / 4), dim3(NTHREADS), 0, s, x, dact,
                          dw, NB);

@@ -111,6 +111,10 @@ class LeNet5(nn.Module):
                                w_t=self._compT("fc2_w"))
         return logits
 
+    def predictions(self, logits):
+        """Softmax class probabilities (reference mnist.py:166-167)."""
+        return torch.softmax(logits.float(), dim=1)
+
     def loss_and_accuracy(self, logits, labels):
         """(mean CE loss, mean top-1 accuracy) — mnist.py:149-164."""
         loss, correct = Fx.softmax_xent(logits, labels)

@@ -68,6 +68,10 @@ class MLP(nn.Module):
                              db_out=self._gout("fc2_b"),
                              w_t=self._compT("fc2_w"))
 
+    def predictions(self, logits):
+        """Softmax class probabilities (reference mnist.py:166-167)."""
+        return torch.softmax(logits.float(), dim=1)
+
     def loss_and_accuracy(self, logits, labels):
         loss, correct = Fx.softmax_xent(logits, labels)
         return loss, correct / logits.shape[0]

@@ -95,3 +95,11 @@ def test_mlp_forward_backward():
 def test_build_model_names():
     assert isinstance(build_model("lenet"), LeNet5)
     assert isinstance(build_model("mlp"), MLP)
+
+
+def test_predictions_softmax():
+    m = MLP()
+    x = torch.rand(4, 28, 28, 1) - 0.5
+    p = m.predictions(m(x, train=False))
+    assert p.shape == (4, 10)
+    torch.testing.assert_close(p.sum(1), torch.ones(4), rtol=1e-5, atol=1e-5)
