@@ -350,20 +350,21 @@ void conv_dw_slab_kernel(const u16* __restrict__ x,
       bshift[ni] = (khkw / 5) * WP + (khkw % 5);
     }
 
+    // zero-fill ONCE per tile pass: the halo/pad zeros survive across
+    // images (staging overwrites exactly the valid pixel region each time)
+    __syncthreads();
+    for (int c = tid; c < COUT * (KPAD / 8); c += NTHREADS) {
+      *reinterpret_cast<short8*>(&dslab[c / (KPAD / 8)][(c % (KPAD / 8)) * 8]) =
+          short8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+    for (int c = tid; c < CIN * (XROW / 8); c += NTHREADS) {
+      *reinterpret_cast<short8*>(&xslab[c / (XROW / 8)][(c % (XROW / 8)) * 8]) =
+          short8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
     for (int g = 0; g < G; ++g) {
       int img = img0 + g;
       if (img >= NB) break;
-      // ---- stage slabs (zero-fill + transpose scatter) ----
-      __syncthreads();  // previous compute done before overwrite
-      for (int c = tid; c < COUT * (KPAD / 8); c += NTHREADS) {
-        *reinterpret_cast<short8*>(&dslab[c / (KPAD / 8)][(c % (KPAD / 8)) * 8]) =
-            short8{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-      for (int c = tid; c < CIN * (XROW / 8); c += NTHREADS) {
-        *reinterpret_cast<short8*>(&xslab[c / (XROW / 8)][(c % (XROW / 8)) * 8]) =
-            short8{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-      __syncthreads();
+      __syncthreads();  // previous compute / zero pass done before overwrite
       const u16* di = dact + (size_t)img * H * W * COUT;
       const u16* xi = x + (size_t)img * H * W * CIN;
       for (int c = tid; c < H * W * (COUT / 8); c += NTHREADS) {
@@ -491,16 +492,15 @@ void conv1_dw_slab_kernel(const u16* __restrict__ x,
 
   f32x4 acc = f32x4{0.f, 0.f, 0.f, 0.f};
 
+  for (int c = tid; c < COUT * (KPAD / 8); c += NTHREADS)
+    *reinterpret_cast<short8*>(&dslab[c / (KPAD / 8)][(c % (KPAD / 8)) * 8]) =
+        short8{0, 0, 0, 0, 0, 0, 0, 0};
+  for (int c = tid; c < XROW / 8; c += NTHREADS)
+    *reinterpret_cast<short8*>(&xslab[c * 8]) = short8{0, 0, 0, 0, 0, 0, 0, 0};
   for (int g = 0; g < G; ++g) {
     int img = img0 + g;
     if (img >= NB) break;
-    __syncthreads();
-    for (int c = tid; c < COUT * (KPAD / 8); c += NTHREADS)
-      *reinterpret_cast<short8*>(&dslab[c / (KPAD / 8)][(c % (KPAD / 8)) * 8]) =
-          short8{0, 0, 0, 0, 0, 0, 0, 0};
-    for (int c = tid; c < XROW / 8; c += NTHREADS)
-      *reinterpret_cast<short8*>(&xslab[c * 8]) = short8{0, 0, 0, 0, 0, 0, 0, 0};
-    __syncthreads();
+    __syncthreads();  // previous compute / zero pass done before overwrite
     const u16* di = dact + (size_t)img * H * W * COUT;
     const u16* xi = x + (size_t)img * H * W;
     for (int c = tid; c < H * W * (COUT / 8); c += NTHREADS) {
