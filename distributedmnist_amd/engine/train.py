@@ -211,11 +211,8 @@ class Trainer:
         begin_time = time.time()
         writer = None
         if flags.should_summarize and self.is_chief:
-            try:
-                from torch.utils.tensorboard import SummaryWriter
-                writer = SummaryWriter(flags.train_dir)
-            except Exception:
-                log.info("tensorboard unavailable; --should_summarize ignored")
+            from ..utils.tbwriter import make_writer
+            writer = make_writer(flags.train_dir)
         next_summary_time = time.time() + flags.save_summaries_secs
         while self.step < max_steps:
             start_time = time.time()

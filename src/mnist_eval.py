@@ -34,12 +34,8 @@ def main(argv=None):
     else:
         dataset = load_mnist(flags.data_dir, fake_data=flags.fake_data,
                              shard=False).validation
-    writer = None
-    try:
-        from torch.utils.tensorboard import SummaryWriter
-        writer = SummaryWriter(flags.eval_dir)
-    except Exception:
-        pass
+    from distributedmnist_amd.utils.tbwriter import make_writer
+    writer = make_writer(flags.eval_dir)
     evaluate(dataset, flags, writer=writer)
     return 0
 
