@@ -40,7 +40,11 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
   constexpr int WM = 112, WN = BN / 2;
   constexpr int MI = WM / 16, NI = WN / 16;
   constexpr int K = 25 * CIN;
-  __shared__ __align__(16) u16 slab[HP * WP * CIN];
+  // pixel stride padded 32->40 elems: an A-fragment's 16 lanes read 16
+  // DIFFERENT pixels at this stride; 64B stride = 16-way ds_read_b128 bank
+  // conflict, 80B = conflict-free ((a/4)%64 steps of 20)
+  constexpr int PST = CIN + 8;
+  __shared__ __align__(16) u16 slab[HP * WP * PST];
   __shared__ __align__(16) u16 Bs[BN][LDK];
 
   const int tid = threadIdx.x;
@@ -59,7 +63,7 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
     int sy = yy - 2, sx = xx - 2;
     if (sy >= 0 && sy < H && sx >= 0 && sx < W)
       v = *reinterpret_cast<const short8*>(xi + ((size_t)sy * W + sx) * CIN + ci);
-    *reinterpret_cast<short8*>(&slab[(yy * WP + xx) * CIN + ci]) = v;
+    *reinterpret_cast<short8*>(&slab[(yy * WP + xx) * PST + ci]) = v;
   }
 
   // per-lane A-fragment slab offsets for this wave's MI row-fragments:
@@ -73,7 +77,7 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
     if (lm >= M) { wo = 0; ho = 0; pos = 0; }
     int oy = ho * 2 + (pos >> 1) + 2;   // +2: padded coords
     int ox = wo * 2 + (pos & 1) + 2;
-    arow_off[mi] = (oy * WP + ox) * CIN;
+    arow_off[mi] = (oy * WP + ox) * PST;
   }
 
   f32x4 acc[MI][NI];
@@ -103,7 +107,7 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
       int kbase = kt + kh2 * 32 + kq;       // k = khkw*CIN + ci
       int khkw = kbase / CIN, ci = kbase % CIN;
       int kh = khkw / 5, kw = khkw % 5;
-      int shift = ((kh - 2) * WP + (kw - 2)) * CIN + ci;
+      int shift = ((kh - 2) * WP + (kw - 2)) * PST + ci;
       short8 af[MI], bf[NI];
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
@@ -167,7 +171,8 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
   constexpr int WM = 112, WN = BN / 2; // 16
   constexpr int MI = WM / 16, NI = WN / 16;  // 7, 1
   constexpr int K = 25 * COUT;
-  __shared__ __align__(16) u16 slab[HP * WP * COUT];
+  constexpr int PST = COUT + 8;  // 144B pixel stride: conflict-free b128
+  __shared__ __align__(16) u16 slab[HP * WP * PST];
   __shared__ __align__(16) u16 Bs[BN][LDK];
 
   const int tid = threadIdx.x;
@@ -185,7 +190,7 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
     int sy = yy - 2, sx = xx - 2;
     if (sy >= 0 && sy < H && sx >= 0 && sx < W)
       v = *reinterpret_cast<const short8*>(di + ((size_t)sy * W + sx) * COUT + co);
-    *reinterpret_cast<short8*>(&slab[(yy * WP + xx) * COUT + co]) = v;
+    *reinterpret_cast<short8*>(&slab[(yy * WP + xx) * PST + co]) = v;
   }
 
   int arow_off[MI];
@@ -194,7 +199,7 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
     int lm = wr * WM + mi * 16 + (lane & 15);
     int wi = lm % W, hi = (lm / W) % H;
     if (lm >= M) { wi = 0; hi = 0; }
-    arow_off[mi] = ((hi + 2) * WP + (wi + 2)) * COUT;  // padded (hi,wi)
+    arow_off[mi] = ((hi + 2) * WP + (wi + 2)) * PST;  // padded (hi,wi)
   }
 
   f32x4 acc[MI][NI];
@@ -224,7 +229,7 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
       int khkw = kbase / COUT, co = kbase % COUT;
       int kh = khkw / 5, kw = khkw % 5;
       // dact[hi - kh + 2, wi - kw + 2] -> padded offset shift
-      int shift = (-(kh - 2) * WP - (kw - 2)) * COUT + co;
+      int shift = (-(kh - 2) * WP - (kw - 2)) * PST + co;
       short8 af[MI], bf[NI];
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
@@ -313,12 +318,15 @@ void conv_dw_slab_kernel(const u16* __restrict__ x,
                          float* __restrict__ dw, int NB) {
   constexpr int HP = H + 4, WP = W + 4;           // 18 x 18
   constexpr int KPAD = 384;                       // 324 padded to 6 BK-steps
+  constexpr int DST = KPAD + 8;                   // row stride: the 16 lanes
+  // of an A-fragment read 16 co ROWS; a 768B stride is a 16-way b128 bank
+  // conflict, 784B is conflict-free
   constexpr int XROW = 472;                       // max addr 466 + margin
   constexpr int NP = 25 * CIN;                    // 800
   constexpr int BN = 448;                         // 2 tiles
   constexpr int WN = 112, NI = WN / 16;           // 7
   constexpr int MI = COUT / 16;                   // 4 (WM = COUT = 64)
-  __shared__ __align__(16) u16 dslab[COUT][KPAD];
+  __shared__ __align__(16) u16 dslab[COUT][DST];
   __shared__ __align__(16) u16 xslab[CIN][XROW];
 
   const int tid = threadIdx.x;
@@ -353,8 +361,8 @@ void conv_dw_slab_kernel(const u16* __restrict__ x,
     // zero-fill ONCE per tile pass: the halo/pad zeros survive across
     // images (staging overwrites exactly the valid pixel region each time)
     __syncthreads();
-    for (int c = tid; c < COUT * (KPAD / 8); c += NTHREADS) {
-      *reinterpret_cast<short8*>(&dslab[c / (KPAD / 8)][(c % (KPAD / 8)) * 8]) =
+    for (int c = tid; c < COUT * (DST / 8); c += NTHREADS) {
+      *reinterpret_cast<short8*>(&dslab[c / (DST / 8)][(c % (DST / 8)) * 8]) =
           short8{0, 0, 0, 0, 0, 0, 0, 0};
     }
     for (int c = tid; c < CIN * (XROW / 8); c += NTHREADS) {
@@ -474,8 +482,9 @@ void conv1_dw_slab_kernel(const u16* __restrict__ x,
                           float* __restrict__ dw, int NB) {
   constexpr int HP = H + 4, WP = W + 4;            // 32 x 32
   constexpr int KPAD = HP * WP;                    // 1024 = 16 BK-steps
+  constexpr int DST = KPAD + 8;  // 2064B row stride: conflict-free b128
   constexpr int XROW = 1168;  // max addr (4*32+4) + 1023 + 8 margin
-  __shared__ __align__(16) u16 dslab[COUT][KPAD];
+  __shared__ __align__(16) u16 dslab[COUT][DST];
   __shared__ __align__(16) u16 xslab[XROW];
 
   const int tid = threadIdx.x;
@@ -492,8 +501,8 @@ void conv1_dw_slab_kernel(const u16* __restrict__ x,
 
   f32x4 acc = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (int c = tid; c < COUT * (KPAD / 8); c += NTHREADS)
-    *reinterpret_cast<short8*>(&dslab[c / (KPAD / 8)][(c % (KPAD / 8)) * 8]) =
+  for (int c = tid; c < COUT * (DST / 8); c += NTHREADS)
+    *reinterpret_cast<short8*>(&dslab[c / (DST / 8)][(c % (DST / 8)) * 8]) =
         short8{0, 0, 0, 0, 0, 0, 0, 0};
   for (int c = tid; c < XROW / 8; c += NTHREADS)
     *reinterpret_cast<short8*>(&xslab[c * 8]) = short8{0, 0, 0, 0, 0, 0, 0, 0};

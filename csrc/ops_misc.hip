@@ -333,31 +333,29 @@ extern "C" __global__ __launch_bounds__(256)
 void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
                              const float* bias, ushort_t* y, uint8_t* amax,
                              int NB, int H, int W, int Cout, int Mpool) {
-  // 64 pooled pixels per block (8 per thread-slot); the 25 filter taps for
-  // this thread's Cout live in REGISTERS, windows in LDS (broadcast reads
-  // across the 32 co lanes); 100 fp32 FMA per pixel.
-  __shared__ float win[64][6][6];
+  // One IMAGE per block (H=W=28): x staged once as a fp32 [32][32] padded
+  // slab (4 KB); 256 threads = 32 co x 8 slots, each slot walking a
+  // CONTIGUOUS run of pooled pixels so the 6x6 window shifts by 2 columns
+  // per step — 12 incremental LDS reads (broadcast across the 32 co lanes)
+  // instead of 36.  The 25 filter taps live in registers (co is fixed per
+  // thread).  K=25 with Cin=1 is too small for MFMA to win; this VALU form
+  // runs at ~100 fp32 FMA per output.
+  __shared__ float slab[32][32];
   __shared__ float wl[25][32];
   const int Ho = H / 2, Wo = W / 2;
   const int tid = threadIdx.x;
-  const long q0 = (long)blockIdx.x * 64;
+  const int img = blockIdx.x;
+  const int qpi = Ho * Wo;              // pooled pixels per image (196)
   for (int i = tid; i < 25 * Cout; i += 256)
     wl[i / Cout][i % Cout] = bf2f(w[i]);
-  for (int i = tid; i < 64 * 36; i += 256) {
-    int p = i / 36, e = i % 36;
-    int wr = e / 6, wc = e % 6;
-    long q = q0 + p;
+  const ushort_t* xi = x + (size_t)img * H * W;
+  for (int i = tid; i < 32 * 32; i += 256) {
+    int xx = i % 32, yy = i / 32;
+    int sy = yy - 2, sx = xx - 2;
     float v = 0.f;
-    if (q < Mpool) {
-      int wo = q % Wo;
-      long t = q / Wo;
-      int ho = t % Ho;
-      int n = t / Ho;
-      int yy = ho * 2 - 2 + wr, xx = wo * 2 - 2 + wc;
-      if (yy >= 0 && yy < H && xx >= 0 && xx < W)
-        v = bf2f(x[((size_t)n * H + yy) * W + xx]);
-    }
-    win[p][wr][wc] = v;
+    if (sy >= 0 && sy < H && sx >= 0 && sx < W)
+      v = bf2f(xi[(size_t)sy * W + sx]);
+    slab[yy][xx] = v;
   }
   __syncthreads();
   const int co = tid & 31, slot = tid >> 5;
@@ -366,24 +364,43 @@ void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
 #pragma unroll
   for (int t = 0; t < 25; ++t) wreg[t] = wl[t][co];
   const float b = bias[co];
+  const int run = (qpi + 7) / 8;        // 25 pooled pixels per slot
+  int q0 = slot * run;
+  int q1 = min(qpi, q0 + run);
+  ushort_t* yi = y + (size_t)img * qpi * Cout;
+  uint8_t* ai = amax + (size_t)img * qpi * Cout;
+  float wv[6][6];
+  int prev_ho = -9;
+  for (int q = q0; q < q1; ++q) {
+    int wo = q % Wo, ho = q / Wo;
+    int oy = ho * 2, ox = wo * 2;       // slab coords = conv coords + 2 pad
+    if (ho != prev_ho || wo == 0) {
+      // full window load (row change breaks column adjacency)
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    int p = slot * 8 + j;
-    long q = q0 + p;
-    if (q >= Mpool) break;
-    float wv[36];
+      for (int r = 0; r < 6; ++r)
 #pragma unroll
-    for (int e = 0; e < 36; ++e) wv[e] = win[p][e / 6][e % 6];
+        for (int c = 0; c < 6; ++c) wv[r][c] = slab[oy + r][ox + c];
+      prev_ho = ho;
+    } else {
+      // shift left by 2 columns, read the 2 new columns
+#pragma unroll
+      for (int r = 0; r < 6; ++r) {
+#pragma unroll
+        for (int c = 0; c < 4; ++c) wv[r][c] = wv[r][c + 2];
+        wv[r][4] = slab[oy + r][ox + 4];
+        wv[r][5] = slab[oy + r][ox + 5];
+      }
+    }
     float acc0 = b, acc1 = b, acc2 = b, acc3 = b;
 #pragma unroll
     for (int kh = 0; kh < 5; ++kh)
 #pragma unroll
       for (int kw = 0; kw < 5; ++kw) {
         float wc_ = wreg[kh * 5 + kw];
-        acc0 += wv[kh * 6 + kw] * wc_;
-        acc1 += wv[kh * 6 + kw + 1] * wc_;
-        acc2 += wv[(kh + 1) * 6 + kw] * wc_;
-        acc3 += wv[(kh + 1) * 6 + kw + 1] * wc_;
+        acc0 += wv[kh][kw] * wc_;
+        acc1 += wv[kh][kw + 1] * wc_;
+        acc2 += wv[kh + 1][kw] * wc_;
+        acc3 += wv[kh + 1][kw + 1] * wc_;
       }
     float vals[4] = {acc0, acc1, acc2, acc3};
     float best = -1.0f / 0.0f;
@@ -393,8 +410,8 @@ void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
       float v = vals[pz] > 0.f ? vals[pz] : 0.f;
       if (v > best) { best = v; barg = pz; }
     }
-    y[(size_t)q * Cout + co] = f2bf(best);
-    amax[(size_t)q * Cout + co] = (uint8_t)barg;
+    yi[(size_t)q * Cout + co] = f2bf(best);
+    ai[(size_t)q * Cout + co] = (uint8_t)barg;
   }
 }
 
@@ -410,7 +427,6 @@ void launch_conv1_direct_fwd(const unsigned short* x, const unsigned short* w,
                              uint8_t* amax, int NB, int H, int W, int Cout,
                              hipStream_t s) {
   long mpool = (long)NB * (H / 2) * (W / 2);
-  dim3 grid((mpool + 63) / 64);
-  hipLaunchKernelGGL(conv1_direct_fwd_kernel, grid, dim3(256), 0, s, x, w,
-                     bias, y, amax, NB, H, W, Cout, (int)mpool);
+  hipLaunchKernelGGL(conv1_direct_fwd_kernel, dim3(NB), dim3(256), 0, s, x,
+                     w, bias, y, amax, NB, H, W, Cout, (int)mpool);
 }
