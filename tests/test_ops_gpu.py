@@ -355,3 +355,35 @@ def test_conv_dw_slab_image_groups(ext, force_g, monkeypatch):
         assert_close_bf16(dx, dx_ref, scale=float(dx_ref.abs().max()))
     finally:
         del os.environ["DMNIST_DW_G"]
+
+
+def test_train_loop_and_eval_gpu(ext, tmp_path):
+    """Full engine on GPU: train() with checkpointing, then the evaluator
+    entry point consumes the checkpoint (BASELINE configs 2 plumbing)."""
+    import os
+    import re
+    import subprocess
+    import sys
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+    td = str(tmp_path / "train")
+    flags = build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", td, "--batch_size", "256",
+         "--max_steps", "10", "--model", "lenet",
+         "--save_interval_secs", "0", "--save_results_period", "5"])
+    t = Trainer(flags, device=torch.device("cuda:0"))
+    hist = t.train(make_dataset(flags, 0, 1, t.device, t.compute_dtype))
+    assert len(hist) == 10
+    assert os.path.exists(os.path.join(td, "checkpoint"))
+    assert os.path.exists(os.path.join(td, "worker0_time_acc.npy"))
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "src", "mnist_eval.py"),
+         "--checkpoint_dir", td, "--eval_dir", str(tmp_path / "eval"),
+         "--run_once", "--synthetic_data", "--model", "lenet"],
+        capture_output=True, text=True, timeout=300, cwd=root)
+    assert out.returncode == 0, out.stderr[-2000:]
+    m = re.search(r"Num examples: (\d+)  Precision @ 1: ([\d.]+) Loss: ([\d.]+)",
+                  out.stdout)
+    assert m, out.stdout
+    assert (tmp_path / "eval").exists()
