@@ -43,7 +43,7 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
   // pixel stride padded 32->40 elems: an A-fragment's 16 lanes read 16
   // DIFFERENT pixels at this stride; 64B stride = 16-way ds_read_b128 bank
   // conflict, 80B = conflict-free ((a/4)%64 steps of 20)
-  constexpr int PST = CIN + 4;  // 72B pixel stride measured best here
+  constexpr int PST = CIN + 8;  // 80B pixel stride: 16B-aligned b128, conflict-reduced
   __shared__ __align__(16) u16 slab[HP * WP * PST];
   __shared__ __align__(16) u16 Bs[BN][LDK];
 
