@@ -22,6 +22,27 @@
 
 typedef ushort_t u16;
 
+// read elements [s, s+8) of an LDS row via two ALIGNED ds_read_b128 + a
+// constant shuffle: s is wave-uniform per call site (the 8-case switch is a
+// uniform scalar branch), so this replaces 8 scalar ds_read_u16 per
+// fragment (the conv-dW x operand is pixel-shifted by the filter tap and
+// never 16B-aligned).
+DEV short8 lds_row_shifted(const u16* row, int s) {
+  int s0 = s & ~7;
+  short8 lo = *reinterpret_cast<const short8*>(row + s0);
+  short8 hi = *reinterpret_cast<const short8*>(row + s0 + 8);
+  switch (s & 7) {
+    case 0: return lo;
+    case 1: return __builtin_shufflevector(lo, hi, 1, 2, 3, 4, 5, 6, 7, 8);
+    case 2: return __builtin_shufflevector(lo, hi, 2, 3, 4, 5, 6, 7, 8, 9);
+    case 3: return __builtin_shufflevector(lo, hi, 3, 4, 5, 6, 7, 8, 9, 10);
+    case 4: return __builtin_shufflevector(lo, hi, 4, 5, 6, 7, 8, 9, 10, 11);
+    case 5: return __builtin_shufflevector(lo, hi, 5, 6, 7, 8, 9, 10, 11, 12);
+    case 6: return __builtin_shufflevector(lo, hi, 6, 7, 8, 9, 10, 11, 12, 13);
+    default: return __builtin_shufflevector(lo, hi, 7, 8, 9, 10, 11, 12, 13, 14);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // conv2 fwd: y = maxpool(relu(conv5x5(x) + b)), one image per block.
 // x slab: [H+4][W+4][Cin] bf16, zero halo. K = 25*Cin, BK = 64.
@@ -404,15 +425,13 @@ void conv_dw_slab_kernel(const u16* __restrict__ x,
           for (int mi = 0; mi < MI; ++mi)
             af[mi] = *reinterpret_cast<const short8*>(
                 &dslab[mi * 16 + lcol][k0]);
-          // B fragments just-in-time (one live at a time — the 8 unaligned
-          // u16 reads per fragment otherwise blow register liveness to
-          // 1 wave/SIMD)
+          // B fragments just-in-time; bshift is UNIFORM per fragment
+          // (a 16-column block never straddles a khkw boundary), so the
+          // shifted read is two aligned b128 + a uniform constant shuffle
 #pragma unroll
           for (int ni = 0; ni < NI; ++ni) {
-            const u16* base = &xslab[bci[ni]][bshift[ni] + k0];
-            short8 bf;
-#pragma unroll
-            for (int e = 0; e < 8; ++e) bf[e] = (short)base[e];
+            short8 bf = lds_row_shifted(&xslab[bci[ni]][0],
+                                        bshift[ni] + k0);
 #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
               acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
