@@ -33,20 +33,33 @@ TEST_IMAGES = "t10k-images-idx3-ubyte.gz"
 TEST_LABELS = "t10k-labels-idx1-ubyte.gz"
 
 
-def extract_images(path: str, num_images: int) -> np.ndarray:
-    """idx3-ubyte.gz -> [N,28,28,1] float32 normalized to [-0.5, 0.5]."""
+def extract_images(path: str, num_images: int | None = None) -> np.ndarray:
+    """idx3-ubyte.gz -> [N,28,28,1] float32 normalized to [-0.5, 0.5].
+
+    The image count comes from the idx header (the reference hardcoded
+    60000/10000; reading the header makes subsets/test fixtures loadable)."""
+    import struct
     with gzip.open(path) as f:
-        f.read(16)
-        buf = f.read(IMAGE_SIZE * IMAGE_SIZE * num_images * NUM_CHANNELS)
+        magic, n, rows, cols = struct.unpack(">IIII", f.read(16))
+        if magic != 2051:
+            raise ValueError(f"{path}: bad idx3 magic {magic}")
+        if num_images is not None:
+            n = min(n, num_images)
+        buf = f.read(rows * cols * n * NUM_CHANNELS)
         data = np.frombuffer(buf, dtype=np.uint8).astype(np.float32)
         data = (data - PIXEL_DEPTH / 2.0) / PIXEL_DEPTH
-        return data.reshape(num_images, IMAGE_SIZE, IMAGE_SIZE, NUM_CHANNELS)
+        return data.reshape(n, rows, cols, NUM_CHANNELS)
 
 
-def extract_labels(path: str, num_images: int) -> np.ndarray:
+def extract_labels(path: str, num_images: int | None = None) -> np.ndarray:
+    import struct
     with gzip.open(path) as f:
-        f.read(8)
-        buf = f.read(num_images)
+        magic, n = struct.unpack(">II", f.read(8))
+        if magic != 2049:
+            raise ValueError(f"{path}: bad idx1 magic {magic}")
+        if num_images is not None:
+            n = min(n, num_images)
+        buf = f.read(n)
         return np.frombuffer(buf, dtype=np.uint8).astype(np.int64)
 
 
@@ -169,10 +182,10 @@ def read_data_sets(train_dir: str, fake_data: bool = False,
             f"MNIST files missing under {train_dir}: {missing}. This "
             "environment has no network; place the idx-gz files there or use "
             "--subset/fake data / synthetic mode.")
-    train_images = extract_images(paths["ti"], 60000)
-    train_labels = extract_labels(paths["tl"], 60000)
-    test_images = extract_images(paths["ei"], 10000)
-    test_labels = extract_labels(paths["el"], 10000)
+    train_images = extract_images(paths["ti"])
+    train_labels = extract_labels(paths["tl"])
+    test_images = extract_images(paths["ei"])
+    test_labels = extract_labels(paths["el"])
     # reference quirk kept: validation IS the test set (mnist_data.py:200-201)
     return Datasets(
         DataSet(train_images, train_labels, worker_id=worker_id,

@@ -93,3 +93,23 @@ def test_deterministic_training_cpu(tmp_path):
     l2, w2 = run("b")
     assert l1 == l2
     assert torch.equal(w1, w2)
+
+
+def test_real_data_training_path(tmp_path):
+    """End-to-end: idx-gz files on disk -> load_mnist -> Trainer (the
+    non-synthetic data path the reference used, minus the download)."""
+    write_idx(tmp_path, n_train=128, n_test=64)
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+    flags = build_train_parser().parse_args(
+        ["--data_dir", str(tmp_path), "--train_dir", str(tmp_path / "td"),
+         "--batch_size", "16", "--max_steps", "4", "--model", "lenet",
+         "--device", "cpu", "--save_interval_secs", "100000"])
+    t = Trainer(flags)
+    ds = make_dataset(flags, 0, 1, t.device, t.compute_dtype)
+    assert ds.num_examples == 128
+    hist = t.train(ds)
+    assert len(hist) == 4
+    # sharded 2-worker view of the same files
+    ds2 = make_dataset(flags, 1, 2, t.device, t.compute_dtype)
+    assert ds2.num_examples == 64
