@@ -238,7 +238,8 @@ DEV void writeB(ushort_t (*Bs)[LDK], int tid, const short8 (&regs)[CH]) {
 //   huge-grid conv gathers where 4-5 blocks/CU of TLP hide latency better
 //   than the in-wave pipeline (measured: conv2 fwd 103us PIPE0 vs 120us
 //   PIPE1; fc1 fwd 94us PIPE0 vs 59us PIPE1).
-template <int BM, int BN, int AMODE, int BMODE, int EPI, int OUT, int PIPE>
+template <int BM, int BN, int AMODE, int BMODE, int EPI, int OUT, int PIPE,
+          int SWZ = 0>
 __global__ __launch_bounds__(NTHREADS)
 void gemm_tile_kernel(GemmParams p) {
   constexpr int WM = BM / 2, WN = BN / 2;
@@ -254,12 +255,30 @@ void gemm_tile_kernel(GemmParams p) {
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int wr = wave >> 1, wc = wave & 1;
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  if (SWZ) {
+    // bijective XCD-chunk remap (cdna_hip_programming.md T1): blocks with
+    // the same split-K slice / N-tile land on ONE XCD so the operand slice
+    // they share stays in that XCD's L2 instead of being re-fetched via L3.
+    // Default dispatch puts block b on XCD b%8; give each XCD a CONTIGUOUS
+    // chunk of the (x-fastest) flattened grid instead.
+    int gx = gridDim.x, gz = gridDim.z;
+    int total = gx * gridDim.y * gz;
+    int flat = bx + gx * (by + gridDim.y * bz);
+    int q = total / 8, r = total % 8;
+    int xcd = flat % 8, pos = flat / 8;
+    int flat2 = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    bx = flat2 % gx;
+    int rest = flat2 / gx;
+    by = rest % gridDim.y;
+    bz = rest / gridDim.y;
+  }
+  const int m0 = bx * BM;
+  const int n0 = by * BN;
 
   int ksteps_total = (p.K + BK - 1) / BK;
   int steps_per = (ksteps_total + p.splitk - 1) / p.splitk;
-  int kbeg = (int)blockIdx.z * steps_per * BK;
+  int kbeg = bz * steps_per * BK;
   int kend = min(p.K, kbeg + steps_per * BK);
   int nt = (kend - kbeg + BK - 1) / BK;
   if (nt <= 0) nt = 0;
@@ -392,6 +411,13 @@ static inline int cdiv_host(int a, int b) { return (a + b - 1) / b; }
     hipLaunchKernelGGL((gemm_tile_kernel<BM, BN, AM, BMo, EPI, OUT, PIPE>),  \
                        grid, dim3(NTHREADS), 0, s, p);                       \
   }
+#define GEMM_ENTRY_SWZ(name, BM, BN, AM, BMo, EPI, OUT, PIPE)                \
+  void name(const GemmParams& p, hipStream_t s) {                            \
+    dim3 grid(cdiv_host(p.M, BM), cdiv_host(p.N, BN), p.splitk);             \
+    hipLaunchKernelGGL((gemm_tile_kernel<BM, BN, AM, BMo, EPI, OUT, PIPE,    \
+                                         1>),                                \
+                       grid, dim3(NTHREADS), 0, s, p);                       \
+  }
 
 // fc GEMMs: grids of 32-400 WGs (occupancy grid-limited) -> PIPE=1
 GEMM_ENTRY(gemm_fwd_bias_128, 128, 128, A_N, B_KMAJ, EPI_BIAS, OUT_BF16, 1)
@@ -402,8 +428,8 @@ GEMM_ENTRY(gemm_fwd_drop_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF1
 GEMM_ENTRY(gemm_fwd_drop_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dx_128, 128, 128, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dx_64, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
-GEMM_ENTRY(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
-GEMM_ENTRY(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+GEMM_ENTRY_SWZ(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+GEMM_ENTRY_SWZ(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
 // B-transposed (pre-transposed weight) fwd variants: ldb = K, vector staging
 GEMM_ENTRY(gemm_fwd_bias_128_bt, 128, 128, A_N, B_NMAJ, EPI_BIAS, OUT_BF16, 1)
 GEMM_ENTRY(gemm_fwd_bias_64_bt, 64, 64, A_N, B_NMAJ, EPI_BIAS, OUT_BF16, 1)
@@ -419,6 +445,6 @@ GEMM_ENTRY(conv_dx_gemm, 128, 32, A_CONV_DX, B_CONV_DX_W, EPI_NONE, OUT_BF16, 0)
 // conv dW: gather-staged -> occupancy (PIPE=0) beats the in-wave pipeline
 // (measured 2069us PIPE=1 vs 1309 at BM=64 PIPE=0); BM=128 halves the
 // per-M-tile dact re-reads
-GEMM_ENTRY(conv_dw_gemm, 128, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)
+GEMM_ENTRY_SWZ(conv_dw_gemm, 128, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)
 // conv1 dW: M=25 -> 32x32 tile (78% M-utilization vs 39% at BM=64)
-GEMM_ENTRY(conv1_dw_gemm, 32, 32, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)
+GEMM_ENTRY_SWZ(conv1_dw_gemm, 32, 32, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)
