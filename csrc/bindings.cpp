@@ -468,6 +468,20 @@ void sgd_step_dev(torch::Tensor master, torch::Tensor grad,
                       cur_stream());
 }
 
+void transpose_bf16_batch(std::vector<torch::Tensor> srcs,
+                          std::vector<torch::Tensor> dsts) {
+  TORCH_CHECK(srcs.size() == dsts.size() && srcs.size() <= 4,
+              "up to 4 transpose pairs");
+  TransposeDesc d[4];
+  for (size_t i = 0; i < srcs.size(); ++i) {
+    CHECK_BF16(srcs[i]); CHECK_CONTIG(srcs[i]);
+    CHECK_BF16(dsts[i]); CHECK_CONTIG(dsts[i]);
+    d[i] = {bf16_ptr(srcs[i]), bf16_mut(dsts[i]),
+            (int)srcs[i].size(0), (int)srcs[i].size(1)};
+  }
+  launch_transpose_bf16_batch(d, (int)srcs.size(), cur_stream());
+}
+
 void transpose_bf16(torch::Tensor src, torch::Tensor dst) {
   CHECK_CUDA(src); CHECK_BF16(src); CHECK_CONTIG(src);
   CHECK_BF16(dst); CHECK_CONTIG(dst);
@@ -494,6 +508,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("p_keep"), py::arg("seed"), py::arg("offset"),
         py::arg("wT") = c10::nullopt);
   m.def("transpose_bf16", &transpose_bf16, "bf16 2-D transpose (wT refresh)");
+  m.def("transpose_bf16_batch", &transpose_bf16_batch,
+        "up to 4 transposes in one launch");
   m.def("mask_db", &mask_db, "relu/dropout grad mask + bias-grad column sums");
   m.def("linear_dw_into", &linear_dw_into, "dW = x^T dyeff into bucket view");
   m.def("linear_dx", &linear_dx, "dx = dyeff @ W^T");

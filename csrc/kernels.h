@@ -59,6 +59,18 @@ void launch_step_advance(long* step_dev, float* lr_scale_dev, float lr0,
                          hipStream_t);
 void launch_transpose_bf16(const unsigned short* src, unsigned short* dst,
                            int R, int C, hipStream_t);
+struct TransposeDesc {
+  const unsigned short* src;
+  unsigned short* dst;
+  int R, C;
+};
+struct TransposeBatchArgs {
+  TransposeDesc d[4];
+  int tile0[4];
+  int n, total;
+};
+void launch_transpose_bf16_batch(const TransposeDesc* descs, int n,
+                                 hipStream_t);
 void launch_conv1_direct_fwd(const unsigned short* x, const unsigned short* w,
                              const float* bias, unsigned short* y,
                              uint8_t* amax, int NB, int H, int W, int Cout,

@@ -4,6 +4,7 @@
 // + bf16 shadow refresh).  All memory-bound: bf16 payloads, grid-stride.
 
 #include "common.h"
+#include "kernels.h"
 
 // ---------------------------------------------------------------------------
 // dyeff[b,n] = dy * (y>0) / p_keep ; db[n] += column sums (fp32)
@@ -226,8 +227,35 @@ extern "C" __global__ void step_advance_kernel(long* step_dev,
   }
 }
 
+// batched variant: all per-step weight re-transposes in ONE launch (three
+// separate ~5us launches showed up as ~20us of the B=1024 graph replay)
+extern "C" __global__ __launch_bounds__(256)
+void transpose_bf16_batch_kernel(TransposeBatchArgs a) {
+  __shared__ ushort_t tile[32][33];
+  int t = blockIdx.x;
+  int which = 0;
+  while (which + 1 < a.n && t >= a.tile0[which + 1]) ++which;
+  int local = t - a.tile0[which];
+  const TransposeDesc d = a.d[which];
+  int ctiles = (d.C + 31) / 32;
+  int tr0 = (local / ctiles) * 32;
+  int tc0 = (local % ctiles) * 32;
+  int lx = threadIdx.x & 31, ly = threadIdx.x >> 5;
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+    int r = tr0 + ly + rr * 8, c = tc0 + lx;
+    tile[ly + rr * 8][lx] = (r < d.R && c < d.C)
+        ? d.src[(size_t)r * d.C + c] : (ushort_t)0;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+    int c = tc0 + ly + rr * 8, r = tr0 + lx;
+    if (c < d.C && r < d.R) d.dst[(size_t)c * d.R + r] = tile[lx][ly + rr * 8];
+  }
+}
+
 // ---- host wrappers --------------------------------------------------------
-#include "kernels.h"
 
 static inline int cdivh(long a, long b) { return (int)((a + b - 1) / b); }
 
@@ -420,6 +448,21 @@ void launch_transpose_bf16(const unsigned short* src, unsigned short* dst,
   dim3 grid((C + 31) / 32, (R + 31) / 32);
   hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, s, src, dst,
                      R, C);
+}
+
+void launch_transpose_bf16_batch(const TransposeDesc* descs, int n,
+                                 hipStream_t s) {
+  TransposeBatchArgs a{};
+  int total = 0;
+  for (int i = 0; i < n && i < 4; ++i) {
+    a.d[i] = descs[i];
+    a.tile0[i] = total;
+    total += ((descs[i].C + 31) / 32) * ((descs[i].R + 31) / 32);
+  }
+  a.n = n < 4 ? n : 4;
+  a.total = total;
+  hipLaunchKernelGGL(transpose_bf16_batch_kernel, dim3(total), dim3(256), 0,
+                     s, a);
 }
 
 void launch_conv1_direct_fwd(const unsigned short* x, const unsigned short* w,

@@ -102,8 +102,8 @@ class FlatParams:
         if self.flat_shadow.is_cuda:
             from .. import _C
             ext = _C.ext()
-            for v2, t in self._t_pairs:
-                ext.transpose_bf16(v2, t)
+            ext.transpose_bf16_batch([v2 for v2, _ in self._t_pairs],
+                                     [t for _, t in self._t_pairs])
         else:
             for v2, t in self._t_pairs:
                 t.copy_(v2.t())
