@@ -332,7 +332,9 @@ void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
 // Block: G=16 images accumulated into the same registers (flush atomics
 // /16); 2 N'-tiles of 448 (waves 1x4, wave tile 64x112, acc 4x7).
 // ---------------------------------------------------------------------------
-template <int H, int W, int CIN, int COUT, int G>
+// ABL: perf-ablation variants (0=full, 1=skip MFMA loop, 2=skip staging,
+// 3=skip flush) — selected by DMNIST_DW_ABL, numerically wrong except 0.
+template <int H, int W, int CIN, int COUT, int G, int ABL = 0>
 __global__ __launch_bounds__(NTHREADS, 2)  // force <=256 regs: 2 blocks/CU
 void conv_dw_slab_kernel(const u16* __restrict__ x,
                          const u16* __restrict__ dact,
@@ -396,6 +398,7 @@ void conv_dw_slab_kernel(const u16* __restrict__ x,
       __syncthreads();  // previous compute / zero pass done before overwrite
       const u16* di = dact + (size_t)img * H * W * COUT;
       const u16* xi = x + (size_t)img * H * W * CIN;
+      if (ABL == 2) goto compute;  // slab holds zeros/stale: staging ablated
       for (int c = tid; c < H * W * (COUT / 8); c += NTHREADS) {
         int co0 = (c % (COUT / 8)) * 8;
         int pix = c / (COUT / 8);
@@ -414,7 +417,9 @@ void conv_dw_slab_kernel(const u16* __restrict__ x,
 #pragma unroll
         for (int e = 0; e < 8; ++e) xslab[ci0 + e][pk] = v[e];
       }
+compute:
       __syncthreads();
+      if (ABL == 1) continue;
       // ---- 6 K-steps over the padded pixel space ----
       for (int kt = 0; kt < KPAD; kt += BK) {
 #pragma unroll
@@ -441,6 +446,7 @@ void conv_dw_slab_kernel(const u16* __restrict__ x,
       }
     }
     // ---- flush: dW'[co][n'] -> atomicAdd dW[(khkw*CIN+ci)*COUT + co] ----
+    if (ABL == 3) { __syncthreads(); continue; }
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi) {
 #pragma unroll
