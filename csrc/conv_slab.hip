@@ -480,10 +480,18 @@ void launch_conv_dw_slab(const unsigned short* x, const unsigned short* dact,
     }();
     int nb_eff = force_g ? (force_g >= 16 ? 8192 : (force_g >= 4 ? 2048 : 0))
                          : NB;
+    static int abl = [] {
+      const char* e = getenv("DMNIST_DW_ABL");
+      return e ? atoi(e) : 0;
+    }();
     if (nb_eff >= 8192) {
-      hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16>),
-                         dim3((NB + 15) / 16), dim3(NTHREADS), 0, s, x, dact,
-                         dw, NB);
+      dim3 grid((NB + 15) / 16);
+      switch (abl) {
+        case 1: hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16, 1>), grid, dim3(NTHREADS), 0, s, x, dact, dw, NB); break;
+        case 2: hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16, 2>), grid, dim3(NTHREADS), 0, s, x, dact, dw, NB); break;
+        case 3: hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16, 3>), grid, dim3(NTHREADS), 0, s, x, dact, dw, NB); break;
+        default: hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 16, 0>), grid, dim3(NTHREADS), 0, s, x, dact, dw, NB); break;
+      }
     } else if (nb_eff >= 2048) {
       hipLaunchKernelGGL((conv_dw_slab_kernel<14, 14, 32, 64, 4>),
                          dim3((NB + 3) / 4), dim3(NTHREADS), 0, s, x, dact,
