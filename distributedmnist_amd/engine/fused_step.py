@@ -19,6 +19,7 @@ All dW/db land directly in the flat fp32 all-reduce bucket (pre-zeroed).
 from __future__ import annotations
 
 import torch
+import torch.distributed as dist
 
 from .. import _C
 
@@ -36,6 +37,15 @@ class FusedLeNetStep:
         self.side = torch.cuda.Stream()
         self.p_keep = 0.5
         self.seed = t.flags.seed
+        # bucketed all-reduce: the fc gradients (96.5% of the payload,
+        # SURVEY.md §2.4) are complete long before the conv backward — their
+        # collective overlaps it on the side stream.  The flat bucket layout
+        # (conv*, then fc*) makes the split two contiguous slices.
+        fp = t.fp
+        fc0 = fp.offsets[fp.names.index("fc1_w")]
+        self.conv_slice = fp.flat_grad[:fc0]
+        self.fc_slice = fp.flat_grad[fc0:]
+        self.overlap_allreduce = t.world > 1 and dist.is_initialized()
 
     def __call__(self, x, labels, step_dev):
         """One fwd+bwd; gradients land in the flat bucket.
@@ -74,6 +84,10 @@ class FusedLeNetStep:
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
+            if self.overlap_allreduce:
+                # every fc gradient (fc2 via the earlier s1 work + the db's
+                # ordered by wait_stream) is final here
+                dist.all_reduce(self.fc_slice, op=dist.ReduceOp.SUM)
         dx1 = ext.linear_dx(dyeff1, sh["fc1_w"]).view(B, 7, 7, 64)
 
         dact2 = ext.pool_scatter(dx1, y2, am2, gv("conv2_b"), 14, 14)
@@ -86,6 +100,8 @@ class FusedLeNetStep:
         ext.conv_dw_into(x, dact1, gv("conv1_w"))
 
         s0.wait_stream(s1)
+        if self.overlap_allreduce:
+            dist.all_reduce(self.conv_slice, op=dist.ReduceOp.SUM)
         # keep the side-stream consumers alive until the join (capture-safe)
         self._keep = (a1, h2, dyeff1, dyeff2, y1, dact2)
         return loss, correct

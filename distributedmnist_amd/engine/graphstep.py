@@ -93,15 +93,17 @@ class GraphedStep:
         t = self.t
         fp = t.fp
         fp.flat_grad.zero_()
+        fused_reduced = False
         if self._fused is not None:
             loss, correct = self._fused(self.static_x, self.static_y,
                                         self.step_dev)
             acc = correct / self.static_x.shape[0]
+            fused_reduced = self._fused.overlap_allreduce
         else:
             logits = t.model(self.static_x, train=True)
             loss, acc = t.model.loss_and_accuracy(logits, self.static_y)
             loss.backward()
-        if t.world > 1 and dist.is_initialized():
+        if not fused_reduced and t.world > 1 and dist.is_initialized():
             dist.all_reduce(fp.flat_grad, op=dist.ReduceOp.SUM)
         self._ext.sgd_step_dev(fp.flat_master, fp.flat_grad,
                                fp.flat_shadow if fp.flat_shadow is not None
