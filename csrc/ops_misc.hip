@@ -52,30 +52,32 @@ extern "C" __global__ __launch_bounds__(256)
 void pool_bwd_scatter_kernel(const ushort_t* dy, const ushort_t* y,
                              const uint8_t* amax, ushort_t* dact, float* db,
                              int Mpool, int C, int H, int W, int Wo) {
-  // fixed-size grid, grid-stride; per-column fp32 partials accumulated in
-  // LDS across the whole loop, ONE global atomic per column per block
-  // (a naive per-element global atomicAdd on db was 1.6M atomics on 64
-  // addresses — 300+ us; this form is ~64 atomics per block).
-  // vectorized: one thread handles 8 consecutive channels of one pooled
-  // window -> all reads/writes are 16B; C % 8 == 0 (32 or 64 here)
+  // one thread handles 8 consecutive channels of one pooled window; all
+  // reads/writes are 16B (C % 8 == 0).  The grid is sized so each thread
+  // walks >=4 granules (single-shot threads were pure latency-bound:
+  // 0.8-1.5 TB/s measured); index math is 32-bit with CB8 a power of two.
   __shared__ float partial[64];
   if (threadIdx.x < 64) partial[threadIdx.x] = 0.f;
   __syncthreads();
-  const int CB8 = C / 8;
-  size_t total = (size_t)Mpool * CB8;
-  size_t stride = (size_t)gridDim.x * blockDim.x;
+  const unsigned CB8 = (unsigned)C / 8;
+  const unsigned cb8_sh = (CB8 == 8) ? 3u : (CB8 == 4 ? 2u : 0u);
+  const bool cb8_pow2 = (CB8 & (CB8 - 1)) == 0;
+  unsigned total = (unsigned)Mpool * CB8;
+  unsigned stride = gridDim.x * blockDim.x;
   float local[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   int mycol = -1;
-  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+  const unsigned Ho = (unsigned)H / 2;
+  for (unsigned i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
-    int cb = (int)(i % CB8);
-    size_t q = i / CB8;
-    int c0 = cb * 8;
-    int wo = q % Wo;
-    size_t t = q / Wo;
-    int ho = t % (H / 2);
-    int n = t / (H / 2);
-    size_t base = q * C + c0;
+    unsigned cb, q;
+    if (cb8_pow2 && cb8_sh) { cb = i & (CB8 - 1); q = i >> cb8_sh; }
+    else { cb = i % CB8; q = i / CB8; }
+    int c0 = (int)cb * 8;
+    unsigned wo = q % (unsigned)Wo;
+    unsigned t = q / (unsigned)Wo;
+    unsigned ho = t % Ho;
+    unsigned n = t / Ho;
+    size_t base = (size_t)q * C + c0;
     short8 dyv = *reinterpret_cast<const short8*>(dy + base);
     short8 yv = *reinterpret_cast<const short8*>(y + base);
     uint64_t am8;
@@ -277,8 +279,11 @@ void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
                              float* db, int Mpool, int C, int H, int W, int Wo,
                              hipStream_t s) {
   long total = (long)Mpool * (C / 8);
-  int blocks = cdivh(total, 256);
-  if (blocks > 2048) blocks = 2048;  // grid-stride; bounds db atomics
+  // >=4 granules per thread for memory-level parallelism; >=256 blocks to
+  // fill the chip; <=2048 to bound the db atomics
+  int blocks = cdivh(total, 256 * 4);
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 256) blocks = cdivh(total, 256) < 256 ? cdivh(total, 256) : 256;
   hipLaunchKernelGGL(pool_bwd_scatter_kernel, dim3(blocks), dim3(256), 0, s,
                      dy, y, amax, dact, db, Mpool, C, H, W, Wo);
 }
