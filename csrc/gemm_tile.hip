@@ -260,18 +260,20 @@ void gemm_tile_kernel(GemmParams p) {
     // bijective XCD-chunk remap (cdna_hip_programming.md T1): blocks with
     // the same split-K slice / N-tile land on ONE XCD so the operand slice
     // they share stays in that XCD's L2 instead of being re-fetched via L3.
-    // Default dispatch puts block b on XCD b%8; give each XCD a CONTIGUOUS
-    // chunk of the (x-fastest) flattened grid instead.
-    int gx = gridDim.x, gz = gridDim.z;
-    int total = gx * gridDim.y * gz;
-    int flat = bx + gx * (by + gridDim.y * bz);
+    // SWZ launches use a 1-D grid (blockIdx.x only) so the observed
+    // b -> XCD b%8 mapping is unambiguous; the logical (x,y,z) shape is
+    // recomputed from the problem dims.
+    int gx = (p.M + BM - 1) / BM;
+    int gy = (p.N + BN - 1) / BN;
+    int total = gx * gy * p.splitk;
+    int flat = blockIdx.x;
     int q = total / 8, r = total % 8;
     int xcd = flat % 8, pos = flat / 8;
     int flat2 = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
     bx = flat2 % gx;
     int rest = flat2 / gx;
-    by = rest % gridDim.y;
-    bz = rest / gridDim.y;
+    by = rest % gy;
+    bz = rest / gy;
   }
   const int m0 = bx * BM;
   const int n0 = by * BN;
@@ -413,7 +415,7 @@ static inline int cdiv_host(int a, int b) { return (a + b - 1) / b; }
   }
 #define GEMM_ENTRY_SWZ(name, BM, BN, AM, BMo, EPI, OUT, PIPE)                \
   void name(const GemmParams& p, hipStream_t s) {                            \
-    dim3 grid(cdiv_host(p.M, BM), cdiv_host(p.N, BN), p.splitk);             \
+    dim3 grid(cdiv_host(p.M, BM) * cdiv_host(p.N, BN) * p.splitk);           \
     hipLaunchKernelGGL((gemm_tile_kernel<BM, BN, AM, BMo, EPI, OUT, PIPE,    \
                                          1>),                                \
                        grid, dim3(NTHREADS), 0, s, p);                       \
