@@ -443,29 +443,41 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
 
 void sgd_step(torch::Tensor master, torch::Tensor grad, torch::Tensor shadow,
               bool has_shadow, double lr, double scale, double dc_keep,
-              int64_t seed, int64_t offset) {
+              int64_t seed, int64_t offset,
+              c10::optional<torch::Tensor> momentum, double mu) {
   CHECK_CUDA(master); CHECK_F32(master); CHECK_CONTIG(master);
   CHECK_F32(grad); CHECK_CONTIG(grad);
+  float* mom = nullptr;
+  if (momentum.has_value() && momentum->defined()) {
+    CHECK_F32((*momentum)); CHECK_CONTIG((*momentum));
+    mom = momentum->data_ptr<float>();
+  }
   launch_sgd_step(master.data_ptr<float>(), grad.data_ptr<float>(),
                   has_shadow ? bf16_mut(shadow) : nullptr,
                   has_shadow ? 1 : 0, master.numel(),
                   (float)(lr * scale), (float)dc_keep, (uint64_t)seed,
-                  (uint64_t)offset, cur_stream());
+                  (uint64_t)offset, mom, (float)mu, cur_stream());
 }
 
 void sgd_step_dev(torch::Tensor master, torch::Tensor grad,
                   torch::Tensor shadow, bool has_shadow,
                   torch::Tensor lr_scale_dev, double dc_keep, int64_t seed,
-                  torch::Tensor offset_dev) {
+                  torch::Tensor offset_dev,
+                  c10::optional<torch::Tensor> momentum, double mu) {
   CHECK_CUDA(master); CHECK_F32(master); CHECK_CONTIG(master);
   CHECK_F32(grad); CHECK_CONTIG(grad);
   CHECK_F32(lr_scale_dev);
+  float* mom = nullptr;
+  if (momentum.has_value() && momentum->defined()) {
+    CHECK_F32((*momentum)); CHECK_CONTIG((*momentum));
+    mom = momentum->data_ptr<float>();
+  }
   launch_sgd_step_dev(master.data_ptr<float>(), grad.data_ptr<float>(),
                       has_shadow ? bf16_mut(shadow) : nullptr,
                       has_shadow ? 1 : 0, master.numel(),
                       lr_scale_dev.data_ptr<float>(), (float)dc_keep,
                       (uint64_t)seed, offset_dev.data_ptr<long>(),
-                      cur_stream());
+                      mom, (float)mu, cur_stream());
 }
 
 void transpose_bf16_batch(std::vector<torch::Tensor> srcs,
@@ -525,9 +537,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_pool_bwd_into", &conv_pool_bwd_into,
         "conv backward accumulating dw/db into bucket views");
   m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax-CE (+grad)");
-  m.def("sgd_step", &sgd_step, "fused flat SGD apply");
+  m.def("sgd_step", &sgd_step, "fused flat SGD(+momentum) apply",
+        py::arg("master"), py::arg("grad"), py::arg("shadow"),
+        py::arg("has_shadow"), py::arg("lr"), py::arg("scale"),
+        py::arg("dc_keep"), py::arg("seed"), py::arg("offset"),
+        py::arg("momentum") = c10::nullopt, py::arg("mu") = 0.0);
   m.def("sgd_step_dev", &sgd_step_dev,
-        "SGD apply with device-side lr/offset (hipGraph-capturable)");
+        "SGD apply with device-side lr/offset (hipGraph-capturable)",
+        py::arg("master"), py::arg("grad"), py::arg("shadow"),
+        py::arg("has_shadow"), py::arg("lr_scale_dev"), py::arg("dc_keep"),
+        py::arg("seed"), py::arg("offset_dev"),
+        py::arg("momentum") = c10::nullopt, py::arg("mu") = 0.0);
   m.def("step_advance", &step_advance,
         "device-side staircase LR + step increment (inside the graph)");
   m.def("linear_act_fwd_dev", &linear_act_fwd_dev,

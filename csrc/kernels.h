@@ -48,12 +48,14 @@ void launch_softmax_xent(const unsigned short* logits, const long* labels,
                          hipStream_t);
 void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
-                     uint64_t seed, uint64_t offset, hipStream_t);
+                     uint64_t seed, uint64_t offset, float* momentum, float mu,
+                     hipStream_t);
 // graph-capturable variants: lr_scale / RNG offset read from device memory
 void launch_sgd_step_dev(float* master, const float* grad,
                          unsigned short* shadow, int has_shadow, long n,
                          const float* lr_scale_dev, float dc_keep,
-                         uint64_t seed, const long* offset_dev, hipStream_t);
+                         uint64_t seed, const long* offset_dev,
+                         float* momentum, float mu, hipStream_t);
 void launch_step_advance(long* step_dev, float* lr_scale_dev, float lr0,
                          float decay, int decay_steps, float inv_contrib,
                          hipStream_t);

@@ -174,7 +174,10 @@ extern "C" __global__ __launch_bounds__(256)
 void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
                      uint64_t seed, uint64_t offset,
-                     const float* lr_scale_dev, const long* offset_dev) {
+                     const float* lr_scale_dev, const long* offset_dev,
+                     float* momentum, float mu) {
+  // momentum (optional): v = mu*v + g; w -= lr*v  (v=nullptr => plain SGD,
+  // the reference's GradientDescentOptimizer semantics)
   if (lr_scale_dev) lr_scale = *lr_scale_dev;
   if (offset_dev) offset = (uint64_t)*offset_dev;
   long i = (long)(blockIdx.x) * blockDim.x + threadIdx.x;
@@ -192,6 +195,13 @@ void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
         g.z *= (ph.z * s < dc_keep) ? 1.f : 0.f;
         g.w *= (ph.w * s < dc_keep) ? 1.f : 0.f;
       }
+      if (momentum) {
+        float4 v = *reinterpret_cast<const float4*>(momentum + base);
+        v.x = mu * v.x + g.x; v.y = mu * v.y + g.y;
+        v.z = mu * v.z + g.z; v.w = mu * v.w + g.w;
+        *reinterpret_cast<float4*>(momentum + base) = v;
+        g = v;
+      }
       m.x -= lr_scale * g.x; m.y -= lr_scale * g.y;
       m.z -= lr_scale * g.z; m.w -= lr_scale * g.w;
       *reinterpret_cast<float4*>(master + base) = m;
@@ -206,6 +216,11 @@ void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
       for (long j = base; j < n; ++j) {
         float g = grad[j];
         if (dc_keep > 0.f) g *= (u[j - base] < dc_keep) ? 1.f : 0.f;
+        if (momentum) {
+          float v = mu * momentum[j] + g;
+          momentum[j] = v;
+          g = v;
+        }
         float m = master[j] - lr_scale * g;
         master[j] = m;
         if (has_shadow) shadow[j] = f2bf(m);
@@ -298,26 +313,28 @@ void launch_softmax_xent(const unsigned short* logits, const long* labels,
 
 void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
-                     uint64_t seed, uint64_t offset, hipStream_t s) {
+                     uint64_t seed, uint64_t offset, float* momentum, float mu,
+                     hipStream_t s) {
   long groups = (n + 3) / 4;
   int blocks = (int)min((long)2048, (groups + 255) / 256);
   if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(sgd_step_kernel, dim3(blocks), dim3(256), 0, s, master,
                      grad, shadow, has_shadow, n, lr_scale, dc_keep, seed,
-                     offset, (const float*)nullptr, (const long*)nullptr);
+                     offset, (const float*)nullptr, (const long*)nullptr,
+                     momentum, mu);
 }
 
 void launch_sgd_step_dev(float* master, const float* grad,
                          unsigned short* shadow, int has_shadow, long n,
                          const float* lr_scale_dev, float dc_keep,
                          uint64_t seed, const long* offset_dev,
-                         hipStream_t s) {
+                         float* momentum, float mu, hipStream_t s) {
   long groups = (n + 3) / 4;
   int blocks = (int)min((long)2048, (groups + 255) / 256);
   if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(sgd_step_kernel, dim3(blocks), dim3(256), 0, s, master,
                      grad, shadow, has_shadow, n, 0.f, dc_keep, seed, 0,
-                     lr_scale_dev, offset_dev);
+                     lr_scale_dev, offset_dev, momentum, mu);
 }
 
 void launch_step_advance(long* step_dev, float* lr_scale_dev, float lr0,

@@ -110,7 +110,9 @@ class GraphedStep:
                                else fp.flat_master,
                                fp.flat_shadow is not None,
                                self.lr_scale_dev, self.dc_keep,
-                               t.flags.seed, self.step_dev)
+                               t.flags.seed, self.step_dev,
+                               momentum=t.flat_momentum,
+                               mu=t.flags.momentum)
         fp.refresh_transposes()
         # advance step + LR on-device for the next replay
         self._ext.step_advance(self.step_dev, self.lr_scale_dev,

@@ -109,6 +109,8 @@ class Trainer:
         self._num_examples = 60000  # overwritten by train(); LR-schedule default
         self._graph = None
         self._graph_tried = False
+        self.flat_momentum = (torch.zeros_like(self.fp.flat_master)
+                              if flags.momentum > 0 else None)
         # parameter-init parity across ranks (broadcast once)
         if world > 1 and dist.is_initialized():
             dist.broadcast(self.fp.flat_master, src=0)
@@ -187,7 +189,8 @@ class Trainer:
                         drop_connect_keep=(flags.drop_connect_probability
                                            if flags.drop_connect else None),
                         seed=flags.seed, offset=self.step,
-                        shadow=self.fp.flat_shadow)
+                        shadow=self.fp.flat_shadow,
+                        momentum=self.flat_momentum, mu=flags.momentum)
             self.fp.refresh_transposes()
             self.num_contributors = contributors
         self.step += 1

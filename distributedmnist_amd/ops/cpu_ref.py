@@ -151,20 +151,27 @@ def softmax_xent_fwd(logits, labels):
 
 def sgd_step(master, grad, lr: float, grad_scale: float = 1.0,
              drop_connect_keep: float | None = None,
-             gen: torch.Generator | None = None, shadow=None):
-    """In-place w -= lr * grad_scale * g  (flat fp32 tensors).
+             gen: torch.Generator | None = None, shadow=None,
+             momentum=None, mu: float = 0.0):
+    """In-place w -= lr * grad_scale * (g or momentum-filtered g).
 
     drop_connect_keep: if set, per-element Bernoulli(keep) mask on the grad
     (reference drop_connect: grad * mask, NO rescale —
     distributed_train.py:414-416).
+    momentum/mu: optional heavyweight-ball momentum v = mu*v + g; w -= lr*v
+    (the reference used plain GradientDescentOptimizer; momentum is the
+    north-star's optional extension).
     shadow: optional flat low-precision copy refreshed after the update.
     """
-    g = grad.float()
+    g = grad.float() * grad_scale
     if drop_connect_keep is not None:
         mask = (torch.rand(g.shape, device=g.device, dtype=torch.float32,
                            generator=gen) < drop_connect_keep).float()
         g = g * mask
-    master.add_(g, alpha=-(lr * grad_scale))
+    if momentum is not None:
+        momentum.mul_(mu).add_(g)
+        g = momentum
+    master.add_(g, alpha=-lr)
     if shadow is not None:
         shadow.copy_(master.to(shadow.dtype))
     return master

@@ -173,7 +173,8 @@ def softmax_xent(logits, labels):
 
 def sgd_step(master: torch.Tensor, grad: torch.Tensor, lr: float,
              grad_scale: float = 1.0, drop_connect_keep=None,
-             seed: int = 0, offset: int = 0, shadow=None):
+             seed: int = 0, offset: int = 0, shadow=None,
+             momentum=None, mu: float = 0.0):
     """Fused flat SGD apply: master -= lr*scale*(grad [* bernoulli(keep)]).
 
     Optionally refreshes the bf16 `shadow` copy (one kernel on GPU).
@@ -185,11 +186,13 @@ def sgd_step(master: torch.Tensor, grad: torch.Tensor, lr: float,
                           shadow if shadow is not None else master,
                           shadow is not None, float(lr), float(grad_scale),
                           -1.0 if drop_connect_keep is None else float(drop_connect_keep),
-                          int(seed), int(offset))
+                          int(seed), int(offset),
+                          momentum=momentum, mu=float(mu))
     else:
         gen = None
         if drop_connect_keep is not None:
             gen = torch.Generator(device="cpu")
             gen.manual_seed((seed * 0x9E3779B97F4A7C15 + offset) % (2**63))
         cpu_ref.sgd_step(master, grad, lr, grad_scale,
-                         drop_connect_keep, gen, shadow)
+                         drop_connect_keep, gen, shadow,
+                         momentum=momentum, mu=mu)

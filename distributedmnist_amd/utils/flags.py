@@ -46,6 +46,8 @@ def build_train_parser() -> argparse.ArgumentParser:
     p.add_argument("--save_interval_secs", type=int, default=20)
     p.add_argument("--save_summaries_secs", type=int, default=300)
     p.add_argument("--initial_learning_rate", type=float, default=0.01)
+    p.add_argument("--momentum", type=float, default=0.0,
+                   help="SGD momentum mu (0 = reference GradientDescent parity)")
     p.add_argument("--num_epochs_per_decay", type=float, default=1.0)
     p.add_argument("--learning_rate_decay_factor", type=float, default=0.95)
     p.add_argument("--drop_connect_probability", type=float, default=0.9,

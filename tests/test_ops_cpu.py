@@ -160,3 +160,30 @@ def test_sgd_step_shadow_refresh():
     grad = torch.randn(64)
     Fx.sgd_step(master, grad, lr=0.01, shadow=shadow)
     torch.testing.assert_close(shadow, master.to(torch.bfloat16))
+
+
+def test_sgd_momentum():
+    """v = mu*v + scale*g; w -= lr*v (heavyweight-ball)."""
+    master = torch.zeros(16)
+    mom = torch.zeros(16)
+    g = torch.ones(16)
+    Fx.sgd_step(master, g, lr=0.1, grad_scale=1.0, momentum=mom, mu=0.9)
+    torch.testing.assert_close(master, torch.full((16,), -0.1))
+    torch.testing.assert_close(mom, torch.ones(16))
+    Fx.sgd_step(master, g, lr=0.1, grad_scale=1.0, momentum=mom, mu=0.9)
+    torch.testing.assert_close(mom, torch.full((16,), 1.9))
+    torch.testing.assert_close(master, torch.full((16,), -0.29))
+
+
+def test_trainer_momentum_flag(tmp_path):
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+    flags = build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", str(tmp_path), "--batch_size",
+         "16", "--max_steps", "4", "--model", "mlp", "--device", "cpu",
+         "--momentum", "0.9", "--save_interval_secs", "100000"])
+    t = Trainer(flags)
+    assert t.flat_momentum is not None
+    ds = make_dataset(flags, 0, 1, t.device, t.compute_dtype)
+    t.train(ds)
+    assert float(t.flat_momentum.abs().sum()) > 0

@@ -388,3 +388,19 @@ def test_train_loop_and_eval_gpu(ext, tmp_path):
                   out.stdout)
     assert m, out.stdout
     assert (tmp_path / "eval").exists()
+
+
+def test_sgd_momentum_gpu(ext):
+    master = torch.zeros(100001).cuda()
+    mom = torch.zeros(100001).cuda()
+    g = torch.randn(100001).cuda()
+    shadow = torch.zeros(100001, dtype=bf16).cuda()
+    ext.sgd_step(master, g, shadow, True, 0.1, 1.0, -1.0, 0, 0,
+                 momentum=mom, mu=0.9)
+    torch.testing.assert_close(mom, g)
+    torch.testing.assert_close(master, -0.1 * g)
+    ext.sgd_step(master, g, shadow, True, 0.1, 1.0, -1.0, 0, 0,
+                 momentum=mom, mu=0.9)
+    torch.testing.assert_close(mom, 1.9 * g)
+    torch.testing.assert_close(master, -0.29 * g, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(shadow, master.to(bf16))
