@@ -207,6 +207,10 @@ class Trainer:
         if restored is not None:
             step0, payload = restored
             self.fp.load_flat(payload["flat_master"])
+            if (self.flat_momentum is not None
+                    and payload.get("flat_momentum") is not None):
+                self.flat_momentum.copy_(
+                    payload["flat_momentum"].to(self.device))
             self.step = step0
             if self.is_chief:
                 log.info("Restored checkpoint at step %d", step0)
@@ -274,6 +278,8 @@ class Trainer:
     def checkpoint_payload(self):
         return {
             "flat_master": self.fp.flat_master.detach().cpu().clone(),
+            "flat_momentum": (self.flat_momentum.detach().cpu().clone()
+                              if self.flat_momentum is not None else None),
             "model_state": {k: v.cpu() for k, v in
                             self.fp.state_dict_params().items()},
             "model": self.flags.model,
