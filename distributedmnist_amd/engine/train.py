@@ -143,6 +143,17 @@ class Trainer:
         if eligible:
             from .graphstep import try_graph
             self._graph = try_graph(self, tuple(images.shape))
+        if self.world > 1 and dist.is_initialized():
+            # ALL ranks must agree on graph-vs-eager: the graphed fused step
+            # issues a bucketed 2-collective sequence, eager issues one —
+            # a mixed fleet would deadlock the communicator.
+            ok = torch.tensor(
+                [1.0 if self._graph is not None else 0.0],
+                device=self.device if self.device.type == "cuda" else "cpu")
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if float(ok.item()) < 0.5 and self._graph is not None:
+                log.warning("hipGraph disabled: another rank failed capture")
+                self._graph = None
         return self._graph
 
     def graph_or_eager_step(self, images, labels):
