@@ -238,6 +238,11 @@ DEV void writeB(ushort_t (*Bs)[LDK], int tid, const short8 (&regs)[CH]) {
 //   huge-grid conv gathers where 4-5 blocks/CU of TLP hide latency better
 //   than the in-wave pipeline (measured: conv2 fwd 103us PIPE0 vs 120us
 //   PIPE1; fc1 fwd 94us PIPE0 vs 59us PIPE1).
+// PIPE=2: single buffer + register-carried next tile (T14 write-after-
+//   barrier): next tile's gather loads issue BEFORE the current tile's
+//   MFMAs so their latency hides under compute, at PIPE=0's LDS footprint
+//   (the dW GEMMs were memory-level-parallelism-bound: ~6 16B loads in
+//   flight per thread serial-blocked at the ds_write vmcnt = ~2 TB/s).
 template <int BM, int BN, int AMODE, int BMODE, int EPI, int OUT, int PIPE,
           int SWZ = 0>
 __global__ __launch_bounds__(NTHREADS)
@@ -246,7 +251,7 @@ void gemm_tile_kernel(GemmParams p) {
   constexpr int MI = WM / 16, NI = WN / 16;
   constexpr int CHA = (BM * BK / 8) / NTHREADS;
   constexpr int CHB = (BN * BK / 8) / NTHREADS;
-  constexpr int DB = PIPE ? 2 : 1;
+  constexpr int DB = (PIPE == 1) ? 2 : 1;
   static_assert(CHA >= 1 && CHB >= 1, "tile too small for 256 threads");
   __shared__ __align__(16) ushort_t As[DB][BM][LDK];
   __shared__ __align__(16) ushort_t Bs[DB][BN][LDK];
@@ -296,6 +301,48 @@ void gemm_tile_kernel(GemmParams p) {
 
   short8 ra[CHA], rb[CHB];
   int cur = 0;
+  if (PIPE == 2) {
+    if (nt > 0) {
+      loadA<BM, AMODE, CHA>(p, m0, kbeg, kend, tid, ra);
+      loadB<BN, BMODE, CHB>(p, n0, kbeg, kend, tid, rb);
+      writeA<BM, AMODE, CHA>(As[0], tid, ra);
+      writeB<BN, BMODE, CHB>(Bs[0], tid, rb);
+      __syncthreads();
+    }
+    for (int t = 0; t < nt; ++t) {
+      const bool have_next = (t + 1) < nt;
+      if (have_next) {
+        int kn = kbeg + (t + 1) * BK;
+        loadA<BM, AMODE, CHA>(p, m0, kn, kend, tid, ra);  // hides under MFMA
+        loadB<BN, BMODE, CHB>(p, n0, kn, kend, tid, rb);
+      }
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+        short8 af[MI], bf[NI];
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+          af[mi] = *reinterpret_cast<const short8*>(
+              &As[0][wr * WM + mi * 16 + lrow][kh * 32 + kq]);
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni)
+          bf[ni] = *reinterpret_cast<const short8*>(
+              &Bs[0][wc * WN + ni * 16 + lrow][kh * 32 + kq]);
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < NI; ++ni)
+            acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      }
+      if (have_next) {
+        __syncthreads();   // all waves done reading the buffer
+        writeA<BM, AMODE, CHA>(As[0], tid, ra);
+        writeB<BN, BMODE, CHB>(Bs[0], tid, rb);
+        __syncthreads();
+      }
+    }
+    goto epilogue;
+  }
   if (PIPE && nt > 0) {
     loadA<BM, AMODE, CHA>(p, m0, kbeg, kend, tid, ra);
     loadB<BN, BMODE, CHB>(p, n0, kbeg, kend, tid, rb);
@@ -352,6 +399,7 @@ void gemm_tile_kernel(GemmParams p) {
   }
 
   // ---- epilogue ----------------------------------------------------------
+epilogue:
   const int frow = (lane >> 4) * 4;  // C/D: row=(lane>>4)*4+reg, col=lane&15
   const int fcol = lane & 15;
 #pragma unroll
@@ -430,8 +478,8 @@ GEMM_ENTRY(gemm_fwd_drop_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF1
 GEMM_ENTRY(gemm_fwd_drop_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dx_128, 128, 128, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dx_64, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
-GEMM_ENTRY_SWZ(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
-GEMM_ENTRY_SWZ(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 1)
+GEMM_ENTRY_SWZ(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 2)
+GEMM_ENTRY_SWZ(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 2)
 // B-transposed (pre-transposed weight) fwd variants: ldb = K, vector staging
 GEMM_ENTRY(gemm_fwd_bias_128_bt, 128, 128, A_N, B_NMAJ, EPI_BIAS, OUT_BF16, 1)
 GEMM_ENTRY(gemm_fwd_bias_64_bt, 64, 64, A_N, B_NMAJ, EPI_BIAS, OUT_BF16, 1)
@@ -447,6 +495,6 @@ GEMM_ENTRY(conv_dx_gemm, 128, 32, A_CONV_DX, B_CONV_DX_W, EPI_NONE, OUT_BF16, 0)
 // conv dW: gather-staged -> occupancy (PIPE=0) beats the in-wave pipeline
 // (measured 2069us PIPE=1 vs 1309 at BM=64 PIPE=0); BM=128 halves the
 // per-M-tile dact re-reads
-GEMM_ENTRY_SWZ(conv_dw_gemm, 128, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)
+GEMM_ENTRY_SWZ(conv_dw_gemm, 128, 64, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 2)
 // conv1 dW: M=25 -> 32x32 tile (78% M-utilization vs 39% at BM=64)
-GEMM_ENTRY_SWZ(conv1_dw_gemm, 32, 32, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 0)
+GEMM_ENTRY_SWZ(conv1_dw_gemm, 32, 32, A_CONV_DW, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 2)
