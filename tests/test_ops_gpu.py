@@ -141,8 +141,13 @@ def test_conv_pool_fwd(ext, NB, H, W, Cin, Cout):
     b = torch.randn(Cout) * 0.1
     from distributedmnist_amd.ops import cpu_ref
     y_ref, amax_ref = cpu_ref.conv_pool_fwd(x, w, b)
-    y, amax = ext.conv_pool_fwd(to_gpu_bf16(x), to_gpu_bf16(w),
-                                b.cuda().float())
+    xg, wg, bg = to_gpu_bf16(x), to_gpu_bf16(w), b.cuda().float()
+    y, amax = ext.conv_pool_fwd(xg, wg, bg)
+    # race detector: repeated launches must be bitwise identical (the
+    # kernel has no atomics — any variation is a synchronization bug)
+    for _ in range(4):
+        y2, amax2 = ext.conv_pool_fwd(xg, wg, bg)
+        assert torch.equal(y, y2) and torch.equal(amax, amax2),             "conv_pool_fwd is nondeterministic: RACE in the kernel"
     assert y.shape == y_ref.shape
     assert_close_bf16(y, y_ref, scale=float(y_ref.abs().max()))
     # argmax routing: fp32-ref vs bf16-kernel near-ties legitimately pick
