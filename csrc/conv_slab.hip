@@ -127,6 +127,13 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
     for (int kh2 = 0; kh2 < 2; ++kh2) {
       int kbase = kt + kh2 * 32 + kq;       // k = khkw*CIN + ci
       int khkw = kbase / CIN, ci = kbase % CIN;
+      // K-tail guard: K=25*CIN is not a multiple of BK, so the last
+      // K-step's upper fragments decode khkw >= 25.  Their B operand is
+      // zero (staged with a k<K bound) so the product vanishes — but the
+      // A read MUST stay inside the slab: an out-of-allocation LDS read
+      // can return a NaN bit pattern and NaN*0 = NaN (this was a
+      // run-order-dependent test flake).
+      if (khkw >= 25) { khkw = 0; ci = 0; }
       int kh = khkw / 5, kw = khkw % 5;
       int shift = ((kh - 2) * WP + (kw - 2)) * PST + ci;
       short8 af[MI], bf[NI];
