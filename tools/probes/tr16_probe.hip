@@ -20,7 +20,7 @@ __global__ void tr16_probe(unsigned short* out, int scheme) {
   }
   uint2 v;
   asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0\n\ts_waitcnt lgkmcnt(0)"
-               : "=v"(v) : "v"(addr));
+               : "=v"(v) : "v"(addr) : "memory");
   __builtin_amdgcn_sched_barrier(0);
   out[tid * 4 + 0] = (unsigned short)(v.x & 0xffff);
   out[tid * 4 + 1] = (unsigned short)(v.x >> 16);
