@@ -3,11 +3,15 @@
 // per SCHEME; we dump the 4 u16 elements every lane receives.
 #include <hip/hip_runtime.h>
 #include <stdio.h>
+#include <stdint.h>
 
 __global__ void tr16_probe(unsigned short* out, int scheme) {
   __shared__ __align__(16) unsigned short lds[2048];
   int tid = threadIdx.x;
   for (int i = tid; i < 2048; i += 64) lds[i] = (unsigned short)i;
+  // escape hatch: the tr16 asm below only sees an INTEGER offset, so LLVM
+  // would otherwise prove the array dead and delete the fill
+  asm volatile("" : : "v"((unsigned)(uintptr_t)&lds[0]) : "memory");
   __syncthreads();
   unsigned addr = 0;  // BYTE address into LDS
   int l = tid;
