@@ -35,6 +35,16 @@ int pick_splitk(int mtiles, int ntiles, int ksteps) {
   return want;
 }
 
+// glds + tr16 dW staging (dw_tr.hip); DMNIST_DW_TR=0 reverts to the
+// scatter-staged gemm_tile path for A/B comparison
+bool dw_tr_enabled() {
+  static int v = [] {
+    const char* e = getenv("DMNIST_DW_TR");
+    return e ? atoi(e) : 1;
+  }();
+  return v != 0;
+}
+
 // --------------------------------------------------------------------------
 torch::Tensor linear_act_fwd_impl(torch::Tensor x, torch::Tensor w,
                                   torch::Tensor b, bool relu, double p_keep,
@@ -134,10 +144,16 @@ std::vector<torch::Tensor> linear_act_bwd_impl(torch::Tensor dy, torch::Tensor x
     p.M = K; p.N = N; p.K = B;
     p.lda = K;  // A_T: A[m][k'] = x[k'*lda + m]
     p.ldb = N; p.ldc = N;
-    bool big = cdiv(K, 128) * cdiv(N, 128) >= 128;
-    int bm = big ? 128 : 64;
-    p.splitk = pick_splitk(cdiv(K, bm), cdiv(N, bm), cdiv(B, 32));
-    (big ? gemm_dw_128 : gemm_dw_64)(p, s);
+    if (dw_tr_enabled() && K % 8 == 0 && N % 64 == 0) {
+      int bn = (N % 128 == 0) ? 128 : 64;
+      p.splitk = pick_splitk(cdiv(K, 128), cdiv(N, bn), cdiv(B, 32));
+      (bn == 128 ? gemm_dw_tr_128 : gemm_dw_tr_64)(p, s);
+    } else {
+      bool big = cdiv(K, 128) * cdiv(N, 128) >= 128;
+      int bm = big ? 128 : 64;
+      p.splitk = pick_splitk(cdiv(K, bm), cdiv(N, bm), cdiv(B, 32));
+      (big ? gemm_dw_128 : gemm_dw_64)(p, s);
+    }
   }
   torch::Tensor dx;
   if (need_dx) {
@@ -267,7 +283,10 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
     } else {
       int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
       p.splitk = std::min(cdiv(p.K, 64), std::max(1, 2048 / tiles));
-      conv_dw_gemm(p, s);
+      if (dw_tr_enabled() && Cin % 8 == 0 && Cout % 64 == 0)
+        conv_dw_tr(p, s);
+      else
+        conv_dw_gemm(p, s);
     }
   }
   torch::Tensor dx;
@@ -338,6 +357,12 @@ void linear_dw_into(torch::Tensor x, torch::Tensor dyeff,
   p.C = dw_out.data_ptr();
   p.M = K; p.N = N; p.K = B;
   p.lda = K; p.ldb = N; p.ldc = N;
+  if (dw_tr_enabled() && K % 8 == 0 && N % 64 == 0) {
+    int bn = (N % 128 == 0) ? 128 : 64;
+    p.splitk = pick_splitk(cdiv(K, 128), cdiv(N, bn), cdiv(B, 32));
+    (bn == 128 ? gemm_dw_tr_128 : gemm_dw_tr_64)(p, cur_stream());
+    return;
+  }
   bool big = cdiv(K, 128) * cdiv(N, 128) >= 128;
   int bm = big ? 128 : 64;
   p.splitk = pick_splitk(cdiv(K, bm), cdiv(N, bm), cdiv(B, 32));
@@ -396,7 +421,10 @@ void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
   } else {
     int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
     p.splitk = std::min(cdiv(p.K, 64), std::max(1, 2048 / tiles));
-    conv_dw_gemm(p, s);
+    if (dw_tr_enabled() && Cin % 8 == 0 && Cout % 64 == 0)
+      conv_dw_tr(p, s);
+    else
+      conv_dw_gemm(p, s);
   }
 }
 

@@ -24,7 +24,7 @@ def main():
     abi = int(torch.compiled_with_cxx11_abi())
 
     srcs = [os.path.join(ROOT, f) for f in
-            ("gemm_tile.hip", "conv_slab.hip", "ops_misc.hip", "bindings.cpp")]
+            ("gemm_tile.hip", "conv_slab.hip", "dw_tr.hip", "ops_misc.hip", "bindings.cpp")]
     out = os.path.join(ROOT, "_dmnist_hip.so")
 
     cmd = [hipcc, "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",

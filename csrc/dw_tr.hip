@@ -1,0 +1,229 @@
+// tr16 + glds dW GEMM for gfx950 — the "designed around glds" staging the
+// guide prescribes for k-strided operands (cdna_hip_programming.md: glds
+// pipelining + ds_read_b64_tr_b16; semantics pinned empirically in
+// profiles/r01_tr16_probe.md with tools/probes/tr16_probe.hip).
+//
+// dW = A^T-ish reductions over the batch-pixel axis K: both operands are
+// K-MAJOR in memory (x / h2 rows are per-pixel, dact / dyeff rows are
+// per-pixel).  The old path (gemm_tile.hip A_T/A_CONV_DW) had to SCATTER
+// each staged row element-by-element into an m-major LDS image so the MFMA
+// fragment reads were contiguous — 8 ds_write_b16 per 16B of operand, and
+// the ds_write's vmcnt wait serialized the global loads (~2 TB/s MLP wall).
+// Here the LDS image stays K-MAJOR and LINEAR:
+//   * staging is pure glds (global_load_lds_dwordx4): no VGPR round-trip,
+//     no scatter, fire-and-forget — a whole tile is in flight while the
+//     previous tile's MFMAs run (2-buffer overlap, one barrier/K-step);
+//   * fragment reads use ds_read_b64_tr_b16, the LDS transpose read: within
+//     each 16-lane group, lane g supplies &img[kq + (g>>2)][n0 + 4*(g&3)]
+//     and lane i receives img[kq + j][n0 + i] (j = 0..3) — exactly an MFMA
+//     operand fragment, with ZERO staging-time transpose work.
+// Out-of-bounds / K-tail lanes redirect their glds source to a zeroed
+// device page (glds cannot conditionally mask, and every lane of the wave
+// must participate).
+//
+// Modes: AM_PLAIN  A[m][k] = Asrc[k*lda + m]                (fc dW: x^T)
+//        AM_CONV5  A[m=(khkw,ci)][k=pixel] 5x5-SAME gather  (conv dW)
+// Requirements: M % 8 == 0, N % BN == 0, Cin % 8 == 0 (conv), 16B-aligned
+// sources.  Output: fp32 split-K atomics straight into the grad bucket.
+
+#include "common.h"
+#include "kernels.h"
+
+#define NT 256
+#define TBK 64
+#define TBM 128
+
+namespace {
+
+__device__ __align__(16) unsigned short g_zero_page[8];  // zero-init
+
+// two transpose reads -> one MFMA short8 fragment (k-contiguous per lane)
+DEV short8 tr16_frag(unsigned a0, unsigned a1) {
+  uint2 lo, hi;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+      "ds_read_b64_tr_b16 %1, %3 offset:0\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(lo), "=&v"(hi)  // early-clobber: written before %3 is read
+      : "v"(a0), "v"(a1)
+      : "memory");
+  union { unsigned u[4]; short8 s; } r;
+  r.u[0] = lo.x; r.u[1] = lo.y; r.u[2] = hi.x; r.u[3] = hi.y;
+  return r.s;
+}
+
+enum { AM_PLAIN = 0, AM_CONV5 = 1 };
+
+template <int BN, int AMODE>
+__global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
+  constexpr int WM = TBM / 2, WN = BN / 2;
+  constexpr int MI = WM / 16, NI = WN / 16;
+  constexpr int AQ = TBM / 32;        // A glds per wave (1KB each)
+  constexpr int BQ = BN / 32;         // B glds per wave
+  constexpr int ALPR = TBM / 8;       // lanes per A row
+  constexpr int BLPR = BN / 8;        // lanes per B row
+  __shared__ __align__(16) ushort_t Aimg[2][TBK][TBM];  // k-major, LINEAR
+  __shared__ __align__(16) ushort_t Bimg[2][TBK][BN];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+
+  // XCD-chunk swizzle on the 1-D grid (same remap as gemm_tile SWZ): blocks
+  // sharing a split-K slice land on one XCD so the operand slice stays in
+  // that XCD's L2.
+  int gx = (p.M + TBM - 1) / TBM;
+  int gy = (p.N + BN - 1) / BN;
+  int total = gx * gy * p.splitk;
+  int flat = blockIdx.x;
+  int q8 = total / 8, r8 = total % 8;
+  int xcd = flat % 8, pos = flat / 8;
+  int flat2 = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + pos;
+  const int m0 = (flat2 % gx) * TBM;
+  int rest = flat2 / gx;
+  const int n0 = (rest % gy) * BN;
+  const int bz = rest / gy;
+
+  int ksteps_total = (p.K + TBK - 1) / TBK;
+  int steps_per = (ksteps_total + p.splitk - 1) / p.splitk;
+  int kbeg = bz * steps_per * TBK;
+  int kend = min(p.K, kbeg + steps_per * TBK);
+  int nt = (kend - kbeg + TBK - 1) / TBK;
+  if (nt < 0) nt = 0;
+
+  f32x4 acc[MI][NI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // ---- per-lane glds source decode (loop-invariant parts) ----------------
+  // A: glds gi = wave*AQ+q stages rows gi*(512/TBM) + lane/ALPR at columns
+  //    (lane%ALPR)*8..+7; LDS dest = &Aimg[buf][row0][0] + lane*16 (linear).
+  const int amc = m0 + (lane % ALPR) * 8;  // m column this lane stages
+  const bool am_ok = amc < p.M;
+  int akh = 0, akw = 0, aci = 0;
+  if (AMODE == AM_CONV5) {
+    int khkw = amc / p.Cin;
+    aci = amc % p.Cin;
+    akh = khkw / 5 - 2;
+    akw = khkw % 5 - 2;
+  }
+  const int bnc = n0 + (lane % BLPR) * 8;  // n column this lane stages
+  const bool bn_ok = bnc < p.N;
+
+  const auto issue = [&](int buf, int kt) {
+#pragma unroll
+    for (int q = 0; q < AQ; ++q) {
+      int gi = wave * AQ + q;
+      int row0 = gi * (512 / TBM);
+      int k = kt + row0 + lane / ALPR;
+      const ushort_t* src = g_zero_page;
+      if (k < kend && am_ok) {
+        if (AMODE == AM_PLAIN) {
+          src = p.A + (size_t)k * p.lda + amc;
+        } else {
+          int w_ = k % p.CW;
+          int t2 = k / p.CW;
+          int h_ = t2 % p.CH;
+          int n_ = t2 / p.CH;
+          int y = h_ + akh, x = w_ + akw;
+          if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
+            src = p.A + (((size_t)n_ * p.CH + y) * p.CW + x) * p.Cin + aci;
+        }
+      }
+      __builtin_amdgcn_global_load_lds(src, &Aimg[buf][row0][0], 16, 0, 0);
+    }
+#pragma unroll
+    for (int q = 0; q < BQ; ++q) {
+      int gi = wave * BQ + q;
+      int row0 = gi * (512 / BN);
+      int k = kt + row0 + lane / BLPR;
+      const ushort_t* src = (k < kend && bn_ok)
+                                ? p.B + (size_t)k * p.ldb + bnc
+                                : g_zero_page;
+      __builtin_amdgcn_global_load_lds(src, &Bimg[buf][row0][0], 16, 0, 0);
+    }
+  };
+
+  // ---- tr16 fragment addressing (loop-invariant parts) -------------------
+  // within each 16-lane group: lane g addresses (row kq + (g>>2), col
+  // frag_n0 + 4*(g&3)); the second read is 4 rows further down.
+  const int slot_r = (lane & 15) >> 2;
+  const int slot_c = (lane & 3) * 4;
+  const int kgrp = (lane >> 4) * 8;  // this group's k-offset within the half
+
+  int cur = 0;
+  if (nt > 0) issue(0, kbeg);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt) issue(cur ^ 1, kbeg + (t + 1) * TBK);  // flight under MFMA
+    unsigned abase = (unsigned)(uintptr_t)&Aimg[cur][0][0];
+    unsigned bbase = (unsigned)(uintptr_t)&Bimg[cur][0][0];
+#pragma unroll
+    for (int kh2 = 0; kh2 < 2; ++kh2) {
+      int krow = kh2 * 32 + kgrp + slot_r;
+      short8 af[MI], bf[NI];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi) {
+        unsigned a0 = abase + (unsigned)((krow * TBM + wr * WM + mi * 16 + slot_c) * 2);
+        af[mi] = tr16_frag(a0, a0 + 4 * TBM * 2);
+      }
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni) {
+        unsigned b0 = bbase + (unsigned)((krow * BN + wc * WN + ni * 16 + slot_c) * 2);
+        bf[ni] = tr16_frag(b0, b0 + 4 * BN * 2);
+      }
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    // one barrier per K-step; the implicit vmcnt(0) drain at __syncthreads
+    // completes tile t+1's glds (2-buffer overlap)
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int frow = (lane >> 4) * 4;
+  const int fcol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) {
+      int gc = n0 + wc * WN + ni * 16 + fcol;
+      if (gc >= p.N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = m0 + wr * WM + mi * 16 + frow + r;
+        if (gr >= p.M) continue;
+        atomicAdd(reinterpret_cast<float*>(p.C) + (size_t)gr * p.ldc + gc,
+                  acc[mi][ni][r]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+static inline int cdiv_h(int a, int b) { return (a + b - 1) / b; }
+
+void conv_dw_tr(const GemmParams& p, hipStream_t s) {
+  dim3 grid(cdiv_h(p.M, TBM) * cdiv_h(p.N, 64) * p.splitk);
+  hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5>), grid, dim3(NT), 0, s, p);
+}
+
+void gemm_dw_tr_128(const GemmParams& p, hipStream_t s) {
+  dim3 grid(cdiv_h(p.M, TBM) * cdiv_h(p.N, 128) * p.splitk);
+  hipLaunchKernelGGL((dw_tr_kernel<128, AM_PLAIN>), grid, dim3(NT), 0, s, p);
+}
+
+void gemm_dw_tr_64(const GemmParams& p, hipStream_t s) {
+  dim3 grid(cdiv_h(p.M, TBM) * cdiv_h(p.N, 64) * p.splitk);
+  hipLaunchKernelGGL((dw_tr_kernel<64, AM_PLAIN>), grid, dim3(NT), 0, s, p);
+}

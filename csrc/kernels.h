@@ -34,6 +34,10 @@ void conv_fwd_pool(const GemmParams&, hipStream_t);
 void conv1_fwd_pool(const GemmParams&, hipStream_t);
 void conv_dx_gemm(const GemmParams&, hipStream_t);
 void conv_dw_gemm(const GemmParams&, hipStream_t);
+// dw_tr.hip — glds + ds_read_b64_tr_b16 k-major dW GEMMs (no staging scatter)
+void conv_dw_tr(const GemmParams&, hipStream_t);    // 5x5-SAME gather A
+void gemm_dw_tr_128(const GemmParams&, hipStream_t);  // plain k-major A, BN=128
+void gemm_dw_tr_64(const GemmParams&, hipStream_t);   // plain k-major A, BN=64
 
 // ops_misc.hip — host wrappers
 void launch_relu_drop_bwd(const unsigned short* dy, const unsigned short* y,
