@@ -33,9 +33,14 @@
 #define TBK 64
 #define TBM 128
 
-namespace {
-
 __device__ __align__(16) unsigned short g_zero_page[8];  // zero-init
+
+// non-template wrapper: the target builtin inside a function TEMPLATE makes
+// hipcc's host pass silently skip emitting the kernel's device stub
+// (deferred-diagnostic quirk); hoisting it into a plain DEV helper fixes it
+DEV void glds16(const ushort_t* src, ushort_t* dst) {
+  __builtin_amdgcn_global_load_lds(src, dst, 16, 0, 0);
+}
 
 // two transpose reads -> one MFMA short8 fragment (k-contiguous per lane)
 DEV short8 tr16_frag(unsigned a0, unsigned a1) {
@@ -133,7 +138,7 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
             src = p.A + (((size_t)n_ * p.CH + y) * p.CW + x) * p.Cin + aci;
         }
       }
-      __builtin_amdgcn_global_load_lds(src, &Aimg[buf][row0][0], 16, 0, 0);
+      glds16(src, &Aimg[buf][row0][0]);
     }
 #pragma unroll
     for (int q = 0; q < BQ; ++q) {
@@ -143,7 +148,7 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
       const ushort_t* src = (k < kend && bn_ok)
                                 ? p.B + (size_t)k * p.ldb + bnc
                                 : g_zero_page;
-      __builtin_amdgcn_global_load_lds(src, &Bimg[buf][row0][0], 16, 0, 0);
+      glds16(src, &Bimg[buf][row0][0]);
     }
   };
 
@@ -208,8 +213,6 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
     }
   }
 }
-
-}  // namespace
 
 static inline int cdiv_h(int a, int b) { return (a + b - 1) / b; }
 
