@@ -37,13 +37,14 @@ int pick_splitk(int mtiles, int ntiles, int ksteps) {
 
 // glds + tr16 dW staging (dw_tr.hip); DMNIST_DW_TR=0 reverts to the
 // scatter-staged gemm_tile path for A/B comparison
-bool dw_tr_enabled() {
+int dw_tr_level() {
   static int v = [] {
     const char* e = getenv("DMNIST_DW_TR");
     return e ? atoi(e) : 1;
   }();
-  return v != 0;
+  return v;
 }
+bool dw_tr_enabled() { return dw_tr_level() != 0; }
 
 // --------------------------------------------------------------------------
 torch::Tensor linear_act_fwd_impl(torch::Tensor x, torch::Tensor w,
@@ -274,7 +275,7 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
       // k-chain is short and the chip stays full
       p.splitk = std::min(cdiv(p.K, 64), 2048);
       conv1_dw_gemm(p, s);
-    } else if (conv_slab_supported(H, W, Cin, Cout) &&
+    } else if (conv_slab_supported(H, W, Cin, Cout) && dw_tr_level() < 2 &&
                (NB >= 2048 || getenv("DMNIST_DW_G"))) {
       // per-image-group slab dW: wins when the flush atomics amortize over
       // >=4 images/block; below that the implicit-GEMM form is faster
@@ -414,7 +415,7 @@ void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
   } else if (Cin == 1) {
     p.splitk = std::min(cdiv(p.K, 64), 2048);
     conv1_dw_gemm(p, s);
-  } else if (conv_slab_supported(H, W, Cin, Cout) &&
+  } else if (conv_slab_supported(H, W, Cin, Cout) && dw_tr_level() < 2 &&
              (NB >= 2048 || getenv("DMNIST_DW_G"))) {
     launch_conv_dw_slab(bf16_ptr(x), bf16_ptr(dact), dw_out.data_ptr<float>(),
                         NB, H, W, Cin, Cout, s);
