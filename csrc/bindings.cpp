@@ -275,8 +275,8 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
       // k-chain is short and the chip stays full
       p.splitk = std::min(cdiv(p.K, 64), 2048);
       conv1_dw_gemm(p, s);
-    } else if (conv_slab_supported(H, W, Cin, Cout) && dw_tr_level() < 2 &&
-               (NB >= 2048 || getenv("DMNIST_DW_G"))) {
+    } else if (conv_slab_supported(H, W, Cin, Cout) &&
+               ((dw_tr_level() == 0 && NB >= 2048) || getenv("DMNIST_DW_G"))) {
       // per-image-group slab dW: wins when the flush atomics amortize over
       // >=4 images/block; below that the implicit-GEMM form is faster
       launch_conv_dw_slab(bf16_ptr(x), bf16_ptr(dact), dw.data_ptr<float>(),
@@ -415,8 +415,8 @@ void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
   } else if (Cin == 1) {
     p.splitk = std::min(cdiv(p.K, 64), 2048);
     conv1_dw_gemm(p, s);
-  } else if (conv_slab_supported(H, W, Cin, Cout) && dw_tr_level() < 2 &&
-             (NB >= 2048 || getenv("DMNIST_DW_G"))) {
+  } else if (conv_slab_supported(H, W, Cin, Cout) &&
+             ((dw_tr_level() == 0 && NB >= 2048) || getenv("DMNIST_DW_G"))) {
     launch_conv_dw_slab(bf16_ptr(x), bf16_ptr(dact), dw_out.data_ptr<float>(),
                         NB, H, W, Cin, Cout, s);
   } else {
