@@ -283,7 +283,9 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
                           NB, H, W, Cin, Cout, s);
     } else {
       int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
-      p.splitk = std::min(cdiv(p.K, 64), std::max(1, 2048 / tiles));
+      int want = 2048;  // total blocks target; DMNIST_DW_BLOCKS sweeps it
+      if (const char* e = getenv("DMNIST_DW_BLOCKS")) want = atoi(e);
+      p.splitk = std::min(cdiv(p.K, 64), std::max(1, want / tiles));
       if (dw_tr_enabled() && Cin % 8 == 0 && Cout % 64 == 0)
         conv_dw_tr(p, s);
       else
@@ -421,7 +423,9 @@ void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
                         NB, H, W, Cin, Cout, s);
   } else {
     int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
-    p.splitk = std::min(cdiv(p.K, 64), std::max(1, 2048 / tiles));
+    int want = 2048;  // total blocks target; DMNIST_DW_BLOCKS sweeps it
+    if (const char* e = getenv("DMNIST_DW_BLOCKS")) want = atoi(e);
+    p.splitk = std::min(cdiv(p.K, 64), std::max(1, want / tiles));
     if (dw_tr_enabled() && Cin % 8 == 0 && Cout % 64 == 0)
       conv_dw_tr(p, s);
     else
