@@ -106,17 +106,43 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
   // ---- per-lane glds source decode (loop-invariant parts) ----------------
   // A: glds gi = wave*AQ+q stages rows gi*(512/TBM) + lane/ALPR at columns
   //    (lane%ALPR)*8..+7; LDS dest = &Aimg[buf][row0][0] + lane*16 (linear).
-  const int amc = m0 + (lane % ALPR) * 8;  // m column this lane stages
-  const bool am_ok = amc < p.M;
-  int akh = 0, akw = 0, aci = 0;
-  if (AMODE == AM_CONV5) {
-    int khkw = amc / p.Cin;
-    aci = amc % p.Cin;
-    akh = khkw / 5 - 2;
-    akw = khkw % 5 - 2;
+  //
+  // XOR bank swizzle: un-swizzled, every LDS row is a 256B/128B stride =
+  // a whole bank wrap, so the tr16 slot reads (4 rows x same columns) were
+  // ~8-way bank-conflicted (measured conf/idx = 0.85, profiles/).  The LDS
+  // DEST of glds is fixed (base + lane*16), so the swizzle is applied to
+  // the SOURCE: the lane staging physical 16B-chunk pc of tile-row rt
+  // fetches LOGICAL chunk pc ^ xs(rt), with xs(rt) = (rt ^ (rt>>4)) & 15
+  // for A (16 chunks/row) and (rt ^ (rt>>3)) & 7 for B.  rt depends only
+  // on (wave, q, lane) — never on the K-tile — so decodes stay loop-
+  // invariant.  The tr16 addresses apply the same xs per slot row.
+  int amc[AQ];     // logical m column this lane stages, per glds instr
+  bool am_ok[AQ];
+  int akh[AQ], akw[AQ], aci[AQ];
+#pragma unroll
+  for (int q = 0; q < AQ; ++q) {
+    int rt = (wave * AQ + q) * (512 / TBM) + lane / ALPR;
+    int xs = (rt ^ (rt >> 4)) & (ALPR - 1);
+    amc[q] = m0 + ((lane % ALPR) ^ xs) * 8;
+    am_ok[q] = amc[q] < p.M;
+    if (AMODE == AM_CONV5) {
+      int khkw = amc[q] / p.Cin;
+      aci[q] = amc[q] % p.Cin;
+      akh[q] = khkw / 5 - 2;
+      akw[q] = khkw % 5 - 2;
+    } else {
+      akh[q] = akw[q] = aci[q] = 0;
+    }
   }
-  const int bnc = n0 + (lane % BLPR) * 8;  // n column this lane stages
-  const bool bn_ok = bnc < p.N;
+  int bnc[BQ];
+  bool bn_ok[BQ];
+#pragma unroll
+  for (int q = 0; q < BQ; ++q) {
+    int rt = (wave * BQ + q) * (512 / BN) + lane / BLPR;
+    int xs = (rt ^ (rt >> 3)) & (BLPR - 1);
+    bnc[q] = n0 + ((lane % BLPR) ^ xs) * 8;
+    bn_ok[q] = bnc[q] < p.N;
+  }
 
   const auto issue = [&](int buf, int kt) {
 #pragma unroll
@@ -125,17 +151,17 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
       int row0 = gi * (512 / TBM);
       int k = kt + row0 + lane / ALPR;
       const ushort_t* src = g_zero_page;
-      if (k < kend && am_ok) {
+      if (k < kend && am_ok[q]) {
         if (AMODE == AM_PLAIN) {
-          src = p.A + (size_t)k * p.lda + amc;
+          src = p.A + (size_t)k * p.lda + amc[q];
         } else {
           int w_ = k % p.CW;
           int t2 = k / p.CW;
           int h_ = t2 % p.CH;
           int n_ = t2 / p.CH;
-          int y = h_ + akh, x = w_ + akw;
+          int y = h_ + akh[q], x = w_ + akw[q];
           if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
-            src = p.A + (((size_t)n_ * p.CH + y) * p.CW + x) * p.Cin + aci;
+            src = p.A + (((size_t)n_ * p.CH + y) * p.CW + x) * p.Cin + aci[q];
         }
       }
       glds16(src, &Aimg[buf][row0][0]);
@@ -145,8 +171,8 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
       int gi = wave * BQ + q;
       int row0 = gi * (512 / BN);
       int k = kt + row0 + lane / BLPR;
-      const ushort_t* src = (k < kend && bn_ok)
-                                ? p.B + (size_t)k * p.ldb + bnc
+      const ushort_t* src = (k < kend && bn_ok[q])
+                                ? p.B + (size_t)k * p.ldb + bnc[q]
                                 : g_zero_page;
       glds16(src, &Bimg[buf][row0][0]);
     }
@@ -172,15 +198,26 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
     for (int kh2 = 0; kh2 < 2; ++kh2) {
       int krow = kh2 * 32 + kgrp + slot_r;
       short8 af[MI], bf[NI];
+      // slot address with the matching XOR chunk swizzle (see staging note)
+      const auto aswz = [&](int r, int cb) {
+        int xs = (r ^ (r >> 4)) & (ALPR - 1);
+        return abase + (unsigned)(r * (TBM * 2) + (((cb >> 4) ^ xs) << 4) +
+                                  (cb & 15));
+      };
+      const auto bswz = [&](int r, int cb) {
+        int xs = (r ^ (r >> 3)) & (BLPR - 1);
+        return bbase + (unsigned)(r * (BN * 2) + (((cb >> 4) ^ xs) << 4) +
+                                  (cb & 15));
+      };
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi) {
-        unsigned a0 = abase + (unsigned)((krow * TBM + wr * WM + mi * 16 + slot_c) * 2);
-        af[mi] = tr16_frag(a0, a0 + 4 * TBM * 2);
+        int cb = (wr * WM + mi * 16 + slot_c) * 2;
+        af[mi] = tr16_frag(aswz(krow, cb), aswz(krow + 4, cb));
       }
 #pragma unroll
       for (int ni = 0; ni < NI; ++ni) {
-        unsigned b0 = bbase + (unsigned)((krow * BN + wc * WN + ni * 16 + slot_c) * 2);
-        bf[ni] = tr16_frag(b0, b0 + 4 * BN * 2);
+        int cb = (wc * WN + ni * 16 + slot_c) * 2;
+        bf[ni] = tr16_frag(bswz(krow, cb), bswz(krow + 4, cb));
       }
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
