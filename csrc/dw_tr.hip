@@ -42,18 +42,22 @@ DEV void glds16(const ushort_t* src, ushort_t* dst) {
   __builtin_amdgcn_global_load_lds(src, dst, 16, 0, 0);
 }
 
-// two transpose reads -> one MFMA short8 fragment (k-contiguous per lane)
-DEV short8 tr16_frag(unsigned a0, unsigned a1) {
-  uint2 lo, hi;
-  asm volatile(
-      "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
-      "ds_read_b64_tr_b16 %1, %3 offset:0\n\t"
-      "s_waitcnt lgkmcnt(0)"
-      : "=&v"(lo), "=&v"(hi)  // early-clobber: written before %3 is read
-      : "v"(a0), "v"(a1)
-      : "memory");
+// un-waited transpose read: issue-only, so a whole k-half's reads overlap;
+// drain with pack_wait before the MFMAs consume them
+DEV uint2 tr16_issue(unsigned a) {
+  uint2 d;
+  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+               : "=&v"(d) : "v"(a) : "memory");
+  return d;
+}
+
+// pack two 4-element halves into an MFMA fragment; the wait is tied to the
+// packed value so no MFMA reading it can be scheduled before the drain
+// (repeat waits after the first are ~free: the counter is already 0)
+DEV short8 pack_wait(uint2 lo, uint2 hi) {
   union { unsigned u[4]; short8 s; } r;
   r.u[0] = lo.x; r.u[1] = lo.y; r.u[2] = hi.x; r.u[3] = hi.y;
+  asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(r.s));
   return r.s;
 }
 
@@ -144,6 +148,26 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
     bn_ok[q] = bnc[q] < p.N;
   }
 
+  // incremental pixel decode (AM_CONV5): the k this (lane, q) stages
+  // advances by exactly TBK per issue() call, so a mixed-radix add with
+  // carries replaces the per-tile runtime div/mod chain
+  int aw[AQ], ah[AQ], an[AQ];
+  int dw64 = 0, dh64 = 0, dn64 = 0;
+  if (AMODE == AM_CONV5) {
+    dw64 = TBK % p.CW;
+    int t64 = TBK / p.CW;
+    dh64 = t64 % p.CH;
+    dn64 = t64 / p.CH;
+#pragma unroll
+    for (int q = 0; q < AQ; ++q) {
+      int k0 = kbeg + (wave * AQ + q) * (512 / TBM) + lane / ALPR;
+      aw[q] = k0 % p.CW;
+      int t2 = k0 / p.CW;
+      ah[q] = t2 % p.CH;
+      an[q] = t2 / p.CH;
+    }
+  }
+
   const auto issue = [&](int buf, int kt) {
 #pragma unroll
     for (int q = 0; q < AQ; ++q) {
@@ -155,16 +179,19 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
         if (AMODE == AM_PLAIN) {
           src = p.A + (size_t)k * p.lda + amc[q];
         } else {
-          int w_ = k % p.CW;
-          int t2 = k / p.CW;
-          int h_ = t2 % p.CH;
-          int n_ = t2 / p.CH;
-          int y = h_ + akh[q], x = w_ + akw[q];
+          int y = ah[q] + akh[q], x = aw[q] + akw[q];
           if (y >= 0 && y < p.CH && x >= 0 && x < p.CW)
-            src = p.A + (((size_t)n_ * p.CH + y) * p.CW + x) * p.Cin + aci[q];
+            src = p.A + (((size_t)an[q] * p.CH + y) * p.CW + x) * p.Cin + aci[q];
         }
       }
       glds16(src, &Aimg[buf][row0][0]);
+      if (AMODE == AM_CONV5) {  // advance to the pixel this lane stages next
+        aw[q] += dw64;
+        if (aw[q] >= p.CW) { aw[q] -= p.CW; ah[q] += 1; }
+        ah[q] += dh64;
+        if (ah[q] >= p.CH) { ah[q] -= p.CH; an[q] += 1; }
+        an[q] += dn64;
+      }
     }
 #pragma unroll
     for (int q = 0; q < BQ; ++q) {
@@ -209,16 +236,23 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
         return bbase + (unsigned)(r * (BN * 2) + (((cb >> 4) ^ xs) << 4) +
                                   (cb & 15));
       };
+      uint2 ar[MI][2], br[NI][2];
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi) {
         int cb = (wr * WM + mi * 16 + slot_c) * 2;
-        af[mi] = tr16_frag(aswz(krow, cb), aswz(krow + 4, cb));
+        ar[mi][0] = tr16_issue(aswz(krow, cb));
+        ar[mi][1] = tr16_issue(aswz(krow + 4, cb));
       }
 #pragma unroll
       for (int ni = 0; ni < NI; ++ni) {
         int cb = (wc * WN + ni * 16 + slot_c) * 2;
-        bf[ni] = tr16_frag(bswz(krow, cb), bswz(krow + 4, cb));
+        br[ni][0] = tr16_issue(bswz(krow, cb));
+        br[ni][1] = tr16_issue(bswz(krow + 4, cb));
       }
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi) af[mi] = pack_wait(ar[mi][0], ar[mi][1]);
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni) bf[ni] = pack_wait(br[ni][0], br[ni][1]);
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
