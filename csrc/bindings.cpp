@@ -267,7 +267,11 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
     p.M = 25 * Cin; p.N = Cout; p.K = NB * H * W;
     p.ldb = Cout; p.ldc = Cout;
     p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
-    if (Cin == 1 && conv1_slab_supported(H, W, Cin, Cout)) {
+    if (Cin == 1 && H <= 28 && W <= 28 && Cout <= 32 &&
+        !getenv("DMNIST_DW1_SLAB")) {
+      launch_conv1_dw_direct(bf16_ptr(x), bf16_ptr(dact),
+                             dw.data_ptr<float>(), NB, H, W, Cout, s);
+    } else if (Cin == 1 && conv1_slab_supported(H, W, Cin, Cout)) {
       launch_conv1_dw_slab(bf16_ptr(x), bf16_ptr(dact), dw.data_ptr<float>(),
                            NB, H, W, Cout, s);
     } else if (Cin == 1) {
@@ -411,7 +415,11 @@ void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
   p.M = 25 * Cin; p.N = Cout; p.K = NB * H * W;
   p.ldb = Cout; p.ldc = Cout;
   p.CB = NB; p.CH = H; p.CW = W; p.Cin = Cin; p.Cout = Cout;
-  if (Cin == 1 && conv1_slab_supported(H, W, Cin, Cout)) {
+  if (Cin == 1 && H <= 28 && W <= 28 && Cout <= 32 &&
+      !getenv("DMNIST_DW1_SLAB")) {
+    launch_conv1_dw_direct(bf16_ptr(x), bf16_ptr(dact),
+                           dw_out.data_ptr<float>(), NB, H, W, Cout, s);
+  } else if (Cin == 1 && conv1_slab_supported(H, W, Cin, Cout)) {
     launch_conv1_dw_slab(bf16_ptr(x), bf16_ptr(dact),
                          dw_out.data_ptr<float>(), NB, H, W, Cout, s);
   } else if (Cin == 1) {

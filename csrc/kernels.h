@@ -105,3 +105,7 @@ bool conv1_slab_supported(int H, int W, int Cin, int Cout);
 void launch_conv1_dw_slab(const unsigned short* x, const unsigned short* dact,
                           float* dw, int NB, int H, int W, int Cout,
                           hipStream_t);
+// ops_misc.hip — direct-VALU conv1 dW (Cin==1, H,W<=28, Cout<=32)
+void launch_conv1_dw_direct(const unsigned short* x,
+                            const unsigned short* dact, float* dw, int NB,
+                            int H, int W, int Cout, hipStream_t s);

@@ -495,3 +495,93 @@ void launch_conv1_direct_fwd(const unsigned short* x, const unsigned short* w,
   hipLaunchKernelGGL(conv1_direct_fwd_kernel, dim3(NB), dim3(256), 0, s, x,
                      w, bias, y, amax, NB, H, W, Cout, (int)mpool);
 }
+
+// ---------------------------------------------------------------------------
+// conv1 backward dW, direct VALU form (Cin == 1, K = 25 taps).
+// dw[kh][kw][co] = sum_{n,h,w} x[n, h+kh-2, w+kw-2] * dact[n,h,w,co].
+// MFMA loses at K=25 (M-util 25/32 and the A gather is per-element); here
+// each lane owns one co and 25 fp32 accumulators; the x image is staged once
+// per block as the same padded fp32 [32][32] slab the conv1 forward uses,
+// and the 5x5 x-window walks each row with a 5-phase cyclic column buffer —
+// 5 broadcast LDS reads + 25 v_fmac per pixel.  A block accumulates G
+// images into registers and flushes ONCE via an LDS-reduced 800-value
+// atomicAdd (the per-image flush is what made the old slab path atomics-
+// bound below NB=2048).
+__global__ __launch_bounds__(256)
+void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
+                            float* dw, int NB, int H, int W, int Cout,
+                            int G) {
+  __shared__ float slab[32][32];
+  __shared__ float red[25][32];
+  const int tid = threadIdx.x;
+  const int co = tid & 31;
+  const int hw = tid >> 5;  // half-wave 0..7: rows hw, hw+8, ...
+  float acc[25];
+#pragma unroll
+  for (int t = 0; t < 25; ++t) acc[t] = 0.f;
+
+  const int img0 = blockIdx.x * G;
+  for (int g = 0; g < G; ++g) {
+    const int img = img0 + g;
+    if (img >= NB) break;  // uniform across the block
+    const ushort_t* xi = x + (size_t)img * H * W;
+    for (int i = tid; i < 32 * 32; i += 256) {
+      int xx = i % 32, yy = i / 32;
+      int sy = yy - 2, sx = xx - 2;
+      float v = 0.f;
+      if (sy >= 0 && sy < H && sx >= 0 && sx < W)
+        v = bf2f(xi[(size_t)sy * W + sx]);
+      slab[yy][xx] = v;
+    }
+    __syncthreads();
+    const ushort_t* di = dact + (size_t)img * H * W * Cout + co;
+    if (co < Cout) {
+      for (int r = hw; r < H; r += 8) {
+        float win[5][5];  // win[colslot][kh]; colslot = (x_col) mod 5
+#pragma unroll
+        for (int pc = 0; pc < 4; ++pc)  // preload x cols -2..1 (slab 0..3)
+#pragma unroll
+          for (int kr = 0; kr < 5; ++kr)
+            win[(pc + 3) % 5][kr] = slab[r + kr][pc];
+        const ushort_t* drow = di + (size_t)r * W * Cout;
+        for (int cc = 0; cc < W; cc += 5) {
+#pragma unroll
+          for (int p = 0; p < 5; ++p) {
+            int c = cc + p;
+            if (c >= W) break;
+#pragma unroll
+            for (int kr = 0; kr < 5; ++kr)  // new col c+2 -> slot (c+2)%5
+              win[(p + 2) % 5][kr] = slab[r + kr][c + 4];
+            float gv = bf2f(drow[(size_t)c * Cout]);
+#pragma unroll
+            for (int kh = 0; kh < 5; ++kh)
+#pragma unroll
+              for (int kw = 0; kw < 5; ++kw)
+                acc[kh * 5 + kw] += win[(p + kw + 3) % 5][kh] * gv;
+          }
+        }
+      }
+    }
+    __syncthreads();  // slab is re-staged next image
+  }
+
+  // block reduction: 8 half-waves -> LDS, then one global flush
+  for (int i = tid; i < 25 * 32; i += 256) red[i / 32][i % 32] = 0.f;
+  __syncthreads();
+  if (co < Cout)
+#pragma unroll
+    for (int t = 0; t < 25; ++t) atomicAdd(&red[t][co], acc[t]);
+  __syncthreads();
+  for (int i = tid; i < 25 * Cout; i += 256)
+    atomicAdd(&dw[(size_t)(i / Cout) * Cout + (i % Cout)],
+              red[i / Cout][i % Cout]);
+}
+
+void launch_conv1_dw_direct(const unsigned short* x,
+                            const unsigned short* dact, float* dw, int NB,
+                            int H, int W, int Cout, hipStream_t s) {
+  int G = NB >= 8192 ? 8 : (NB >= 2048 ? NB / 1024 : 1);
+  int blocks = (NB + G - 1) / G;
+  hipLaunchKernelGGL(conv1_dw_direct_kernel, dim3(blocks), dim3(256), 0, s,
+                     x, dact, dw, NB, H, W, Cout, G);
+}
