@@ -543,9 +543,6 @@ void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
 #pragma unroll
           for (int kr = 0; kr < 5; ++kr)
             win[(pc + 3) % 5][kr] = slab[r + kr][pc];
-        float tmp[5];  // col c+2 for the NEXT pixel, read one pixel ahead
-#pragma unroll
-        for (int kr = 0; kr < 5; ++kr) tmp[kr] = slab[r + kr][4];
         const ushort_t* drow = di + (size_t)r * W * Cout;
         // double-buffered 5-wide dact prefetch: the serial load->25-FMA
         // chain was latency-bound (one load in flight per wave); keeping
@@ -566,24 +563,15 @@ void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
           for (int p = 0; p < 5; ++p) {
             int c = cc + p;
             if (c >= W) break;
-            // issue the NEXT pixel's new-column reads, then commit the
-            // PREVIOUS issue (tmp = col c+2) into its slot — slot (p+2)%5
-            // holds col c-3, outside the current window, so the commit is
-            // safe before the FMAs and the read wait lands after them
-            float tnew[5];
 #pragma unroll
-            for (int kr = 0; kr < 5; ++kr)
-              tnew[kr] = (c + 5 < 32) ? slab[r + kr][c + 5] : 0.f;
-#pragma unroll
-            for (int kr = 0; kr < 5; ++kr) win[(p + 2) % 5][kr] = tmp[kr];
+            for (int kr = 0; kr < 5; ++kr)  // new col c+2 -> slot (c+2)%5
+              win[(p + 2) % 5][kr] = slab[r + kr][c + 4];
             float gv = bf2f(gcur[p]);
 #pragma unroll
             for (int kh = 0; kh < 5; ++kh)
 #pragma unroll
               for (int kw = 0; kw < 5; ++kw)
                 acc[kh * 5 + kw] += win[(p + kw + 3) % 5][kh] * gv;
-#pragma unroll
-            for (int kr = 0; kr < 5; ++kr) tmp[kr] = tnew[kr];
           }
 #pragma unroll
           for (int p = 0; p < 5; ++p) gcur[p] = gnxt[p];
