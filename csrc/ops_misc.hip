@@ -537,49 +537,44 @@ void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
     const ushort_t* di = dact + (size_t)img * H * W * Cout + co;
     if (co < Cout) {
       for (int r = hw; r < H; r += 8) {
-        // PIXEL-PAIR walk (W even): window = 6 columns (slot = x_col mod
-        // 6), 2 new columns + 2 dact values per step -> 50 FMAs per 10 LDS
-        // reads, two independent FMA chains.  No column bounds checks:
-        // c <= W-2 means the furthest read is slab col (W-2)+3+2 = W+3 <=
-        // 31.  dact loads are prefetched one 3-pair group ahead (raw u16
-        // until use) so the serial load->FMA chain is never latency-bound.
-        float win[6][5];  // win[colslot][kh]
+        float win[5][5];  // win[colslot][kh]; colslot = (x_col) mod 5
 #pragma unroll
         for (int pc = 0; pc < 4; ++pc)  // preload x cols -2..1 (slab 0..3)
 #pragma unroll
           for (int kr = 0; kr < 5; ++kr)
-            win[(pc + 4) % 6][kr] = slab[r + kr][pc];
+            win[(pc + 3) % 5][kr] = slab[r + kr][pc];
         const ushort_t* drow = di + (size_t)r * W * Cout;
-        ushort_t gcur[6], gnxt[6];
+        // double-buffered 5-wide dact prefetch: the serial load->25-FMA
+        // chain was latency-bound (one load in flight per wave); keeping
+        // the NEXT column-group's 5 loads in flight under the current
+        // group's 125 FMAs hides it.  Raw u16 until use so the load wait
+        // lands a full group later.
+        ushort_t gcur[5], gnxt[5];
 #pragma unroll
-        for (int p = 0; p < 6; ++p)
+        for (int p = 0; p < 5; ++p)
           gcur[p] = (p < W) ? drow[(size_t)p * Cout] : (ushort_t)0;
-        for (int cc = 0; cc < W; cc += 6) {  // 3 pixel-pairs per group
+        for (int cc = 0; cc < W; cc += 5) {
 #pragma unroll
-          for (int p = 0; p < 6; ++p) {
-            int cn = cc + 6 + p;
+          for (int p = 0; p < 5; ++p) {
+            int cn = cc + 5 + p;
             gnxt[p] = (cn < W) ? drow[(size_t)cn * Cout] : (ushort_t)0;
           }
 #pragma unroll
-          for (int p2 = 0; p2 < 3; ++p2) {
-            int c = cc + p2 * 2;  // even pixel of the pair
+          for (int p = 0; p < 5; ++p) {
+            int c = cc + p;
             if (c >= W) break;
 #pragma unroll
-            for (int kr = 0; kr < 5; ++kr) {  // new cols c+2, c+3
-              win[(p2 * 2 + 2) % 6][kr] = slab[r + kr][c + 4];
-              win[(p2 * 2 + 3) % 6][kr] = slab[r + kr][c + 5];
-            }
-            float ge = bf2f(gcur[p2 * 2]);
-            float go = bf2f(gcur[p2 * 2 + 1]);
+            for (int kr = 0; kr < 5; ++kr)  // new col c+2 -> slot (c+2)%5
+              win[(p + 2) % 5][kr] = slab[r + kr][c + 4];
+            float gv = bf2f(gcur[p]);
 #pragma unroll
             for (int kh = 0; kh < 5; ++kh)
 #pragma unroll
               for (int kw = 0; kw < 5; ++kw)
-                acc[kh * 5 + kw] += win[(p2 * 2 + kw + 4) % 6][kh] * ge +
-                                    win[(p2 * 2 + kw + 5) % 6][kh] * go;
+                acc[kh * 5 + kw] += win[(p + kw + 3) % 5][kh] * gv;
           }
 #pragma unroll
-          for (int p = 0; p < 6; ++p) gcur[p] = gnxt[p];
+          for (int p = 0; p < 5; ++p) gcur[p] = gnxt[p];
         }
       }
     }
