@@ -69,3 +69,33 @@ DEV float philox_uniform(uint64_t seed, uint64_t offset, uint64_t idx) {
 }
 
 DEV int cdiv(int a, int b) { return (a + b - 1) / b; }
+
+// ---- glds + ds_read_b64_tr_b16 staging primitives (gfx950) ----------------
+// Semantics pinned empirically: profiles/r01_tr16_probe.md.  The glds LDS
+// destination is wave-uniform base + lane*16 (hardware); tr16 gives each
+// lane of a 16-lane group img[kq + j][n0 + (l&15)] (j = 0..3) from a
+// K-MAJOR LDS image when lane g of the group addresses
+// &img[kq + (g>>2)][n0 + 4*(g&3)].  Keep these OUT of function templates:
+// a target builtin inside a template makes hipcc's host pass silently skip
+// emitting the kernel's device stub.
+DEV void glds16(const ushort_t* src, ushort_t* dst) {
+  __builtin_amdgcn_global_load_lds(src, dst, 16, 0, 0);
+}
+
+// un-waited transpose read: issue-only, drain with pack_wait before use
+DEV uint2 tr16_issue(unsigned a) {
+  uint2 d;
+  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+               : "=&v"(d) : "v"(a) : "memory");
+  return d;
+}
+
+// pack two 4-element halves into an MFMA fragment; the wait is tied to the
+// packed value so no MFMA reading it can be scheduled before the drain
+// (repeat waits after the first are ~free: the counter is already 0)
+DEV short8 pack_wait(uint2 lo, uint2 hi) {
+  union { unsigned u[4]; short8 s; } r;
+  r.u[0] = lo.x; r.u[1] = lo.y; r.u[2] = hi.x; r.u[3] = hi.y;
+  asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(r.s));
+  return r.s;
+}

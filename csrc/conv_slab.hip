@@ -22,6 +22,8 @@
 
 typedef ushort_t u16;
 
+__device__ __align__(16) static unsigned short g_zero_page_slab[8];  // glds OOB redirect
+
 // read elements [s, s+8) of an LDS row via two ALIGNED ds_read_b128 + a
 // constant shuffle: s is wave-uniform per call site (the 8-case switch is a
 // uniform scalar branch), so this replaces 8 scalar ds_read_u16 per
@@ -66,7 +68,10 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
   // conflict, 80B = conflict-free ((a/4)%64 steps of 20)
   constexpr int PST = CIN + 8;  // 80B pixel stride: 16B-aligned b128, conflict-reduced
   __shared__ __align__(16) u16 slab[HP * WP * PST];
-  __shared__ __align__(16) u16 Bs[BN][LDK];
+  // weight tile: K-MAJOR + double-buffered, staged by glds and read with
+  // ds_read_b64_tr_b16 (same recipe as dw_tr.hip — no scatter-transpose,
+  // the whole next tile is in flight under the current tile's MFMAs)
+  __shared__ __align__(16) u16 Bs[2][BK][BN];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -74,6 +79,26 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
   const int wr = wave >> 1, wc = wave & 1;
   const int img = blockIdx.x;
   const u16* xi = x + (size_t)img * H * W * CIN;
+
+  constexpr int BLPR = BN / 8;   // lanes per Bs row (8)
+  constexpr int BQ = BN / 32;    // glds per wave (2)
+  int bnc[BQ];                   // logical n column this lane stages
+#pragma unroll
+  for (int q = 0; q < BQ; ++q) {
+    int rt = (wave * BQ + q) * (512 / BN) + lane / BLPR;
+    int xs = (rt ^ (rt >> 3)) & (BLPR - 1);  // XOR bank swizzle (see dw_tr)
+    bnc[q] = ((lane % BLPR) ^ xs) * 8;
+  }
+  const auto issueB = [&](int buf, int kt) {
+#pragma unroll
+    for (int q = 0; q < BQ; ++q) {
+      int row0 = (wave * BQ + q) * (512 / BN);
+      int k = kt + row0 + lane / BLPR;
+      const u16* src =
+          (k < K) ? w + (size_t)k * COUT + bnc[q] : g_zero_page_slab;
+      glds16(src, &Bs[buf][row0][0]);
+    }
+  };
 
   // stage padded slab (zero halo): chunks of 8 ci
   for (int c = tid; c < HP * WP * (CIN / 8); c += NTHREADS) {
@@ -108,21 +133,18 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
     for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int kq = (lane >> 4) * 8;
+  const int slot_r = (lane & 15) >> 2;  // tr16 slot addressing (see dw_tr)
+  const int slot_c = (lane & 3) * 4;
+  const int kgrp = (lane >> 4) * 8;
+  issueB(0, 0);    // in flight under the slab-staging barrier's drain
   __syncthreads();
 
-  for (int kt = 0; kt < K; kt += BK) {
-    // stage B tile [BK][BN] -> Bs[n][kk] (scatter-transpose, w is k-major)
-    for (int c = tid; c < BK * (BN / 8); c += NTHREADS) {
-      int kk = c / (BN / 8);
-      int j0 = (c % (BN / 8)) * 8;
-      int k = kt + kk;
-      short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
-      if (k < K)
-        v = *reinterpret_cast<const short8*>(w + (size_t)k * COUT + j0);
-#pragma unroll
-      for (int e = 0; e < 8; ++e) Bs[j0 + e][kk] = v[e];
-    }
-    __syncthreads();
+  constexpr int NT_K = (K + BK - 1) / BK;
+  int cur = 0;
+  for (int t = 0; t < NT_K; ++t) {
+    int kt = t * BK;
+    if (t + 1 < NT_K) issueB(cur ^ 1, kt + BK);  // hides under MFMAs
+    unsigned bbase = (unsigned)(uintptr_t)&Bs[cur][0][0];
 #pragma unroll
     for (int kh2 = 0; kh2 < 2; ++kh2) {
       int kbase = kt + kh2 * 32 + kq;       // k = khkw*CIN + ci
@@ -141,10 +163,21 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
       for (int mi = 0; mi < MI; ++mi)
         af[mi] = *reinterpret_cast<const short8*>(
             &slab[arow_off[mi] + shift]);
+      uint2 br[NI][2];
 #pragma unroll
-      for (int ni = 0; ni < NI; ++ni)
-        bf[ni] = *reinterpret_cast<const short8*>(
-            &Bs[wc * WN + ni * 16 + (lane & 15)][kh2 * 32 + kq]);
+      for (int ni = 0; ni < NI; ++ni) {
+        int cb = (wc * WN + ni * 16 + slot_c) * 2;
+        int r0 = kh2 * 32 + kgrp + slot_r;
+        int xs0 = (r0 ^ (r0 >> 3)) & (BLPR - 1);
+        int r1 = r0 + 4;
+        int xs1 = (r1 ^ (r1 >> 3)) & (BLPR - 1);
+        br[ni][0] = tr16_issue(bbase + (unsigned)(r0 * (BN * 2) +
+                       (((cb >> 4) ^ xs0) << 4) + (cb & 15)));
+        br[ni][1] = tr16_issue(bbase + (unsigned)(r1 * (BN * 2) +
+                       (((cb >> 4) ^ xs1) << 4) + (cb & 15)));
+      }
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni) bf[ni] = pack_wait(br[ni][0], br[ni][1]);
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
@@ -152,7 +185,8 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
-    __syncthreads();
+    __syncthreads();  // drains the in-flight glds (2-buffer overlap)
+    cur ^= 1;
   }
 
   // pool epilogue (per-lane 4-register max, frow multiple of 4)

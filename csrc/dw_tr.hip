@@ -35,32 +35,6 @@
 
 __device__ __align__(16) unsigned short g_zero_page[8];  // zero-init
 
-// non-template wrapper: the target builtin inside a function TEMPLATE makes
-// hipcc's host pass silently skip emitting the kernel's device stub
-// (deferred-diagnostic quirk); hoisting it into a plain DEV helper fixes it
-DEV void glds16(const ushort_t* src, ushort_t* dst) {
-  __builtin_amdgcn_global_load_lds(src, dst, 16, 0, 0);
-}
-
-// un-waited transpose read: issue-only, so a whole k-half's reads overlap;
-// drain with pack_wait before the MFMAs consume them
-DEV uint2 tr16_issue(unsigned a) {
-  uint2 d;
-  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
-               : "=&v"(d) : "v"(a) : "memory");
-  return d;
-}
-
-// pack two 4-element halves into an MFMA fragment; the wait is tied to the
-// packed value so no MFMA reading it can be scheduled before the drain
-// (repeat waits after the first are ~free: the counter is already 0)
-DEV short8 pack_wait(uint2 lo, uint2 hi) {
-  union { unsigned u[4]; short8 s; } r;
-  r.u[0] = lo.x; r.u[1] = lo.y; r.u[2] = hi.x; r.u[3] = hi.y;
-  asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(r.s));
-  return r.s;
-}
-
 enum { AM_PLAIN = 0, AM_CONV5 = 1 };
 
 template <int BN, int AMODE>
