@@ -235,7 +235,13 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
   constexpr int K = 25 * COUT;
   constexpr int PST = COUT + 8;  // 144B pixel stride: conflict-free b128
   __shared__ __align__(16) u16 slab[HP * WP * PST];
-  __shared__ __align__(16) u16 Bs[BN][LDK];
+  // weight tile: n-major rows of BK (w rows are already k-contiguous per
+  // ci), double-buffered and staged by glds — one barrier per K-step, the
+  // next tile in flight under the MFMAs.  The row stride is an unpadded
+  // 128B, so the 16B k-chunks are XOR-permuted per row (staging source and
+  // fragment read agree on xs(row)) to keep the b128 fragment reads off a
+  // single bank pair.
+  __shared__ __align__(16) u16 Bs[2][BN][BK];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -243,6 +249,20 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
   const int wr = wave >> 1, wc = wave & 1;
   const int img = blockIdx.x;
   const u16* di = dact + (size_t)img * H * W * COUT;
+
+  // glds staging decode: one 1KB glds per wave covers rows wave*8..+7
+  const int brow = wave * 8 + lane / 8;            // ci row this lane stages
+  const int bxs = (brow ^ (brow >> 3)) & 7;
+  const int blc = (lane % 8) ^ bxs;                // logical k-chunk
+  const auto issueB = [&](int buf, int kt) {
+    int k = kt + blc * 8;
+    const u16* src = g_zero_page_slab;
+    if (k < K) {
+      int khkw = k / COUT, co = k % COUT;
+      src = w + ((size_t)khkw * CIN + brow) * COUT + co;
+    }
+    glds16(src, &Bs[buf][wave * 8][0]);
+  };
 
   for (int c = tid; c < HP * WP * (COUT / 8); c += NTHREADS) {
     int co = (c % (COUT / 8)) * 8;
@@ -271,20 +291,15 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
     for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int kq = (lane >> 4) * 8;
+  issueB(0, 0);    // drains at the slab-staging barrier
   __syncthreads();
 
-  for (int kt = 0; kt < K; kt += BK) {
-    // Bs[ci][kk] = w[((khkw)*CIN + ci)*COUT + co(k)] — rows16 from w slices
-    for (int c = tid; c < BN * (BK / 8); c += NTHREADS) {
-      int i = c / (BK / 8);
-      int kc = (c % (BK / 8)) * 8;
-      int k = kt + kc;
-      int khkw = k / COUT, co = k % COUT;
-      short8 v = *reinterpret_cast<const short8*>(
-          w + ((size_t)khkw * CIN + i) * COUT + co);
-      *reinterpret_cast<short8*>(&Bs[i][kc]) = v;
-    }
-    __syncthreads();
+  constexpr int NT_K = (K + BK - 1) / BK;
+  int cur = 0;
+  for (int t = 0; t < NT_K; ++t) {
+    int kt = t * BK;
+    if (t + 1 < NT_K) issueB(cur ^ 1, kt + BK);  // hides under MFMAs
+    const u16* bp = &Bs[cur][0][0];
 #pragma unroll
     for (int kh2 = 0; kh2 < 2; ++kh2) {
       int kbase = kt + kh2 * 32 + kq;       // k = khkw*COUT + co
@@ -298,9 +313,13 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
         af[mi] = *reinterpret_cast<const short8*>(
             &slab[arow_off[mi] + shift]);
 #pragma unroll
-      for (int ni = 0; ni < NI; ++ni)
-        bf[ni] = *reinterpret_cast<const short8*>(
-            &Bs[wc * WN + ni * 16 + (lane & 15)][kh2 * 32 + kq]);
+      for (int ni = 0; ni < NI; ++ni) {
+        int n = wc * WN + ni * 16 + (lane & 15);
+        int xs = (n ^ (n >> 3)) & 7;
+        int chunk = (kh2 * 4 + (lane >> 4)) ^ xs;
+        bf[ni] = *reinterpret_cast<const short8*>(bp + (size_t)n * BK +
+                                                  chunk * 8);
+      }
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
@@ -308,7 +327,8 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
-    __syncthreads();
+    __syncthreads();  // drains the in-flight glds
+    cur ^= 1;
   }
 
   const int frow = (lane >> 4) * 4;
