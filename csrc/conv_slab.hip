@@ -349,6 +349,119 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
   }
 }
 
+// High-occupancy variant of conv dX (the pre-glds staging: padded
+// n-major Bs, 2 barriers/K-step, 3 blocks/CU).  The glds-pipelined
+// kernel above is faster at small NB but its extra LDS buffer drops
+// occupancy to 2 blocks/CU, which loses at NB >= 2048 (measured
+// 371 vs 327 us at B=8192); the launcher picks by NB.
+template <int H, int W, int CIN, int COUT>
+__global__ __launch_bounds__(NTHREADS)
+void conv_dx_slab_hi_occ_kernel(const u16* __restrict__ dact,
+                         const u16* __restrict__ w,  // [25*CIN][COUT]
+                         u16* __restrict__ dx, int NB) {
+  constexpr int HP = H + 4, WP = W + 4;
+  constexpr int M = H * W;
+  constexpr int BM = 224;
+  constexpr int BN = CIN;              // 32
+  constexpr int WM = 112, WN = BN / 2; // 16
+  constexpr int MI = WM / 16, NI = WN / 16;  // 7, 1
+  constexpr int K = 25 * COUT;
+  constexpr int PST = COUT + 8;  // 144B pixel stride: conflict-free b128
+  __shared__ __align__(16) u16 slab[HP * WP * PST];
+  __shared__ __align__(16) u16 Bs[BN][LDK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int img = blockIdx.x;
+  const u16* di = dact + (size_t)img * H * W * COUT;
+
+  for (int c = tid; c < HP * WP * (COUT / 8); c += NTHREADS) {
+    int co = (c % (COUT / 8)) * 8;
+    int pix = c / (COUT / 8);
+    int xx = pix % WP, yy = pix / WP;
+    short8 v = short8{0, 0, 0, 0, 0, 0, 0, 0};
+    int sy = yy - 2, sx = xx - 2;
+    if (sy >= 0 && sy < H && sx >= 0 && sx < W)
+      v = *reinterpret_cast<const short8*>(di + ((size_t)sy * W + sx) * COUT + co);
+    *reinterpret_cast<short8*>(&slab[(yy * WP + xx) * PST + co]) = v;
+  }
+
+  int arow_off[MI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+    int lm = wr * WM + mi * 16 + (lane & 15);
+    int wi = lm % W, hi = (lm / W) % H;
+    if (lm >= M) { wi = 0; hi = 0; }
+    arow_off[mi] = ((hi + 2) * WP + (wi + 2)) * PST;  // padded (hi,wi)
+  }
+
+  f32x4 acc[MI][NI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int kq = (lane >> 4) * 8;
+  __syncthreads();
+
+  for (int kt = 0; kt < K; kt += BK) {
+    // Bs[ci][kk] = w[((khkw)*CIN + ci)*COUT + co(k)] — rows16 from w slices
+    for (int c = tid; c < BN * (BK / 8); c += NTHREADS) {
+      int i = c / (BK / 8);
+      int kc = (c % (BK / 8)) * 8;
+      int k = kt + kc;
+      int khkw = k / COUT, co = k % COUT;
+      short8 v = *reinterpret_cast<const short8*>(
+          w + ((size_t)khkw * CIN + i) * COUT + co);
+      *reinterpret_cast<short8*>(&Bs[i][kc]) = v;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kh2 = 0; kh2 < 2; ++kh2) {
+      int kbase = kt + kh2 * 32 + kq;       // k = khkw*COUT + co
+      int khkw = kbase / COUT, co = kbase % COUT;
+      int kh = khkw / 5, kw = khkw % 5;
+      // dact[hi - kh + 2, wi - kw + 2] -> padded offset shift
+      int shift = (-(kh - 2) * WP - (kw - 2)) * PST + co;
+      short8 af[MI], bf[NI];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+        af[mi] = *reinterpret_cast<const short8*>(
+            &slab[arow_off[mi] + shift]);
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni)
+        bf[ni] = *reinterpret_cast<const short8*>(
+            &Bs[wc * WN + ni * 16 + (lane & 15)][kh2 * 32 + kq]);
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int frow = (lane >> 4) * 4;
+  const int fcol = lane & 15;
+  u16* xo = dx + (size_t)img * M * CIN;
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int lm = wr * WM + mi * 16 + frow + r;
+      if (lm >= M) continue;
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni) {
+        int gc = wc * WN + ni * 16 + fcol;
+        xo[(size_t)lm * CIN + gc] = f2bf(acc[mi][ni][r]);
+      }
+    }
+  }
+}
+
 // ---- host wrappers --------------------------------------------------------
 void launch_conv_fwd_slab(const unsigned short* x, const unsigned short* w,
                           const float* bias, unsigned short* y, uint8_t* amax,
@@ -366,12 +479,20 @@ bool conv_slab_supported(int H, int W, int Cin, int Cout) {
   return H == 14 && W == 14 && Cin == 32 && Cout == 64;
 }
 
+
+
 void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
                          unsigned short* dx, int NB, int H, int W, int Cin,
                          int Cout, hipStream_t s) {
   if (H == 14 && W == 14 && Cin == 32 && Cout == 64) {
-    hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64>), dim3(NB),
-                       dim3(NTHREADS), 0, s, dact, w, dx, NB);
+    // glds-pipelined Bs wins below NB=2048; its extra LDS buffer costs one
+    // resident block (3 -> 2 per CU), which TLP-bound large batches feel
+    if (NB < 2048)
+      hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64>), dim3(NB),
+                         dim3(NTHREADS), 0, s, dact, w, dx, NB);
+    else
+      hipLaunchKernelGGL((conv_dx_slab_hi_occ_kernel<14, 14, 32, 64>),
+                         dim3(NB), dim3(NTHREADS), 0, s, dact, w, dx, NB);
   }
 }
 
