@@ -69,7 +69,12 @@ torch::Tensor linear_act_fwd_impl(torch::Tensor x, torch::Tensor w,
   p.offset_dev = offset_dev;
   bool drop = p_keep < 1.0;
   TORCH_CHECK(!drop || relu, "dropout path requires relu epilogue");
-  bool big = cdiv(M, 128) * cdiv(N, 128) >= 128;
+  // 128x128 tiles amortize staging best, but a 128^2 grid below ~2 blocks/CU
+  // is occupancy-starved (fc1 fwd @8192: 256 blocks = 1/CU, MFMA 6.9%);
+  // drop to 64x64 when the 128-grid can't fill the chip twice
+  int g128 = cdiv(M, 128) * cdiv(N, 128);
+  bool big = g128 >= 512;
+  if (const char* e = getenv("DMNIST_FC_TILE")) big = atoi(e) >= 128;
   auto s = cur_stream();
   bool bt = wT.has_value() && wT->defined();
   if (bt) {
