@@ -53,7 +53,7 @@ def main():
         "--batch_size", str(args.batch_size),
         "--train_dir", "/tmp/dmnist_bench",
         "--save_interval_secs", "100000",
-        "--max_steps", str(args.steps + args.warmup + 1),
+        "--max_steps", str(args.steps + args.warmup + 40),
     ])
     trainer = Trainer(tflags, device=device, rank=rank, world=world,
                       local_rank=local_rank)
@@ -71,15 +71,25 @@ def main():
         dist.barrier()
     if on_gpu:
         torch.cuda.synchronize()
-    step_t = np.zeros(args.steps)
     t0 = time.perf_counter()
     for i in range(args.steps):
-        ts = time.perf_counter()
         one_step()
-        step_t[i] = time.perf_counter() - ts
     if on_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
+
+    # per-step latency distribution, sampled SEPARATELY with a device sync
+    # per step (graph replays enqueue asynchronously, so un-synced per-step
+    # wall times are enqueue latencies, not step latencies); the headline
+    # elapsed/steps above stays free of per-step sync overhead
+    n_sample = min(args.steps, 30)
+    step_t = np.zeros(n_sample)
+    for i in range(n_sample):
+        ts = time.perf_counter()
+        one_step()
+        if on_gpu:
+            torch.cuda.synchronize()
+        step_t[i] = time.perf_counter() - ts
     if world > 1:
         dist.barrier()
         t = torch.tensor([elapsed], dtype=torch.float64, device=device if on_gpu else "cpu")
