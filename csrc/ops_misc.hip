@@ -544,18 +544,21 @@ void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
           for (int kr = 0; kr < 5; ++kr)
             win[(pc + 3) % 5][kr] = slab[r + kr][pc];
         const ushort_t* drow = di + (size_t)r * W * Cout;
-        // whole-row dact prefetch: all W loads issue back-to-back (raw u16
-        // until use), so the row pays ONE memory latency instead of a
-        // serial per-pixel load->25-FMA chain (5-wide double buffering
-        // still left this latency-bound)
-        ushort_t grow[28];
+        // double-buffered 5-wide dact prefetch: the serial load->25-FMA
+        // chain was latency-bound (one load in flight per wave); keeping
+        // the NEXT column-group's 5 loads in flight under the current
+        // group's 125 FMAs hides it.  Raw u16 until use so the load wait
+        // lands a full group later.
+        ushort_t gcur[5], gnxt[5];
 #pragma unroll
-        for (int p = 0; p < 28; ++p)
-          grow[p] = (p < W) ? drow[(size_t)p * Cout] : (ushort_t)0;
-        // constant loop bounds so grow[c]/win[...] indices fold to statics
-        // (a runtime index into a register array would spill to scratch)
+        for (int p = 0; p < 5; ++p)
+          gcur[p] = (p < W) ? drow[(size_t)p * Cout] : (ushort_t)0;
+        for (int cc = 0; cc < W; cc += 5) {
 #pragma unroll
-        for (int cc = 0; cc < 28; cc += 5) {
+          for (int p = 0; p < 5; ++p) {
+            int cn = cc + 5 + p;
+            gnxt[p] = (cn < W) ? drow[(size_t)cn * Cout] : (ushort_t)0;
+          }
 #pragma unroll
           for (int p = 0; p < 5; ++p) {
             int c = cc + p;
@@ -563,13 +566,15 @@ void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
 #pragma unroll
             for (int kr = 0; kr < 5; ++kr)  // new col c+2 -> slot (c+2)%5
               win[(p + 2) % 5][kr] = slab[r + kr][c + 4];
-            float gv = bf2f(grow[c]);
+            float gv = bf2f(gcur[p]);
 #pragma unroll
             for (int kh = 0; kh < 5; ++kh)
 #pragma unroll
               for (int kw = 0; kw < 5; ++kw)
                 acc[kh * 5 + kw] += win[(p + kw + 3) % 5][kh] * gv;
           }
+#pragma unroll
+          for (int p = 0; p < 5; ++p) gcur[p] = gnxt[p];
         }
       }
     }
