@@ -99,3 +99,13 @@ DEV short8 pack_wait(uint2 lo, uint2 hi) {
   asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(r.s));
   return r.s;
 }
+
+// packed bf16 dot product: c += a.x*b.x + a.y*b.y (v_dot2c_f32_bf16,
+// 2 MACs/cycle/lane — 2x the fp32 FMA rate)
+typedef short bf16x2_t __attribute__((__vector_size__(2 * sizeof(short))));
+DEV float dot2bf(unsigned a, unsigned b, float c) {
+  union U { unsigned u; bf16x2_t v; };
+  U ua, ub;
+  ua.u = a; ub.u = b;
+  return __builtin_amdgcn_fdot2_f32_bf16(ua.v, ub.v, c, false);
+}

@@ -383,75 +383,90 @@ extern "C" __global__ __launch_bounds__(256)
 void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
                              const float* bias, ushort_t* y, uint8_t* amax,
                              int NB, int H, int W, int Cout, int Mpool) {
-  // One IMAGE per block (H=W=28): x staged once as a fp32 [32][32] padded
-  // slab (4 KB); 256 threads = 32 co x 8 slots, each slot walking a
-  // CONTIGUOUS run of pooled pixels so the 6x6 window shifts by 2 columns
-  // per step — 12 incremental LDS reads (broadcast across the 32 co lanes)
-  // instead of 36.  The 25 filter taps live in registers (co is fixed per
-  // thread).  K=25 with Cin=1 is too small for MFMA to win; this VALU form
-  // runs at ~100 fp32 FMA per output.
-  __shared__ float slab[32][32];
-  __shared__ float wl[25][32];
+  // One IMAGE per block (H=W=28): x staged once as a PACKED bf16 [32][32]
+  // padded slab (2 KB); 256 threads = 32 co x 8 slots, each slot walking a
+  // CONTIGUOUS run of pooled pixels.  K=25 with Cin=1 is too small for
+  // MFMA; the math runs on v_dot2c_f32_bf16 (2 MACs/op): the 6-column
+  // window is held as 3 even-aligned bf16 PAIRS per row (one ds_read_b32
+  // each), odd-based pixels read the cross pairs built with one
+  // v_alignbit, and the kw=4 tail is a dot2 against a zero-padded tap
+  // pair — 15 dot2 per conv pixel, no scalar extraction anywhere.
+  __shared__ ushort_t slab[32][32];
   const int Ho = H / 2, Wo = W / 2;
   const int tid = threadIdx.x;
   const int img = blockIdx.x;
   const int qpi = Ho * Wo;              // pooled pixels per image (196)
-  for (int i = tid; i < 25 * Cout; i += 256)
-    wl[i / Cout][i % Cout] = bf2f(w[i]);
   const ushort_t* xi = x + (size_t)img * H * W;
   for (int i = tid; i < 32 * 32; i += 256) {
     int xx = i % 32, yy = i / 32;
     int sy = yy - 2, sx = xx - 2;
-    float v = 0.f;
-    if (sy >= 0 && sy < H && sx >= 0 && sx < W)
-      v = bf2f(xi[(size_t)sy * W + sx]);
+    ushort_t v = 0;
+    if (sy >= 0 && sy < H && sx >= 0 && sx < W) v = xi[(size_t)sy * W + sx];
     slab[yy][xx] = v;
   }
   __syncthreads();
   const int co = tid & 31, slot = tid >> 5;
   if (co >= Cout) return;
-  float wreg[25];
+  // packed taps: per kh, even pairs (w0,w1)(w2,w3) + tail pairs (w4,0) for
+  // even-based pixels and (0,w4) for odd-based ones
+  unsigned wpk[5][2], wt4e[5], wt4o[5];
 #pragma unroll
-  for (int t = 0; t < 25; ++t) wreg[t] = wl[t][co];
+  for (int kh = 0; kh < 5; ++kh) {
+    unsigned t0 = w[(kh * 5 + 0) * Cout + co], t1 = w[(kh * 5 + 1) * Cout + co];
+    unsigned t2 = w[(kh * 5 + 2) * Cout + co], t3 = w[(kh * 5 + 3) * Cout + co];
+    unsigned t4 = w[(kh * 5 + 4) * Cout + co];
+    wpk[kh][0] = t0 | (t1 << 16);
+    wpk[kh][1] = t2 | (t3 << 16);
+    wt4e[kh] = t4;
+    wt4o[kh] = t4 << 16;
+  }
   const float b = bias[co];
   const int run = (qpi + 7) / 8;        // 25 pooled pixels per slot
   int q0 = slot * run;
   int q1 = min(qpi, q0 + run);
   ushort_t* yi = y + (size_t)img * qpi * Cout;
   uint8_t* ai = amax + (size_t)img * qpi * Cout;
-  float wv[6][6];
+  unsigned wv[6][3];  // window: 3 even-aligned bf16 pairs per row
   int prev_ho = -9;
   for (int q = q0; q < q1; ++q) {
     int wo = q % Wo, ho = q / Wo;
     int oy = ho * 2, ox = wo * 2;       // slab coords = conv coords + 2 pad
     if (ho != prev_ho || wo == 0) {
-      // full window load (row change breaks column adjacency)
 #pragma unroll
       for (int r = 0; r < 6; ++r)
 #pragma unroll
-        for (int c = 0; c < 6; ++c) wv[r][c] = slab[oy + r][ox + c];
+        for (int c = 0; c < 3; ++c)
+          wv[r][c] = *reinterpret_cast<const unsigned*>(&slab[oy + r][ox + c * 2]);
       prev_ho = ho;
     } else {
-      // shift left by 2 columns, read the 2 new columns
+      // shift left by one pair, read the new pair
 #pragma unroll
       for (int r = 0; r < 6; ++r) {
-#pragma unroll
-        for (int c = 0; c < 4; ++c) wv[r][c] = wv[r][c + 2];
-        wv[r][4] = slab[oy + r][ox + 4];
-        wv[r][5] = slab[oy + r][ox + 5];
+        wv[r][0] = wv[r][1];
+        wv[r][1] = wv[r][2];
+        wv[r][2] = *reinterpret_cast<const unsigned*>(&slab[oy + r][ox + 4]);
       }
     }
     float acc0 = b, acc1 = b, acc2 = b, acc3 = b;
 #pragma unroll
-    for (int kh = 0; kh < 5; ++kh)
-#pragma unroll
-      for (int kw = 0; kw < 5; ++kw) {
-        float wc_ = wreg[kh * 5 + kw];
-        acc0 += wv[kh][kw] * wc_;
-        acc1 += wv[kh][kw + 1] * wc_;
-        acc2 += wv[kh + 1][kw] * wc_;
-        acc3 += wv[kh + 1][kw + 1] * wc_;
-      }
+    for (int kh = 0; kh < 5; ++kh) {
+      unsigned o0 = __builtin_amdgcn_alignbit(wv[kh][1], wv[kh][0], 16);
+      unsigned o1 = __builtin_amdgcn_alignbit(wv[kh][2], wv[kh][1], 16);
+      unsigned p0 = __builtin_amdgcn_alignbit(wv[kh + 1][1], wv[kh + 1][0], 16);
+      unsigned p1 = __builtin_amdgcn_alignbit(wv[kh + 1][2], wv[kh + 1][1], 16);
+      acc0 = dot2bf(wv[kh][0], wpk[kh][0],
+             dot2bf(wv[kh][1], wpk[kh][1],
+             dot2bf(wv[kh][2], wt4e[kh], acc0)));
+      acc1 = dot2bf(o0, wpk[kh][0],
+             dot2bf(o1, wpk[kh][1],
+             dot2bf(wv[kh][2], wt4o[kh], acc1)));
+      acc2 = dot2bf(wv[kh + 1][0], wpk[kh][0],
+             dot2bf(wv[kh + 1][1], wpk[kh][1],
+             dot2bf(wv[kh + 1][2], wt4e[kh], acc2)));
+      acc3 = dot2bf(p0, wpk[kh][0],
+             dot2bf(p1, wpk[kh][1],
+             dot2bf(wv[kh + 1][2], wt4o[kh], acc3)));
+    }
     float vals[4] = {acc0, acc1, acc2, acc3};
     float best = -1.0f / 0.0f;
     int barg = 0;
