@@ -526,7 +526,12 @@ __global__ __launch_bounds__(256)
 void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
                             float* dw, int NB, int H, int W, int Cout,
                             int G) {
-  __shared__ float slab[32][32];
+  // math on v_dot2c_f32_bf16: pixel PAIRS (c, c+1) reduce with one dot2
+  // per tap — acc[t] += x[c+kw-2]*g_c + x[c+kw-1]*g_{c+1} where the x pair
+  // is adjacent in the packed bf16 slab (even pairs = one b32 read, odd
+  // pairs = one v_alignbit) and the g pair packs two raw dact loads.
+  // 25 dot2 + 10 alignbit per 2 pixels vs 50 scalar FMAs.
+  __shared__ ushort_t slab[32][32];
   __shared__ float red[25][32];
   const int tid = threadIdx.x;
   const int co = tid & 31;
@@ -543,53 +548,65 @@ void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
     for (int i = tid; i < 32 * 32; i += 256) {
       int xx = i % 32, yy = i / 32;
       int sy = yy - 2, sx = xx - 2;
-      float v = 0.f;
-      if (sy >= 0 && sy < H && sx >= 0 && sx < W)
-        v = bf2f(xi[(size_t)sy * W + sx]);
+      ushort_t v = 0;
+      if (sy >= 0 && sy < H && sx >= 0 && sx < W) v = xi[(size_t)sy * W + sx];
       slab[yy][xx] = v;
     }
     __syncthreads();
     const ushort_t* di = dact + (size_t)img * H * W * Cout + co;
     if (co < Cout) {
       for (int r = hw; r < H; r += 8) {
-        float win[5][5];  // win[colslot][kh]; colslot = (x_col) mod 5
+        // window: 3 even-aligned x pairs per tap row, covering x cols
+        // c-2..c+3 for the current pixel pair (c even)
+        unsigned win[5][3];
 #pragma unroll
-        for (int pc = 0; pc < 4; ++pc)  // preload x cols -2..1 (slab 0..3)
-#pragma unroll
-          for (int kr = 0; kr < 5; ++kr)
-            win[(pc + 3) % 5][kr] = slab[r + kr][pc];
+        for (int kr = 0; kr < 5; ++kr) {  // c=0 pairs: slab cols (0, 2, 4)
+          win[kr][0] = *reinterpret_cast<const unsigned*>(&slab[r + kr][0]);
+          win[kr][1] = *reinterpret_cast<const unsigned*>(&slab[r + kr][2]);
+          win[kr][2] = *reinterpret_cast<const unsigned*>(&slab[r + kr][4]);
+        }
         const ushort_t* drow = di + (size_t)r * W * Cout;
-        // double-buffered 5-wide dact prefetch: the serial load->25-FMA
-        // chain was latency-bound (one load in flight per wave); keeping
-        // the NEXT column-group's 5 loads in flight under the current
-        // group's 125 FMAs hides it.  Raw u16 until use so the load wait
-        // lands a full group later.
-        ushort_t gcur[5], gnxt[5];
+        // 3-pair-deep dact prefetch (6 loads in flight, raw u16 until use)
+        ushort_t gcur[6], gnxt[6];
 #pragma unroll
-        for (int p = 0; p < 5; ++p)
+        for (int p = 0; p < 6; ++p)
           gcur[p] = (p < W) ? drow[(size_t)p * Cout] : (ushort_t)0;
-        for (int cc = 0; cc < W; cc += 5) {
+        for (int cc = 0; cc < 28; cc += 6) {  // 3 pixel pairs per group
 #pragma unroll
-          for (int p = 0; p < 5; ++p) {
-            int cn = cc + 5 + p;
+          for (int p = 0; p < 6; ++p) {
+            int cn = cc + 6 + p;
             gnxt[p] = (cn < W) ? drow[(size_t)cn * Cout] : (ushort_t)0;
           }
 #pragma unroll
-          for (int p = 0; p < 5; ++p) {
-            int c = cc + p;
+          for (int j = 0; j < 3; ++j) {
+            int c = cc + j * 2;
             if (c >= W) break;
+            unsigned gp = (unsigned)gcur[j * 2] |
+                          ((unsigned)gcur[j * 2 + 1] << 16);
 #pragma unroll
-            for (int kr = 0; kr < 5; ++kr)  // new col c+2 -> slot (c+2)%5
-              win[(p + 2) % 5][kr] = slab[r + kr][c + 4];
-            float gv = bf2f(gcur[p]);
+            for (int kh = 0; kh < 5; ++kh) {
+              unsigned o0 = __builtin_amdgcn_alignbit(win[kh][1], win[kh][0], 16);
+              unsigned o1 = __builtin_amdgcn_alignbit(win[kh][2], win[kh][1], 16);
+              acc[kh * 5 + 0] = dot2bf(win[kh][0], gp, acc[kh * 5 + 0]);
+              acc[kh * 5 + 1] = dot2bf(o0, gp, acc[kh * 5 + 1]);
+              acc[kh * 5 + 2] = dot2bf(win[kh][1], gp, acc[kh * 5 + 2]);
+              acc[kh * 5 + 3] = dot2bf(o1, gp, acc[kh * 5 + 3]);
+              acc[kh * 5 + 4] = dot2bf(win[kh][2], gp, acc[kh * 5 + 4]);
+            }
+            // advance window by one pair; clamp the last read inside the
+            // slab (c=26 would index col 32 — the value is discarded but
+            // an out-of-allocation LDS read can return NaN bits)
+            int nc = c + 6 <= 30 ? c + 6 : 30;
 #pragma unroll
-            for (int kh = 0; kh < 5; ++kh)
-#pragma unroll
-              for (int kw = 0; kw < 5; ++kw)
-                acc[kh * 5 + kw] += win[(p + kw + 3) % 5][kh] * gv;
+            for (int kr = 0; kr < 5; ++kr) {
+              win[kr][0] = win[kr][1];
+              win[kr][1] = win[kr][2];
+              win[kr][2] = *reinterpret_cast<const unsigned*>(
+                  &slab[r + kr][nc]);
+            }
           }
 #pragma unroll
-          for (int p = 0; p < 5; ++p) gcur[p] = gnxt[p];
+          for (int p = 0; p < 6; ++p) gcur[p] = gnxt[p];
         }
       }
     }
