@@ -31,13 +31,12 @@
 
 #define NT 256
 #define TBK 64
-#define TBM 128
 
 __device__ __align__(16) unsigned short g_zero_page[8];  // zero-init
 
 enum { AM_PLAIN = 0, AM_CONV5 = 1 };
 
-template <int BN, int AMODE>
+template <int BN, int AMODE, int TBM = 128>
 __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
   constexpr int WM = TBM / 2, WN = BN / 2;
   constexpr int MI = WM / 16, NI = WN / 16;
@@ -262,16 +261,29 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
 static inline int cdiv_h(int a, int b) { return (a + b - 1) / b; }
 
 void conv_dw_tr(const GemmParams& p, hipStream_t s) {
-  dim3 grid(cdiv_h(p.M, TBM) * cdiv_h(p.N, 64) * p.splitk);
-  hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5>), grid, dim3(NT), 0, s, p);
+  // TBM=256 halves the per-M-tile operand re-reads (gx 7 -> 4 at M=800)
+  // at the cost of 80 KB LDS (2 blocks/CU); DMNIST_DW_TBM=128 reverts
+  static int tbm = [] {
+    const char* e = getenv("DMNIST_DW_TBM");
+    return e ? atoi(e) : 256;
+  }();
+  if (tbm >= 256) {
+    dim3 grid(cdiv_h(p.M, 256) * cdiv_h(p.N, 64) * p.splitk);
+    hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5, 256>), grid, dim3(NT), 0,
+                       s, p);
+  } else {
+    dim3 grid(cdiv_h(p.M, 128) * cdiv_h(p.N, 64) * p.splitk);
+    hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5, 128>), grid, dim3(NT), 0,
+                       s, p);
+  }
 }
 
 void gemm_dw_tr_128(const GemmParams& p, hipStream_t s) {
-  dim3 grid(cdiv_h(p.M, TBM) * cdiv_h(p.N, 128) * p.splitk);
+  dim3 grid(cdiv_h(p.M, 128) * cdiv_h(p.N, 128) * p.splitk);
   hipLaunchKernelGGL((dw_tr_kernel<128, AM_PLAIN>), grid, dim3(NT), 0, s, p);
 }
 
 void gemm_dw_tr_64(const GemmParams& p, hipStream_t s) {
-  dim3 grid(cdiv_h(p.M, TBM) * cdiv_h(p.N, 64) * p.splitk);
+  dim3 grid(cdiv_h(p.M, 128) * cdiv_h(p.N, 64) * p.splitk);
   hipLaunchKernelGGL((dw_tr_kernel<64, AM_PLAIN>), grid, dim3(NT), 0, s, p);
 }
