@@ -262,10 +262,13 @@ static inline int cdiv_h(int a, int b) { return (a + b - 1) / b; }
 
 void conv_dw_tr(const GemmParams& p, hipStream_t s) {
   // TBM=256 halves the per-M-tile operand re-reads (gx 7 -> 4 at M=800)
-  // at the cost of 80 KB LDS (2 blocks/CU); DMNIST_DW_TBM=128 reverts
+  // but its 80 KB LDS drops occupancy to 2 blocks/CU and LOSES at both
+  // batch sizes (114 vs 81 us @1024, 511 vs 365 @8192) — same pattern as
+  // the conv dX glds experiment: these latency-bound gather kernels live
+  // on TLP.  Kept instantiated for DMNIST_DW_TBM=256 A/B runs.
   static int tbm = [] {
     const char* e = getenv("DMNIST_DW_TBM");
-    return e ? atoi(e) : 256;
+    return e ? atoi(e) : 128;
   }();
   if (tbm >= 256) {
     dim3 grid(cdiv_h(p.M, 256) * cdiv_h(p.N, 64) * p.splitk);
