@@ -71,9 +71,25 @@ class GraphedStep:
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
 
+        cur_stream = torch.cuda.current_stream()
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
-            self._body()
+        try:
+            with torch.cuda.graph(self.graph):
+                self._body()
+        except Exception:
+            # torch.cuda.graph.__exit__ calls capture_end() BEFORE popping
+            # its stream context: when a body op invalidated the capture,
+            # capture_end raises, the thread is left on the dead capture
+            # stream, and the capture can stay registered — every later op
+            # then fails with hipErrorStreamCaptureUnsupported instead of
+            # falling back to eager.  Terminate the capture and restore the
+            # stream before propagating (try_graph turns this into eager).
+            try:
+                self.graph.capture_end()
+            except Exception:
+                pass
+            torch.cuda.set_stream(cur_stream)
+            raise
 
         # restore pre-warmup state
         t.fp.flat_master.copy_(master0)

@@ -140,6 +140,12 @@ class Trainer:
                     and self.engine.timeout_s is None
                     and flags.inject_slow_rank < 0
                     and getattr(flags, "hip_graph", "auto") != "off")
+        if (eligible and self.world > 1 and dist.is_initialized()
+                and dist.get_backend() != "nccl"
+                and not os.environ.get("DMNIST_FORCE_GRAPH")):
+            # only RCCL collectives are hipGraph-capturable; a host-side
+            # backend (gloo) would invalidate every capture attempt
+            eligible = False
         if eligible:
             from .graphstep import try_graph
             self._graph = try_graph(self, tuple(images.shape))
