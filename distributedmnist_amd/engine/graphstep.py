@@ -59,6 +59,8 @@ class GraphedStep:
 
         # snapshot state: the warmup iterations below really train
         master0 = t.fp.flat_master.clone()
+        mom0 = (t.flat_momentum.clone()
+                if getattr(t, "flat_momentum", None) is not None else None)
         t.model.set_step_dev(self.step_dev)
 
         self._prime(t.step)
@@ -89,10 +91,22 @@ class GraphedStep:
             except Exception:
                 pass
             torch.cuda.set_stream(cur_stream)
+            # best-effort rollback of the warmup training (3 steps on zero
+            # images) so the eager fallback continues from the real weights
+            try:
+                t.fp.flat_master.copy_(master0)
+                if mom0 is not None:
+                    t.flat_momentum.copy_(mom0)
+                t.fp.sync_shadow()
+                t.model.set_step(t.step)
+            except Exception:
+                pass
             raise
 
         # restore pre-warmup state
         t.fp.flat_master.copy_(master0)
+        if mom0 is not None:
+            t.flat_momentum.copy_(mom0)
         t.fp.sync_shadow()
         self._prime(t.step)
         torch.cuda.synchronize()

@@ -66,8 +66,11 @@ class LeNet5(nn.Module):
         self._step_dev = t
 
     def set_step(self, step: int):
-        """Dropout offset — keeps masks deterministic per (seed, step)."""
+        """Dropout offset — keeps masks deterministic per (seed, step).
+        Clears any device-side counter: the host step is authoritative on
+        the eager path (e.g. after a failed hipGraph capture)."""
         self._step = int(step)
+        self._step_dev = None
 
     def _comp(self, name: str) -> torch.Tensor:
         return self.shadows.get(name, getattr(self, name))

@@ -42,6 +42,7 @@ class MLP(nn.Module):
 
     def set_step(self, step: int):
         self._step = int(step)
+        self._step_dev = None  # host step authoritative on the eager path
 
     def _comp(self, name: str) -> torch.Tensor:
         return self.shadows.get(name, getattr(self, name))
