@@ -49,3 +49,24 @@ def test_bench_torchrun_2proc():
     assert j["n_gpus"] == 2
     assert j["config"]["global_batch"] == 64
     assert j["config"]["parallelism"] == "dp2"
+
+
+@pytest.mark.gpu
+def test_bench_two_rank_gloo_on_one_gpu():
+    """Multi-rank GPU path end-to-end on a single device: 2 ranks share
+    cuda:0 over gloo (RCCL refuses duplicate devices), exercising the init
+    broadcast, the fused step, the bucketed all-reduce and the
+    graph-eligibility gate (gloo must fall back to eager cleanly)."""
+    env = dict(os.environ)
+    env.update({"DMNIST_BACKEND": "gloo", "MASTER_ADDR": "127.0.0.1"})
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29381", os.path.join(ROOT, "bench.py"),
+         "--gpus", "2", "--steps", "8", "--warmup", "2"],
+        capture_output=True, text=True, timeout=280, env=env, cwd=ROOT)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0
