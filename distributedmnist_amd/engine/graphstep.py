@@ -12,6 +12,18 @@ advanced INSIDE the graph by the step_advance kernel at the END of the body
 
 Only the full_sync mode is graphable (K-of-N / interval / CDF need
 data-dependent host control flow); Trainer falls back to eager elsewhere.
+
+Two capture scopes:
+  - FULL (world == 1, or --hip_graph full): the whole step including the
+    RCCL all-reduce is one graph.
+  - SPLIT (world > 1 default): only the compute body (fwd + two-stream bwd
+    -> grads in the flat bucket) is captured; the all-reduce and the
+    device-arg optimizer tail (sgd_step_dev, transpose refresh,
+    step_advance) run eagerly after each replay.  A collective that a
+    backend refuses to capture INVALIDATES the capture and leaves the
+    process unrecoverable (see docs/ROUND1_NOTES.md), so the multi-rank
+    default never captures one; the cost is ~4 extra launches + an
+    un-overlapped collective per step.
 """
 
 from __future__ import annotations
@@ -27,8 +39,9 @@ log = logging.getLogger("dmnist.graph")
 
 
 class GraphedStep:
-    def __init__(self, trainer, batch_shape):
+    def __init__(self, trainer, batch_shape, split: bool = False):
         t = trainer
+        self.split = split
         assert t.device.type == "cuda", "GraphedStep requires a GPU"
         assert t.mode == "full_sync" and t.engine.timeout_s is None, \
             "only the full_sync hot path is graph-captured"
@@ -63,6 +76,11 @@ class GraphedStep:
                 if getattr(t, "flat_momentum", None) is not None else None)
         t.model.set_step_dev(self.step_dev)
 
+        if self.split and self._fused is not None:
+            # split scope: the bucketed all-reduce must stay OUT of the
+            # captured region (run eagerly after replay instead)
+            self._fused.overlap_allreduce = False
+
         self._prime(t.step)
         torch.cuda.synchronize()
         side = torch.cuda.Stream()
@@ -77,7 +95,10 @@ class GraphedStep:
         self.graph = torch.cuda.CUDAGraph()
         try:
             with torch.cuda.graph(self.graph):
-                self._body()
+                if self.split:
+                    self._body_grads()
+                else:
+                    self._body()
         except Exception:
             # torch.cuda.graph.__exit__ calls capture_end() BEFORE popping
             # its stream context: when a body op invalidated the capture,
@@ -119,7 +140,9 @@ class GraphedStep:
               flags.learning_rate_decay_factor ** (step // self.decay_steps))
         self.lr_scale_dev.fill_(lr * self.inv_contrib)
 
-    def _body(self):
+    def _body_grads(self):
+        """fwd + bwd: gradients land in the flat bucket; returns True if
+        the fused step already issued the bucketed all-reduce."""
         t = self.t
         fp = t.fp
         fp.flat_grad.zero_()
@@ -133,8 +156,15 @@ class GraphedStep:
             logits = t.model(self.static_x, train=True)
             loss, acc = t.model.loss_and_accuracy(logits, self.static_y)
             loss.backward()
-        if not fused_reduced and t.world > 1 and dist.is_initialized():
-            dist.all_reduce(fp.flat_grad, op=dist.ReduceOp.SUM)
+        self.static_loss = loss.detach()
+        self.static_acc = acc.detach()
+        return fused_reduced
+
+    def _tail(self):
+        """optimizer + next-step device args (all args device-resident, so
+        this is capturable AND replay-equivalent when run eagerly)."""
+        t = self.t
+        fp = t.fp
         self._ext.sgd_step_dev(fp.flat_master, fp.flat_grad,
                                fp.flat_shadow if fp.flat_shadow is not None
                                else fp.flat_master,
@@ -144,27 +174,37 @@ class GraphedStep:
                                momentum=t.flat_momentum,
                                mu=t.flags.momentum)
         fp.refresh_transposes()
-        # advance step + LR on-device for the next replay
+        # advance step + LR on-device for the next body execution
         self._ext.step_advance(self.step_dev, self.lr_scale_dev,
                                t.flags.initial_learning_rate,
                                t.flags.learning_rate_decay_factor,
                                self.decay_steps, self.inv_contrib)
-        self.static_loss = loss.detach()
-        self.static_acc = acc.detach()
+
+    def _body(self):
+        t = self.t
+        fused_reduced = self._body_grads()
+        if not fused_reduced and t.world > 1 and dist.is_initialized():
+            dist.all_reduce(t.fp.flat_grad, op=dist.ReduceOp.SUM)
+        self._tail()
 
     def run(self, images, labels):
         """Replay one step. Returns (loss, acc) static device tensors."""
         self.static_x.copy_(images, non_blocking=True)
         self.static_y.copy_(labels, non_blocking=True)
         self.graph.replay()
+        if self.split:
+            t = self.t
+            if t.world > 1 and dist.is_initialized():
+                dist.all_reduce(t.fp.flat_grad, op=dist.ReduceOp.SUM)
+            self._tail()
         self.t.step += 1
         return self.static_loss, self.static_acc
 
 
-def try_graph(trainer, batch_shape):
+def try_graph(trainer, batch_shape, split: bool = False):
     """Build a GraphedStep, or None if capture is unsupported here."""
     try:
-        return GraphedStep(trainer, batch_shape)
+        return GraphedStep(trainer, batch_shape, split=split)
     except Exception as e:  # noqa: BLE001 — fall back to eager on any failure
         log.warning("hipGraph capture unavailable (%s); running eager", e)
         return None

@@ -140,15 +140,19 @@ class Trainer:
                     and self.engine.timeout_s is None
                     and flags.inject_slow_rank < 0
                     and getattr(flags, "hip_graph", "auto") != "off")
-        if (eligible and self.world > 1 and dist.is_initialized()
-                and dist.get_backend() != "nccl"
-                and not os.environ.get("DMNIST_FORCE_GRAPH")):
-            # only RCCL collectives are hipGraph-capturable; a host-side
-            # backend (gloo) would invalidate every capture attempt
-            eligible = False
         if eligible:
             from .graphstep import try_graph
-            self._graph = try_graph(self, tuple(images.shape))
+            # capture scope: world > 1 captures the COMPUTE only (split) so
+            # no collective is ever inside a capture — a backend that
+            # refuses capture invalidates it unrecoverably (segfault; see
+            # docs/ROUND1_NOTES.md).  --hip_graph full (or the
+            # DMNIST_FORCE_GRAPH diagnostic env) forces whole-step capture
+            # including the RCCL all-reduce.
+            force_full = (getattr(flags, "hip_graph", "auto") == "full"
+                          or bool(os.environ.get("DMNIST_FORCE_GRAPH")))
+            split = (self.world > 1 and dist.is_initialized()
+                     and not force_full)
+            self._graph = try_graph(self, tuple(images.shape), split=split)
         if self.world > 1 and dist.is_initialized():
             # ALL ranks must agree on graph-vs-eager: the graphed fused step
             # issues a bucketed 2-collective sequence, eager issues one —

@@ -72,7 +72,8 @@ def build_train_parser() -> argparse.ArgumentParser:
                    help="sleep this long per step on the injected rank")
     p.add_argument("--seed", type=int, default=66478)
     p.add_argument("--device", default="auto", help="auto|cpu|cuda")
-    p.add_argument("--hip_graph", default="auto", choices=["auto", "off"],
+    p.add_argument("--hip_graph", default="auto",
+                   choices=["auto", "off", "full"],
                    help="capture the training step in a hipGraph (GPU full-sync)")
     p.add_argument("--fused_step", default="auto", choices=["auto", "off"],
                    help="hand-scheduled two-stream LeNet step inside the graph")
