@@ -289,18 +289,29 @@ void launch_relu_drop_bwd(const unsigned short* dy, const unsigned short* y,
                      dyeff, db, B, N, inv_keep, apply_mask, rows_per_slice);
 }
 
+extern "C" __global__ void pool_bwd_gather_kernel(
+    const ushort_t* dy, const ushort_t* y, const uint8_t* amax,
+    ushort_t* dact, float* db, int Mpool, int C, int H, int W, int Wo);
+
 void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
                              const uint8_t* amax, unsigned short* dact,
                              float* db, int Mpool, int C, int H, int W, int Wo,
                              hipStream_t s) {
-  long total = (long)Mpool * (C / 8);
+  // gather form (dense output-ordered stores) unless DMNIST_POOL_SCATTER=1
+  static int use_scatter = [] {
+    const char* e = getenv("DMNIST_POOL_SCATTER");
+    return e ? atoi(e) : 0;
+  }();
+  long total = (long)Mpool * (C / 8) * (use_scatter ? 1 : 4);
   // >=4 granules per thread for memory-level parallelism; >=256 blocks to
   // fill the chip; <=2048 to bound the db atomics
   int blocks = cdivh(total, 256 * 4);
   if (blocks > 2048) blocks = 2048;
   if (blocks < 256) blocks = cdivh(total, 256) < 256 ? cdivh(total, 256) : 256;
-  hipLaunchKernelGGL(pool_bwd_scatter_kernel, dim3(blocks), dim3(256), 0, s,
-                     dy, y, amax, dact, db, Mpool, C, H, W, Wo);
+  hipLaunchKernelGGL(use_scatter ? pool_bwd_scatter_kernel
+                                 : pool_bwd_gather_kernel,
+                     dim3(blocks), dim3(256), 0, s, dy, y, amax, dact, db,
+                     Mpool, C, H, W, Wo);
 }
 
 void launch_softmax_xent(const unsigned short* logits, const long* labels,
@@ -632,4 +643,75 @@ void launch_conv1_dw_direct(const unsigned short* x,
   int blocks = (NB + G - 1) / G;
   hipLaunchKernelGGL(conv1_dw_direct_kernel, dim3(blocks), dim3(256), 0, s,
                      x, dact, dw, NB, H, W, Cout, G);
+}
+
+// ---------------------------------------------------------------------------
+// Maxpool 2x2 backward, GATHER form: one thread-granule per OUTPUT pixel
+// (8 channels).  The scatter form above writes 4 separate 16B stores per
+// window at a 2C-element stride — half-dense store streams that ran
+// 2.6-3.5x above the traffic floor.  Here every thread writes exactly its
+// own 16B of dact in layout order (fully dense, no read-for-ownership
+// partial lines); the pooled dy/y/amax reads are 4x redundant but
+// L2-broadcast (the 4 output pixels of a window read the same 16B).
+// db accumulates from the pos==0 visitor only (each window counted once).
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256)
+void pool_bwd_gather_kernel(const ushort_t* dy, const ushort_t* y,
+                            const uint8_t* amax, ushort_t* dact, float* db,
+                            int Mpool, int C, int H, int W, int Wo) {
+  __shared__ float partial[64];
+  if (threadIdx.x < 64) partial[threadIdx.x] = 0.f;
+  __syncthreads();
+  const unsigned CB8 = (unsigned)C / 8;
+  const unsigned cb8_sh = (CB8 == 8) ? 3u : (CB8 == 4 ? 2u : 0u);
+  const bool cb8_pow2 = (CB8 & (CB8 - 1)) == 0;
+  const unsigned Ho = (unsigned)H / 2;
+  unsigned total = (unsigned)Mpool * 4u * CB8;  // output granules
+  unsigned stride = gridDim.x * blockDim.x;
+  float local[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  int mycol = -1;
+  for (unsigned i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    unsigned cb, pix;
+    if (cb8_pow2 && cb8_sh) { cb = i & (CB8 - 1); pix = i >> cb8_sh; }
+    else { cb = i % CB8; pix = i / CB8; }
+    int c0 = (int)cb * 8;
+    unsigned w_ = pix % (unsigned)W;
+    unsigned t2 = pix / (unsigned)W;
+    unsigned h_ = t2 % (unsigned)H;
+    unsigned n = t2 / (unsigned)H;
+    int pos = (int)((h_ & 1u) * 2u + (w_ & 1u));
+    size_t qb = (((size_t)n * Ho + (h_ >> 1)) * (unsigned)Wo + (w_ >> 1)) * C
+                + c0;
+    short8 dyv = *reinterpret_cast<const short8*>(dy + qb);
+    short8 yv = *reinterpret_cast<const short8*>(y + qb);
+    uint64_t am8;
+    __builtin_memcpy(&am8, amax + qb, 8);
+    short8 out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float g = bf2f((ushort_t)dyv[e]);
+      if (!(bf2f((ushort_t)yv[e]) > 0.f)) g = 0.f;
+      int am = (int)((am8 >> (8 * e)) & 0xff);
+      out[e] = (am == pos) ? (short)f2bf(g) : (short)0;
+      if (pos == 0) local[e] += g;  // one visitor per window counts db
+    }
+    *reinterpret_cast<short8*>(dact + (size_t)i * 8) = out;
+    if (mycol < 0) mycol = c0;
+  }
+  if (mycol >= 0) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      atomicAdd(&partial[(mycol + e) & 63], local[e]);
+  }
+  __syncthreads();
+  if (db && threadIdx.x < 64) {
+    float v = partial[threadIdx.x];
+    if (C >= 64) {
+      if (v != 0.f) atomicAdd(&db[threadIdx.x], v);
+    } else if ((int)threadIdx.x < C) {
+      float v2 = partial[threadIdx.x + 32];
+      if (v + v2 != 0.f) atomicAdd(&db[threadIdx.x], v + v2);
+    }
+  }
 }
