@@ -1,7 +1,8 @@
 // Non-GEMM fused kernels: relu/dropout gradient mask (+ bias-grad column
 // sums), maxpool backward scatter (+ conv bias grad), fused softmax-CE
-// (+top-1 correct count), and the fused flat SGD apply (+drop-connect mask,
-// + bf16 shadow refresh).  All memory-bound: bf16 payloads, grid-stride.
+// (+top-1 correct count), the fused flat SGD apply (+drop-connect mask,
+// + bf16 shadow refresh), bf16 transpose, and the Cin=1 conv layer's
+// direct v_dot2c_f32_bf16 forward/dW (K=25 is below MFMA's win threshold).
 
 #include "common.h"
 #include "kernels.h"
