@@ -30,18 +30,18 @@
 #include "kernels.h"
 
 #define NT 256
-#define TBK 64
 
 __device__ __align__(16) unsigned short g_zero_page[8];  // zero-init
 
 enum { AM_PLAIN = 0, AM_CONV5 = 1 };
 
-template <int BN, int AMODE, int TBM = 128>
+template <int BN, int AMODE, int TBM = 128, int TBK = 64>
 __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
+  static_assert(TBK % 32 == 0, "TBK in 32-k MFMA halves");
   constexpr int WM = TBM / 2, WN = BN / 2;
   constexpr int MI = WM / 16, NI = WN / 16;
-  constexpr int AQ = TBM / 32;        // A glds per wave (1KB each)
-  constexpr int BQ = BN / 32;         // B glds per wave
+  constexpr int AQ = TBM * TBK / 2048;  // A glds per wave (1KB each)
+  constexpr int BQ = BN * TBK / 2048;   // B glds per wave
   constexpr int ALPR = TBM / 8;       // lanes per A row
   constexpr int BLPR = BN / 8;        // lanes per B row
   __shared__ __align__(16) ushort_t Aimg[2][TBK][TBM];  // k-major, LINEAR
@@ -195,7 +195,7 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
     unsigned abase = (unsigned)(uintptr_t)&Aimg[cur][0][0];
     unsigned bbase = (unsigned)(uintptr_t)&Bimg[cur][0][0];
 #pragma unroll
-    for (int kh2 = 0; kh2 < 2; ++kh2) {
+    for (int kh2 = 0; kh2 < TBK / 32; ++kh2) {
       int krow = kh2 * 32 + kgrp + slot_r;
       short8 af[MI], bf[NI];
       // slot address with the matching XOR chunk swizzle (see staging note)
@@ -270,15 +270,22 @@ void conv_dw_tr(const GemmParams& p, hipStream_t s) {
     const char* e = getenv("DMNIST_DW_TBM");
     return e ? atoi(e) : 128;
   }();
-  if (tbm >= 256) {
-    dim3 grid(cdiv_h(p.M, 256) * cdiv_h(p.N, 64) * p.splitk);
+  static int tbk = [] {
+    const char* e = getenv("DMNIST_DW_TBK");
+    return e ? atoi(e) : 64;
+  }();
+  dim3 grid(cdiv_h(p.M, tbm >= 256 ? 256 : 128) * cdiv_h(p.N, 64) * p.splitk);
+  if (tbm >= 256)
     hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5, 256>), grid, dim3(NT), 0,
                        s, p);
-  } else {
-    dim3 grid(cdiv_h(p.M, 128) * cdiv_h(p.N, 64) * p.splitk);
+  else if (tbk <= 32)
+    // 24 KB LDS -> 6 blocks/CU: TLP has been the winning lever on every
+    // latency-bound gather kernel in this file's history
+    hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5, 128, 32>), grid, dim3(NT),
+                       0, s, p);
+  else
     hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5, 128>), grid, dim3(NT), 0,
                        s, p);
-  }
 }
 
 void gemm_dw_tr_128(const GemmParams& p, hipStream_t s) {
