@@ -221,7 +221,7 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
 // conv dX: dx[hi,wi,ci] = sum_{kh,kw,co} dact[hi-kh+2, wi-kw+2, co] *
 //          w[kh,kw,ci,co] — dact slab per image, W read from global per tile.
 // ---------------------------------------------------------------------------
-template <int H, int W, int CIN, int COUT>
+template <int H, int W, int CIN, int COUT, int DB = 2>
 __global__ __launch_bounds__(NTHREADS)
 void conv_dx_slab_kernel(const u16* __restrict__ dact,
                          const u16* __restrict__ w,  // [25*CIN][COUT]
@@ -241,7 +241,12 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
   // 128B, so the 16B k-chunks are XOR-permuted per row (staging source and
   // fragment read agree on xs(row)) to keep the b128 fragment reads off a
   // single bank pair.
-  __shared__ __align__(16) u16 Bs[2][BN][BK];
+  // DB=2: double-buffered, glds for tile t+1 in flight under tile t's
+  // MFMAs, one barrier/K-step — but the extra 4 KB costs a resident block
+  // (2/CU) at this slab size.  DB=1: serial glds drain, two barriers, 3
+  // blocks/CU — the TLP wins at large NB (same threshold the hi-occ
+  // scatter variant sat on).
+  __shared__ __align__(16) u16 Bs[DB][BN][BK];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -291,14 +296,20 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
     for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int kq = (lane >> 4) * 8;
-  issueB(0, 0);    // drains at the slab-staging barrier
+  if (DB == 2) issueB(0, 0);  // drains at the slab-staging barrier
   __syncthreads();
 
   constexpr int NT_K = (K + BK - 1) / BK;
   int cur = 0;
   for (int t = 0; t < NT_K; ++t) {
     int kt = t * BK;
-    if (t + 1 < NT_K) issueB(cur ^ 1, kt + BK);  // hides under MFMAs
+    if (DB == 2) {
+      if (t + 1 < NT_K) issueB(cur ^ 1, kt + BK);  // hides under MFMAs
+    } else {
+      issueB(0, kt);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
     const u16* bp = &Bs[cur][0][0];
 #pragma unroll
     for (int kh2 = 0; kh2 < 2; ++kh2) {
@@ -327,8 +338,8 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
-    __syncthreads();  // drains the in-flight glds
-    cur ^= 1;
+    __syncthreads();  // DB=2: drains the in-flight glds
+    if (DB == 2) cur ^= 1;
   }
 
   const int frow = (lane >> 4) * 4;
