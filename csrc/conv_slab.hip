@@ -496,14 +496,20 @@ void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
                          unsigned short* dx, int NB, int H, int W, int Cin,
                          int Cout, hipStream_t s) {
   if (H == 14 && W == 14 && Cin == 32 && Cout == 64) {
-    // glds-pipelined Bs wins below NB=2048; its extra LDS buffer costs one
-    // resident block (3 -> 2 per CU), which TLP-bound large batches feel
+    // glds-pipelined (double-buffered) Bs wins below NB=2048; its extra
+    // LDS buffer costs a resident block (3 -> 2 per CU), which TLP-bound
+    // large batches feel — there the SINGLE-buffered glds form keeps
+    // 3/CU.  DMNIST_DX_HIOCC=1 reverts large batches to the original
+    // scatter-staged kernel for A/B.
     if (NB < 2048)
-      hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64>), dim3(NB),
+      hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 2>), dim3(NB),
                          dim3(NTHREADS), 0, s, dact, w, dx, NB);
-    else
+    else if (getenv("DMNIST_DX_HIOCC"))
       hipLaunchKernelGGL((conv_dx_slab_hi_occ_kernel<14, 14, 32, 64>),
                          dim3(NB), dim3(NTHREADS), 0, s, dact, w, dx, NB);
+    else
+      hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 1>), dim3(NB),
+                         dim3(NTHREADS), 0, s, dact, w, dx, NB);
   }
 }
 
