@@ -35,17 +35,18 @@ __device__ __align__(16) unsigned short g_zero_page[8];  // zero-init
 
 enum { AM_PLAIN = 0, AM_CONV5 = 1 };
 
-template <int BN, int AMODE, int TBM = 128, int TBK = 64>
+template <int BN, int AMODE, int TBM = 128, int TBK = 64, int NBUF = 2>
 __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
   static_assert(TBK % 32 == 0, "TBK in 32-k MFMA halves");
+  static_assert(NBUF == 2 || NBUF == 3, "2-buf overlap or 3-buf glds span");
   constexpr int WM = TBM / 2, WN = BN / 2;
   constexpr int MI = WM / 16, NI = WN / 16;
   constexpr int AQ = TBM * TBK / 2048;  // A glds per wave (1KB each)
   constexpr int BQ = BN * TBK / 2048;   // B glds per wave
   constexpr int ALPR = TBM / 8;       // lanes per A row
   constexpr int BLPR = BN / 8;        // lanes per B row
-  __shared__ __align__(16) ushort_t Aimg[2][TBK][TBM];  // k-major, LINEAR
-  __shared__ __align__(16) ushort_t Bimg[2][TBK][BN];
+  __shared__ __align__(16) ushort_t Aimg[NBUF][TBK][TBM];  // k-major, LINEAR
+  __shared__ __align__(16) ushort_t Bimg[NBUF][TBK][BN];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -185,15 +186,9 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
   const int slot_c = (lane & 3) * 4;
   const int kgrp = (lane >> 4) * 8;  // this group's k-offset within the half
 
-  int cur = 0;
-  if (nt > 0) issue(0, kbeg);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-
-  for (int t = 0; t < nt; ++t) {
-    if (t + 1 < nt) issue(cur ^ 1, kbeg + (t + 1) * TBK);  // flight under MFMA
-    unsigned abase = (unsigned)(uintptr_t)&Aimg[cur][0][0];
-    unsigned bbase = (unsigned)(uintptr_t)&Bimg[cur][0][0];
+  const auto compute = [&](int buf) {
+    unsigned abase = (unsigned)(uintptr_t)&Aimg[buf][0][0];
+    unsigned bbase = (unsigned)(uintptr_t)&Bimg[buf][0][0];
 #pragma unroll
     for (int kh2 = 0; kh2 < TBK / 32; ++kh2) {
       int krow = kh2 * 32 + kgrp + slot_r;
@@ -233,10 +228,44 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
-    // one barrier per K-step; the implicit vmcnt(0) drain at __syncthreads
-    // completes tile t+1's glds (2-buffer overlap)
+  };
+
+  constexpr int INFLIGHT = AQ + BQ;  // glds per wave per tile
+  if (NBUF == 3) {
+    // 3-buffer glds SPAN (guide: counted vmcnt + raw s_barrier): two tiles
+    // stay in flight across each barrier; each wave waits only until the
+    // tile it is about to READ is complete (its own glds are program-order,
+    // so vmcnt(INFLIGHT) leaves exactly the newest tile outstanding).
+    if (nt > 0) issue(0, kbeg);
+    if (nt > 1) {
+      issue(1 % NBUF, kbeg + TBK);
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(INFLIGHT) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    for (int t = 0; t < nt; ++t) {
+      if (t + 2 < nt) issue((t + 2) % NBUF, kbeg + (t + 2) * TBK);
+      compute(t % NBUF);
+      if (t + 2 < nt)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(INFLIGHT) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  } else {
+    int cur = 0;
+    if (nt > 0) issue(0, kbeg);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    cur ^= 1;
+    for (int t = 0; t < nt; ++t) {
+      if (t + 1 < nt) issue(cur ^ 1, kbeg + (t + 1) * TBK);  // under MFMAs
+      compute(cur);
+      // one barrier per K-step; the implicit vmcnt(0) drain at
+      // __syncthreads completes tile t+1's glds (2-buffer overlap)
+      __syncthreads();
+      cur ^= 1;
+    }
   }
 
   const int frow = (lane >> 4) * 4;
@@ -278,6 +307,10 @@ void conv_dw_tr(const GemmParams& p, hipStream_t s) {
   if (tbm >= 256)
     hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5, 256>), grid, dim3(NT), 0,
                        s, p);
+  else if (tbk <= 32 && getenv("DMNIST_DW_SPAN"))
+    // 3-buffer glds span (36 KB LDS -> 4 blocks/CU)
+    hipLaunchKernelGGL((dw_tr_kernel<64, AM_CONV5, 128, 32, 3>), grid,
+                       dim3(NT), 0, s, p);
   else if (tbk <= 32)
     // 24 KB LDS -> 6 blocks/CU: TLP has been the winning lever on every
     // latency-bound gather kernel in this file's history
