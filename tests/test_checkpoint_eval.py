@@ -78,3 +78,37 @@ def test_ps_role_exits(tmp_path):
         [sys.executable, os.path.join(ROOT, "src", "mnist_distributed_train.py"),
          "--job_name", "ps"], capture_output=True, text=True, timeout=60)
     assert out.returncode == 0
+
+
+def test_momentum_survives_checkpoint_resume(tmp_path):
+    """flat_momentum is part of the checkpoint payload: a resumed trainer
+    continues the momentum filter state bit-for-bit."""
+    flags = build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", str(tmp_path / "train"),
+         "--batch_size", "16", "--max_steps", "5", "--model", "mlp",
+         "--device", "cpu", "--momentum", "0.9",
+         "--save_interval_secs", "0"])
+    t = Trainer(flags)
+    ds = make_dataset(flags, 0, 1, t.device, t.compute_dtype)
+    t.train(ds)
+    assert t.flat_momentum is not None
+    assert float(t.flat_momentum.abs().sum()) > 0
+    t2 = Trainer(flags)
+    ds2 = make_dataset(flags, 0, 1, t2.device, t2.compute_dtype)
+    t2.train(ds2)  # restores at step 5 and exits the loop immediately
+    assert t2.step == 5
+    assert torch.equal(t2.flat_momentum, t.flat_momentum)
+    assert torch.equal(t2.fp.flat_master, t.fp.flat_master)
+
+
+def test_set_step_clears_device_counter():
+    """After a failed hipGraph capture the eager path must not keep using
+    the graph's frozen device-side step counter (the dropout offset would
+    stop advancing): set_step() makes the host step authoritative."""
+    from distributedmnist_amd.models import LeNet5
+    m = LeNet5(seed=1)
+    sentinel = torch.zeros(1, dtype=torch.int64)
+    m.set_step_dev(sentinel)
+    assert m._step_dev is sentinel
+    m.set_step(7)
+    assert m._step_dev is None and m._step == 7
