@@ -39,7 +39,10 @@ def resolve_device(flag_device: str, local_rank: int = 0) -> torch.device:
     if flag_device not in ("auto", ""):
         return torch.device(flag_device)
     if torch.cuda.is_available():
-        return torch.device(f"cuda:{local_rank}")
+        # modulo map so world_size > device_count still runs (exercising
+        # multi-rank paths on a 1-GPU box with --backend gloo); the 8-GPU
+        # launch maps 1:1
+        return torch.device(f"cuda:{local_rank % torch.cuda.device_count()}")
     return torch.device("cpu")
 
 
