@@ -68,7 +68,6 @@ class SyncEngine:
         self.cdf_start_tracking = cdf_start_tracking
         self._compute_times = []      # [(elapsed, worker, iteration), ...]
         self._iter_start_times = []
-        self._times_buf = torch.zeros(world_size, dtype=torch.float64, device="cpu")
 
     @property
     def distributed(self) -> bool:
