@@ -77,6 +77,32 @@ torch::Tensor linear_act_fwd_impl(torch::Tensor x, torch::Tensor w,
   if (const char* e = getenv("DMNIST_FC_TILE")) big = atoi(e) >= 128;
   auto s = cur_stream();
   bool bt = wT.has_value() && wT->defined();
+  // split-K forward for grid-starved shapes (fc1 fwd @B=1024: 128 blocks
+  // = half the chip idle): each k-slice writes its own fp32 [M][N] plane
+  // (deterministic — no atomics), a fused epilogue sums planes + applies
+  // bias/relu/philox-dropout.  DMNIST_FC_NOSPLIT=1 disables.
+  int tiles64 = cdiv(M, 64) * cdiv(N, 64);
+  int ksteps = cdiv(K, 64);
+  if (!big && tiles64 < 256 && ksteps >= 2 && !getenv("DMNIST_FC_NOSPLIT")) {
+    int sk = std::min(ksteps, std::max(2, 512 / tiles64));
+    auto acc = torch::empty({(int64_t)sk, (int64_t)M, (int64_t)N},
+                            x.options().dtype(at::kFloat));
+    p.C = acc.data_ptr();
+    p.splitk = sk;
+    if (bt) {
+      CHECK_BF16((*wT)); CHECK_CONTIG((*wT));
+      TORCH_CHECK(wT->size(0) == N && wT->size(1) == K, "wT shape mismatch");
+      p.B = bf16_ptr(*wT); p.ldb = K;
+      gemm_fwd_slices_64_bt(p, s);
+    } else {
+      p.B = bf16_ptr(w); p.ldb = N;
+      gemm_fwd_slices_64(p, s);
+    }
+    launch_fwd_epilogue(acc.data_ptr<float>(), b.data_ptr<float>(),
+                        bf16_mut(y), M, N, sk, relu ? 1 : 0, (float)p_keep,
+                        (uint64_t)seed, (uint64_t)offset, offset_dev, s);
+    return y;
+  }
   if (bt) {
     // pre-transposed weight copy [N][K] -> vector B staging (B_NMAJ)
     CHECK_BF16((*wT)); CHECK_CONTIG((*wT));

@@ -33,7 +33,7 @@ enum BMode {
 };
 enum Epi { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_RELU = 2, EPI_BIAS_RELU_DROP = 3,
            EPI_POOL = 4 };
-enum OutKind { OUT_BF16 = 0, OUT_F32_ATOMIC = 1 };
+enum OutKind { OUT_BF16 = 0, OUT_F32_ATOMIC = 1, OUT_F32_SLICES = 2 };
 
 #define NTHREADS 256
 #define BK 64
@@ -440,6 +440,13 @@ epilogue:
           }
           if (OUT == OUT_BF16) {
             reinterpret_cast<ushort_t*>(p.C)[(size_t)gr * p.ldc + gc] = f2bf(v);
+          } else if (OUT == OUT_F32_SLICES) {
+            // split-K without atomics: each k-slice owns a full [M][N]
+            // plane; a deterministic epilogue sums the planes (fwd path
+            // must be bitwise-reproducible per (seed, step))
+            float* outp = reinterpret_cast<float*>(p.C) +
+                          (size_t)bz * p.M * p.ldc;
+            outp[(size_t)gr * p.ldc + gc] = v;
           } else {
             atomicAdd(reinterpret_cast<float*>(p.C) + (size_t)gr * p.ldc + gc, v);
           }
@@ -487,6 +494,9 @@ GEMM_ENTRY(gemm_fwd_relu_128_bt, 128, 128, A_N, B_NMAJ, EPI_BIAS_RELU, OUT_BF16,
 GEMM_ENTRY(gemm_fwd_relu_64_bt, 64, 64, A_N, B_NMAJ, EPI_BIAS_RELU, OUT_BF16, 1)
 GEMM_ENTRY(gemm_fwd_drop_128_bt, 128, 128, A_N, B_NMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
 GEMM_ENTRY(gemm_fwd_drop_64_bt, 64, 64, A_N, B_NMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
+// split-K fwd slices (grid-starved small-M cases): deterministic planes
+GEMM_ENTRY(gemm_fwd_slices_64, 64, 64, A_N, B_KMAJ, EPI_NONE, OUT_F32_SLICES, 1)
+GEMM_ENTRY(gemm_fwd_slices_64_bt, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_F32_SLICES, 1)
 // conv gathers: thousands of WGs -> single-buffer, TLP hides latency
 GEMM_ENTRY(conv_fwd_pool, 128, 64, A_CONV_FWD, B_KMAJ, EPI_POOL, OUT_BF16, 0)
 GEMM_ENTRY(conv_fwd_pool_bt, 128, 64, A_CONV_FWD, B_NMAJ, EPI_POOL, OUT_BF16, 0)

@@ -109,3 +109,12 @@ void launch_conv1_dw_slab(const unsigned short* x, const unsigned short* dact,
 void launch_conv1_dw_direct(const unsigned short* x,
                             const unsigned short* dact, float* dw, int NB,
                             int H, int W, int Cout, hipStream_t s);
+// gemm_tile.hip — split-K fwd slice entries + their summing epilogue
+void gemm_fwd_slices_64(const GemmParams&, hipStream_t);
+void gemm_fwd_slices_64_bt(const GemmParams&, hipStream_t);
+// ops_misc.hip — fwd epilogue: y = [drop][relu](sum_slices + bias), bf16
+void launch_fwd_epilogue(const float* acc, const float* bias,
+                         unsigned short* y, int M, int N, int slices,
+                         int relu, float p_keep, uint64_t seed,
+                         uint64_t offset, const long* offset_dev,
+                         hipStream_t);

@@ -718,3 +718,47 @@ void pool_bwd_gather_kernel(const ushort_t* dy, const ushort_t* y,
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Split-K forward epilogue: y[m,n] = [drop][relu](sum_s acc[s][m][n] +
+// bias[n]) in bf16.  The slice sum runs in a FIXED order so the forward is
+// bitwise-reproducible per (seed, offset) — the reason the split-K fwd
+// writes per-slice planes instead of fp32 atomics.  Philox indexing
+// matches the fused GEMM epilogue exactly: idx = m * N + n.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256)
+void fwd_epilogue_kernel(const float* acc, const float* bias, ushort_t* y,
+                         int M, int N, int slices, int relu, float p_keep,
+                         uint64_t seed, uint64_t offset,
+                         const long* offset_dev) {
+  if (offset_dev) offset = (uint64_t)*offset_dev;
+  size_t total = (size_t)M * N;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    float v = 0.f;
+    for (int s = 0; s < slices; ++s) v += acc[(size_t)s * total + i];
+    int n = (int)(i % (size_t)N);
+    v += bias[n];
+    if (relu) v = v > 0.f ? v : 0.f;
+    if (p_keep < 1.0f) {
+      float u = philox_uniform(seed, offset, (uint64_t)i);
+      v = (u < p_keep) ? v / p_keep : 0.f;
+    }
+    y[i] = f2bf(v);
+  }
+}
+
+void launch_fwd_epilogue(const float* acc, const float* bias,
+                         unsigned short* y, int M, int N, int slices,
+                         int relu, float p_keep, uint64_t seed,
+                         uint64_t offset, const long* offset_dev,
+                         hipStream_t s) {
+  long total = (long)M * N;
+  int blocks = cdivh(total, 256 * 4);
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(fwd_epilogue_kernel, dim3(blocks), dim3(256), 0, s, acc,
+                     bias, y, M, N, slices, relu, p_keep, seed, offset,
+                     offset_dev);
+}
