@@ -320,8 +320,11 @@ void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
 void launch_softmax_xent(const unsigned short* logits, const long* labels,
                          unsigned short* dlogits, float* out, int B, int C,
                          hipStream_t s) {
-  dim3 grid(cdivh(B, 256));
-  hipLaunchKernelGGL(softmax_xent_kernel, grid, dim3(256), 0, s, logits,
+  // 64-thread blocks: one row per thread means B=1024 was only 4 blocks
+  // of 256 (256-CU chip ~idle, 16 us of pure latency); 16 blocks of 64
+  // spread the rows across CUs
+  dim3 grid(cdivh(B, 64));
+  hipLaunchKernelGGL(softmax_xent_kernel, grid, dim3(64), 0, s, logits,
                      labels, dlogits, out, B, C);
 }
 
