@@ -320,11 +320,12 @@ void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
 void launch_softmax_xent(const unsigned short* logits, const long* labels,
                          unsigned short* dlogits, float* out, int B, int C,
                          hipStream_t s) {
-  // 64-thread blocks: one row per thread means B=1024 was only 4 blocks
-  // of 256 (256-CU chip ~idle, 16 us of pure latency); 16 blocks of 64
-  // spread the rows across CUs
-  dim3 grid(cdivh(B, 64));
-  hipLaunchKernelGGL(softmax_xent_kernel, grid, dim3(64), 0, s, logits,
+  // one row per thread: B=1024 in 256-thread blocks was only 4 blocks
+  // (256-CU chip ~idle) — 64-thread blocks spread it; at large B the
+  // extra per-block out[] atomics cost more than the spread gains
+  int bt = B >= 2048 ? 256 : 64;
+  dim3 grid(cdivh(B, bt));
+  hipLaunchKernelGGL(softmax_xent_kernel, grid, dim3(bt), 0, s, logits,
                      labels, dlogits, out, B, C);
 }
 
