@@ -321,7 +321,10 @@ std::vector<torch::Tensor> conv_pool_bwd_impl(torch::Tensor dy, torch::Tensor x,
                           NB, H, W, Cin, Cout, s);
     } else {
       int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
-      int want = 2048;  // total blocks target; DMNIST_DW_BLOCKS sweeps it
+      // total blocks target (DMNIST_DW_BLOCKS sweeps it): at BK=32 the
+      // small-batch optimum is 1024 blocks (63 vs 76 us @1024), large
+      // batches want 2048 (tools/dwsweep.py)
+      int want = p.K <= 262144 ? 1024 : 2048;
       if (const char* e = getenv("DMNIST_DW_BLOCKS")) want = atoi(e);
       p.splitk = std::min(cdiv(p.K, 64), std::max(1, want / tiles));
       if (dw_tr_enabled() && Cin % 8 == 0 && Cout % 64 == 0)
@@ -465,7 +468,10 @@ void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
                         NB, H, W, Cin, Cout, s);
   } else {
     int tiles = cdiv(p.M, 128) * cdiv(p.N, 64);
-    int want = 2048;  // total blocks target; DMNIST_DW_BLOCKS sweeps it
+    // total blocks target (DMNIST_DW_BLOCKS sweeps it): at BK=32 the
+    // small-batch optimum is 1024 blocks (63 vs 76 us @1024), large
+    // batches want 2048 (tools/dwsweep.py)
+    int want = p.K <= 262144 ? 1024 : 2048;
     if (const char* e = getenv("DMNIST_DW_BLOCKS")) want = atoi(e);
     p.splitk = std::min(cdiv(p.K, 64), std::max(1, want / tiles));
     if (dw_tr_enabled() && Cin % 8 == 0 && Cout % 64 == 0)
