@@ -113,3 +113,30 @@ def test_real_data_training_path(tmp_path):
     # sharded 2-worker view of the same files
     ds2 = make_dataset(flags, 1, 2, t.device, t.compute_dtype)
     assert ds2.num_examples == 64
+
+
+def test_shard_partition_properties():
+    """Sharding invariants for any (n_examples, n_workers): the shards are
+    disjoint, cover the full set, and sizes differ by at most one (the
+    reference threaded worker_id/num_workers into unused args — SURVEY
+    §2.7 bug — so these properties define the FIXED behavior)."""
+    from hypothesis import given, settings, strategies as st
+    from distributedmnist_amd.data.mnist_data import DataSet
+
+    @settings(max_examples=40, deadline=None)
+    @given(n=st.integers(8, 300), w=st.integers(1, 9))
+    def check(n, w):
+        imgs = np.arange(n, dtype=np.float32).reshape(n, 1, 1, 1)
+        labs = np.arange(n, dtype=np.int64)
+        shards = [DataSet(imgs, labs, worker_id=r, n_workers=w, seed=0)
+                  for r in range(w)]
+        seen = np.concatenate([s.labels for s in shards])
+        assert sorted(seen.tolist()) == list(range(n))  # disjoint + cover
+        sizes = [s.num_examples for s in shards]
+        assert max(sizes) - min(sizes) <= 1              # balanced
+        # shard=False parity: every worker sees the whole set
+        full = DataSet(imgs, labs, worker_id=min(2, w - 1), n_workers=w,
+                       shard=False, seed=0)
+        assert full.num_examples == n
+
+    check()
