@@ -105,3 +105,68 @@ class FusedLeNetStep:
         # keep the side-stream consumers alive until the join (capture-safe)
         self._keep = (a1, h2, dyeff1, dyeff2, y1, dact2)
         return loss, correct
+
+    # ------------------------------------------------------------------
+    # Two-stage form of the SAME schedule, for the two-graph split capture
+    # (graphstep.py): stage_fc ends with every fc gradient final (so the
+    # eager fc-slice all-reduce can launch between the graphs and overlap
+    # stage_conv's replay); stage_conv finishes the conv backward.  The
+    # math and stream schedule per stage are identical to __call__ above —
+    # which stays the single-capture/eager body — except the collectives
+    # live OUTSIDE and the fc stage joins its side stream at the end so
+    # graph A's boundary is well-defined.
+    def stage_fc(self, x, labels, step_dev):
+        ext = self.ext
+        m = self.t.model
+        sh, shT = m.shadows, m.shadows_T
+        B = x.shape[0]
+
+        def gv(name):
+            return getattr(m, name).grad
+
+        s0 = torch.cuda.current_stream()
+        s1 = self.side
+        y1, am1 = ext.conv_pool_fwd(x, sh["conv1_w"], m.conv1_b, None)
+        y2, am2 = ext.conv_pool_fwd(y1, sh["conv2_w"], m.conv2_b,
+                                    shT["conv2_w"])
+        h2 = y2.view(B, 7 * 7 * 64)
+        a1 = ext.linear_act_fwd_dev(h2, sh["fc1_w"], m.fc1_b, True,
+                                    self.p_keep, self.seed, step_dev,
+                                    shT["fc1_w"])
+        logits = ext.linear_act_fwd(a1, sh["fc2_w"], m.fc2_b, False, 1.0,
+                                    0, 0, shT["fc2_w"])
+        loss, correct, dl = ext.softmax_xent_fwd(logits, labels)
+        dyeff2 = ext.mask_db(dl, dl, False, 1.0, gv("fc2_b"))
+        s1.wait_stream(s0)
+        with torch.cuda.stream(s1):
+            ext.linear_dw_into(a1, dyeff2, gv("fc2_w"))
+        dx2 = ext.linear_dx(dyeff2, sh["fc2_w"])
+        dyeff1 = ext.mask_db(dx2, a1, True, self.p_keep, gv("fc1_b"))
+        s1.wait_stream(s0)
+        with torch.cuda.stream(s1):
+            ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
+        dx1 = ext.linear_dx(dyeff1, sh["fc1_w"]).view(B, 7, 7, 64)
+        s0.wait_stream(s1)  # graph A boundary: fc grads complete
+        self._stash = (x, y1, y2, am1, am2, dx1)
+        self._keep = (a1, h2, dyeff1, dyeff2)
+        return loss, correct
+
+    def stage_conv(self):
+        ext = self.ext
+        m = self.t.model
+
+        def gv(name):
+            return getattr(m, name).grad
+
+        x, y1, y2, am1, am2, dx1 = self._stash
+        s0 = torch.cuda.current_stream()
+        s1 = self.side
+        dact2 = ext.pool_scatter(dx1, y2, am2, gv("conv2_b"), 14, 14)
+        s1.wait_stream(s0)
+        with torch.cuda.stream(s1):
+            ext.conv_dw_into(y1, dact2, gv("conv2_w"))
+        dxc = ext.conv_dx(dact2, m.shadows["conv2_w"], 32)
+        dact1 = ext.pool_scatter(dxc, y1, am1, gv("conv1_b"), 28, 28)
+        ext.conv_dw_into(x, dact1, gv("conv1_w"))
+        s0.wait_stream(s1)
+        self._keep2 = (dact2,)

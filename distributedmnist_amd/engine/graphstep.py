@@ -91,14 +91,41 @@ class GraphedStep:
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
 
+        # two-graph split: when the hand-scheduled step is available, the
+        # capture splits at the fc-grads-ready point so the (eager) fc-slice
+        # all-reduce can overlap graph B's conv backward — recovering the
+        # overlap that a single split capture serializes.  DMNIST_SPLIT2=0
+        # falls back to the single compute graph.
+        import os as _os
+        self.two_graph = (self.split and self._fused is not None
+                          and _os.environ.get("DMNIST_SPLIT2", "1") != "0")
+        fp = t.fp
+        fc0 = fp.offsets[fp.names.index("fc1_w")]
+        self.fc_slice = fp.flat_grad[fc0:]
+        self.conv_slice = fp.flat_grad[:fc0]
+        self._comm_stream = torch.cuda.Stream() if self.two_graph else None
         cur_stream = torch.cuda.current_stream()
         self.graph = torch.cuda.CUDAGraph()
+        self.graph_b = torch.cuda.CUDAGraph() if self.two_graph else None
         try:
-            with torch.cuda.graph(self.graph):
-                if self.split:
-                    self._body_grads()
-                else:
-                    self._body()
+            if self.two_graph:
+                with torch.cuda.graph(self.graph):
+                    t.fp.flat_grad.zero_()
+                    loss, correct = self._fused.stage_fc(
+                        self.static_x, self.static_y, self.step_dev)
+                    self.static_loss = loss.detach()
+                    acc = correct / self.static_x.shape[0]
+                    self.static_acc = acc.detach()
+                # graph B shares graph A's memory pool: it reads tensors
+                # graph A allocated (the stage_fc stash)
+                with torch.cuda.graph(self.graph_b, pool=self.graph.pool()):
+                    self._fused.stage_conv()
+            else:
+                with torch.cuda.graph(self.graph):
+                    if self.split:
+                        self._body_grads()
+                    else:
+                        self._body()
         except Exception:
             # torch.cuda.graph.__exit__ calls capture_end() BEFORE popping
             # its stream context: when a body op invalidated the capture,
@@ -107,10 +134,13 @@ class GraphedStep:
             # then fails with hipErrorStreamCaptureUnsupported instead of
             # falling back to eager.  Terminate the capture and restore the
             # stream before propagating (try_graph turns this into eager).
-            try:
-                self.graph.capture_end()
-            except Exception:
-                pass
+            for g in (self.graph, self.graph_b):
+                if g is None:
+                    continue
+                try:
+                    g.capture_end()
+                except Exception:
+                    pass
             torch.cuda.set_stream(cur_stream)
             # best-effort rollback of the warmup training (3 steps on zero
             # images) so the eager fallback continues from the real weights
@@ -192,7 +222,22 @@ class GraphedStep:
         self.static_x.copy_(images, non_blocking=True)
         self.static_y.copy_(labels, non_blocking=True)
         self.graph.replay()
-        if self.split:
+        if self.two_graph:
+            t = self.t
+            main = torch.cuda.current_stream()
+            reduced = t.world > 1 and dist.is_initialized()
+            if reduced:
+                # fc slice reduces on a comm stream WHILE graph B replays
+                # the conv backward on the main stream
+                self._comm_stream.wait_stream(main)
+                with torch.cuda.stream(self._comm_stream):
+                    dist.all_reduce(self.fc_slice, op=dist.ReduceOp.SUM)
+            self.graph_b.replay()
+            if reduced:
+                main.wait_stream(self._comm_stream)
+                dist.all_reduce(self.conv_slice, op=dist.ReduceOp.SUM)
+            self._tail()
+        elif self.split:
             t = self.t
             if t.world > 1 and dist.is_initialized():
                 dist.all_reduce(t.fp.flat_grad, op=dist.ReduceOp.SUM)
