@@ -70,3 +70,8 @@ def test_bench_two_rank_gloo_on_one_gpu():
     d = json.loads(line)
     assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0
+    # the two-graph split capture must actually engage: an eager fallback
+    # would still produce a valid JSON line but lose the graph replay and
+    # the overlapped fc all-reduce at scale
+    combined = out.stdout + out.stderr
+    assert "running eager" not in combined, combined[-1500:]
