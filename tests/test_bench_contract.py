@@ -75,3 +75,35 @@ def test_bench_two_rank_gloo_on_one_gpu():
     # the overlapped fc all-reduce at scale
     combined = out.stdout + out.stderr
     assert "running eager" not in combined, combined[-1500:]
+
+
+@pytest.mark.gpu
+def test_two_graph_split_matches_eager_numerics(tmp_path):
+    """The two-graph split capture must train the same trajectory as the
+    eager multi-rank path (within fp32-atomic noise): same seeds, same
+    synthetic data, 30 steps, 2 ranks sharing one GPU over gloo."""
+    import re
+
+    def run(port, extra):
+        env = dict(os.environ)
+        env["DMNIST_BACKEND"] = "gloo"
+        out = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", str(port),
+             os.path.join(ROOT, "src", "mnist_distributed_train.py"),
+             "--synthetic_data", "--backend", "gloo", "--batch_size", "128",
+             "--max_steps", "30", "--save_interval_secs", "100000",
+             "--train_dir", str(tmp_path / f"t{port}")] + extra,
+            capture_output=True, text=True, timeout=280, env=env, cwd=ROOT)
+        assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+        txt = out.stdout + out.stderr
+        losses = re.findall(r"Worker 0: .*step 29, loss = ([0-9.]+)", txt)
+        assert losses, txt[-1500:]
+        return float(losses[0]), txt
+
+    loss_g, txt_g = run(29392, [])
+    assert "running eager" not in txt_g  # the capture must engage
+    loss_e, _ = run(29393, ["--hip_graph", "off"])
+    assert abs(loss_g - loss_e) < 0.05 * max(1.0, abs(loss_e)), \
+        (loss_g, loss_e)
