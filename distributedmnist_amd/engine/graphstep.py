@@ -83,6 +83,17 @@ class GraphedStep:
 
         self._prime(t.step)
         torch.cuda.synchronize()
+        # Warmup is COLLECTIVE-FREE: it trains 3 throwaway steps on zero
+        # images whose state is rolled back below, so skipping the
+        # all-reduce changes nothing — and it means a rank that failed
+        # earlier in __init__ (e.g. static-buffer OOM) cannot leave the
+        # surviving ranks hung in a warmup collective.  Divergence after
+        # try_graph returns is caught by Trainer.get_graph's agreement
+        # all-reduce.
+        self._warmup = True
+        fused_ar = self._fused.overlap_allreduce if self._fused else False
+        if self._fused is not None:
+            self._fused.overlap_allreduce = False
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -90,6 +101,9 @@ class GraphedStep:
                 self._body()
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
+        self._warmup = False
+        if self._fused is not None:
+            self._fused.overlap_allreduce = fused_ar
 
         # two-graph split: when the hand-scheduled step is available, the
         # capture splits at the fc-grads-ready point so the (eager) fc-slice
@@ -213,7 +227,8 @@ class GraphedStep:
     def _body(self):
         t = self.t
         fused_reduced = self._body_grads()
-        if not fused_reduced and t.world > 1 and dist.is_initialized():
+        if (not fused_reduced and not getattr(self, "_warmup", False)
+                and t.world > 1 and dist.is_initialized()):
             dist.all_reduce(t.fp.flat_grad, op=dist.ReduceOp.SUM)
         self._tail()
 

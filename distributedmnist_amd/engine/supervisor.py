@@ -31,18 +31,24 @@ class Supervisor:
             os.makedirs(train_dir, exist_ok=True)
 
     # -- save ----------------------------------------------------------
-    def maybe_save(self, step: int, payload: dict) -> bool:
+    def maybe_save(self, step: int, payload) -> bool:
+        """`payload` may be a dict or a zero-arg callable returning one.
+        Pass a callable from the train loop: the payload build is a full
+        D2H copy of every buffer, so it must only happen when a save is
+        actually due (chief + interval elapsed), not every step."""
         if not self.is_chief:
             return False
         now = time.time()
         if now - self._last_save < self.save_interval_secs:
             return False
-        self.save(step, payload)
+        self.save(step, payload() if callable(payload) else payload)
         return True
 
-    def save(self, step: int, payload: dict):
+    def save(self, step: int, payload):
         if not self.is_chief:
             return
+        if callable(payload):
+            payload = payload()
         name = f"model.ckpt-{step}"
         path = os.path.join(self.train_dir, name)
         tmp = path + ".tmp"
@@ -89,7 +95,9 @@ class Supervisor:
         if latest is None:
             return None
         step, path = latest
-        payload = torch.load(path, map_location=map_location, weights_only=False)
+        # tensor-only payload (tensors + str/int containers) — loads under
+        # weights_only=True, so a tampered checkpoint cannot execute code
+        payload = torch.load(path, map_location=map_location, weights_only=True)
         return step, payload
 
 

@@ -163,7 +163,20 @@ class Trainer:
             ok = torch.tensor(
                 [1.0 if self._graph is not None else 0.0],
                 device=self.device if self.device.type == "cuda" else "cpu")
-            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            # watchdog: a rank that died during capture would leave the
+            # others hung here forever — bound the wait and fail loudly
+            import datetime as _dt
+            work = dist.all_reduce(ok, op=dist.ReduceOp.MIN, async_op=True)
+            try:
+                done = work.wait(timeout=_dt.timedelta(seconds=180))
+            except Exception as e:  # noqa: BLE001 — backend-specific timeout
+                raise RuntimeError(
+                    "graph-agreement all-reduce failed/timed out: a peer "
+                    "rank likely died during hipGraph capture") from e
+            if done is False:
+                raise RuntimeError(
+                    "graph-agreement all-reduce timed out after 180s: a "
+                    "peer rank likely died during hipGraph capture")
             if float(ok.item()) < 0.5 and self._graph is not None:
                 log.warning("hipGraph disabled: another rank failed capture")
                 self._graph = None
@@ -289,7 +302,7 @@ class Trainer:
                 writer.add_scalar("Train Accuracy", acc_v, self.step)
                 writer.add_scalar("Examples/sec", examples_per_sec, self.step)
                 next_summary_time += flags.save_summaries_secs
-            sv.maybe_save(self.step, self.checkpoint_payload())
+            sv.maybe_save(self.step, self.checkpoint_payload)
         if writer is not None:
             writer.close()
         if self.is_chief:

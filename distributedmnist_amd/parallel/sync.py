@@ -77,18 +77,22 @@ class SyncEngine:
     def step_begin(self, step: int):
         self._t_start = time.time()
         if self.mode == "cdf":
-            self._iter_start_times.append(self._t_start)
+            # (step, t_start) pairs: list position is NOT the step number
+            # after a checkpoint restore (training resumes at step0 > 0)
+            self._iter_start_times.append((step, self._t_start))
 
     def _all_gather_times(self, my_time: float, step: int):
         """Collect per-rank compute times (cdf instrumentation / k_of_n
-        ranking). Uses a small all_gather on the training communicator."""
+        ranking). One small all_gather_into_tensor + ONE host sync (a
+        per-element .item() loop costs `world` separate syncs per step —
+        at 8 ranks that becomes the step time)."""
         if not self.distributed:
             return [my_time]
         t = torch.tensor([my_time], dtype=torch.float64,
                          device=self.flat_grad.device)
-        out = [torch.zeros_like(t) for _ in range(self.world)]
-        dist.all_gather(out, t, group=self.group)
-        return [float(x.item()) for x in out]
+        out = torch.zeros(self.world, dtype=torch.float64, device=t.device)
+        dist.all_gather_into_tensor(out, t, group=self.group)
+        return out.tolist()
 
     # ------------------------------------------------------------------
     def reduce(self, step: int, compute_time_s: float | None = None):
@@ -191,8 +195,8 @@ class SyncEngine:
             return
         elapsed = sorted((t, w, i) for (t, w, i) in self._compute_times
                          if i > self.cdf_start_tracking)
-        starts = [t for idx, t in enumerate(self._iter_start_times)
-                  if idx > self.cdf_start_tracking]
+        starts = [t for (s, t) in self._iter_start_times
+                  if s > self.cdf_start_tracking]
         iter_times = [starts[i + 1] - starts[i] for i in range(len(starts) - 1)]
         log.info("ELAPSED TIMES %s", str(elapsed))
         log.info("ITERATION TIMES %s", str(iter_times))
