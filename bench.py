@@ -26,13 +26,35 @@ import torch.distributed as dist
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=40)
-    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--steps", type=int, default=None,
+                    help="timed steps (default: 2000 GPU / 40 CPU so SMI "
+                         "sampling sees a multi-second timed region)")
+    ap.add_argument("--warmup", type=int, default=None)
     ap.add_argument("--batch_size", type=int, default=1024,
                     help="per-GPU batch (weak scaling)")
     ap.add_argument("--model", default="lenet")
-    ap.add_argument("--mode", default="full_sync")
+    ap.add_argument("--mode", default="full_sync",
+                    choices=["full_sync", "k_of_n", "interval", "cdf"],
+                    help="mapped onto the trainer's DP-mode flags (NOT "
+                         "cosmetic: a cdf bench runs the cdf engine)")
+    ap.add_argument("--replicas_to_aggregate", type=int, default=-1,
+                    help="K for --mode k_of_n (-1 => world size)")
+    ap.add_argument("--interval_ms", type=float, default=100.0,
+                    help="aggregation period for --mode interval")
+    ap.add_argument("--straggler", type=int, default=-1,
+                    help="rank to slow down (straggler injection)")
+    ap.add_argument("--straggler_ms", type=float, default=5.0,
+                    help="per-step sleep on the injected rank")
+    ap.add_argument("--straggler_timeout_ms", type=float, default=0.0,
+                    help=">0: drop a rank's gradient past this deadline")
+    ap.add_argument("--grad_dtype", default="fp32", choices=["fp32", "bf16"],
+                    help="all-reduce wire dtype (fp32 master kept either way)")
     args = ap.parse_args()
+    on_gpu_probe = torch.cuda.is_available()
+    if args.steps is None:
+        args.steps = 2000 if on_gpu_probe else 40
+    if args.warmup is None:
+        args.warmup = 200 if on_gpu_probe else 10
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -52,13 +74,32 @@ def main():
     from distributedmnist_amd.engine.train import Trainer, make_dataset
     from distributedmnist_amd.utils.flags import build_train_parser
 
-    tflags = build_train_parser().parse_args([
+    targv = [
         "--synthetic_data", "--model", args.model,
         "--batch_size", str(args.batch_size),
         "--train_dir", "/tmp/dmnist_bench",
         "--save_interval_secs", "100000",
-        "--max_steps", str(args.steps + args.warmup + 40),
-    ])
+        "--max_steps", str(args.steps + args.warmup + 60),
+        "--grad_dtype", args.grad_dtype,
+    ]
+    # --mode maps onto the trainer's DP flags (reference mode semantics,
+    # SURVEY.md section 2.2); the JSON "mode" field below is therefore the
+    # mode that actually ran
+    if args.mode == "cdf":
+        targv += ["--worker_times_cdf_method"]
+    elif args.mode == "interval":
+        targv += ["--interval_method", "--interval_ms", str(args.interval_ms)]
+    elif args.mode == "k_of_n":
+        k = args.replicas_to_aggregate
+        if k <= 0:
+            k = max(1, int(os.environ.get("WORLD_SIZE", "1")) - 1)
+        targv += ["--num_replicas_to_aggregate", str(k)]
+    if args.straggler >= 0:
+        targv += ["--inject_slow_rank", str(args.straggler),
+                  "--inject_slow_ms", str(args.straggler_ms)]
+    if args.straggler_timeout_ms > 0:
+        targv += ["--straggler_timeout_ms", str(args.straggler_timeout_ms)]
+    tflags = build_train_parser().parse_args(targv)
     trainer = Trainer(tflags, device=device, rank=rank, world=world,
                       local_rank=local_rank)
     trainer._num_examples = 60000
@@ -94,6 +135,11 @@ def main():
         if on_gpu:
             torch.cuda.synchronize()
         step_t[i] = time.perf_counter() - ts
+    if trainer.mode == "interval":
+        # drain outstanding interval aggregations so every rank leaves the
+        # communicator with a matched collective count
+        for grad, c in trainer.engine.finalize_interval():
+            trainer._apply_update(grad, c)
     if world > 1:
         dist.barrier()
         t = torch.tensor([elapsed], dtype=torch.float64, device=device if on_gpu else "cpu")
@@ -126,7 +172,11 @@ def main():
                 "p50_ms_per_step": round(float(np.percentile(step_t, 50)) * 1000, 4),
                 "p95_ms_per_step": round(float(np.percentile(step_t, 95)) * 1000, 4),
                 "p99_ms_per_step": round(float(np.percentile(step_t, 99)) * 1000, 4),
-                "mode": args.mode,
+                # the engine mode that actually ran (k_of_n degenerates to
+                # full_sync at world=1 — reported honestly, never relabeled)
+                "mode": trainer.mode,
+                "mode_requested": args.mode,
+                "grad_dtype": args.grad_dtype,
             },
         }), flush=True)
     if world > 1:

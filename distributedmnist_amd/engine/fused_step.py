@@ -43,6 +43,7 @@ class FusedLeNetStep:
         # (conv*, then fc*) makes the split two contiguous slices.
         fp = t.fp
         fc0 = fp.offsets[fp.names.index("fc1_w")]
+        self.fc_offset = fc0
         self.conv_slice = fp.flat_grad[:fc0]
         self.fc_slice = fp.flat_grad[fc0:]
         self.overlap_allreduce = t.world > 1 and dist.is_initialized()
@@ -87,7 +88,8 @@ class FusedLeNetStep:
             if self.overlap_allreduce:
                 # every fc gradient (fc2 via the earlier s1 work + the db's
                 # ordered by wait_stream) is final here
-                dist.all_reduce(self.fc_slice, op=dist.ReduceOp.SUM)
+                self.t.engine.wire_allreduce(self.fc_slice,
+                                             start=self.fc_offset)
         dx1 = ext.linear_dx(dyeff1, sh["fc1_w"]).view(B, 7, 7, 64)
 
         dact2 = ext.pool_scatter(dx1, y2, am2, gv("conv2_b"), 14, 14)
@@ -101,7 +103,7 @@ class FusedLeNetStep:
 
         s0.wait_stream(s1)
         if self.overlap_allreduce:
-            dist.all_reduce(self.conv_slice, op=dist.ReduceOp.SUM)
+            self.t.engine.wire_allreduce(self.conv_slice, start=0)
         # keep the side-stream consumers alive until the join (capture-safe)
         self._keep = (a1, h2, dyeff1, dyeff2, y1, dact2)
         return loss, correct

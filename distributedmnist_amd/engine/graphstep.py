@@ -115,6 +115,7 @@ class GraphedStep:
                           and _os.environ.get("DMNIST_SPLIT2", "1") != "0")
         fp = t.fp
         fc0 = fp.offsets[fp.names.index("fc1_w")]
+        self.fc_offset = fc0
         self.fc_slice = fp.flat_grad[fc0:]
         self.conv_slice = fp.flat_grad[:fc0]
         self._comm_stream = torch.cuda.Stream() if self.two_graph else None
@@ -246,16 +247,17 @@ class GraphedStep:
                 # the conv backward on the main stream
                 self._comm_stream.wait_stream(main)
                 with torch.cuda.stream(self._comm_stream):
-                    dist.all_reduce(self.fc_slice, op=dist.ReduceOp.SUM)
+                    t.engine.wire_allreduce(self.fc_slice,
+                                            start=self.fc_offset)
             self.graph_b.replay()
             if reduced:
                 main.wait_stream(self._comm_stream)
-                dist.all_reduce(self.conv_slice, op=dist.ReduceOp.SUM)
+                t.engine.wire_allreduce(self.conv_slice, start=0)
             self._tail()
         elif self.split:
             t = self.t
             if t.world > 1 and dist.is_initialized():
-                dist.all_reduce(t.fp.flat_grad, op=dist.ReduceOp.SUM)
+                t.engine.wire_allreduce(t.fp.flat_grad)
             self._tail()
         self.t.step += 1
         return self.static_loss, self.static_acc

@@ -104,7 +104,10 @@ class Trainer:
             replicas_to_aggregate=flags.num_replicas_to_aggregate,
             interval_ms=flags.interval_ms,
             straggler_timeout_ms=(flags.straggler_timeout_ms or None),
-            rank=rank, world_size=world)
+            rank=rank, world_size=world,
+            wire_dtype=(torch.bfloat16
+                        if getattr(flags, "grad_dtype", "fp32") == "bf16"
+                        else None))
         self.timer = StepTimer(self.device)
         self.is_chief = rank == 0
         self.step = 0
@@ -219,19 +222,31 @@ class Trainer:
         compute_time = self.timer.stop() if timed else 0.0
         applied, grad, contributors = self.engine.reduce(self.step, compute_time)
         if applied:
-            lr = lr_at(self.step, flags, self._num_examples,
-                       max(1, self.engine.K))
-            Fx.sgd_step(self.fp.flat_master, grad, lr,
-                        grad_scale=1.0 / max(1, contributors),
-                        drop_connect_keep=(flags.drop_connect_probability
-                                           if flags.drop_connect else None),
-                        seed=flags.seed, offset=self.step,
-                        shadow=self.fp.flat_shadow,
-                        momentum=self.flat_momentum, mu=flags.momentum)
-            self.fp.refresh_transposes()
-            self.num_contributors = contributors
+            self._apply_update(grad, contributors)
         self.step += 1
         return applied, loss.detach(), acc.detach(), compute_time
+
+    def _apply_update(self, grad, contributors):
+        """Fused SGD apply of one aggregated gradient.  Interval mode keys
+        the LR staircase on the aggregation count (the reference's
+        global_step increments once per take_grad apply, not per worker
+        step); every other mode keys it on the local step as before."""
+        flags = self.flags
+        if self.mode == "interval":
+            lr_step = max(0, self.engine.generation - 1)
+        else:
+            lr_step = self.step
+        lr = lr_at(lr_step, flags, self._num_examples,
+                   max(1, self.engine.K))
+        Fx.sgd_step(self.fp.flat_master, grad, lr,
+                    grad_scale=1.0 / max(1, contributors),
+                    drop_connect_keep=(flags.drop_connect_probability
+                                       if flags.drop_connect else None),
+                    seed=flags.seed, offset=lr_step,
+                    shadow=self.fp.flat_shadow,
+                    momentum=self.flat_momentum, mu=flags.momentum)
+        self.fp.refresh_transposes()
+        self.num_contributors = contributors
 
     # ------------------------------------------------------------------
     def train(self, dataset, max_steps=None):
@@ -303,6 +318,12 @@ class Trainer:
                 writer.add_scalar("Examples/sec", examples_per_sec, self.step)
                 next_summary_time += flags.save_summaries_secs
             sv.maybe_save(self.step, self.checkpoint_payload)
+        if self.mode == "interval":
+            # drain: agree on the total generation count out-of-band, post
+            # anything still owed, apply every remaining aggregation — all
+            # ranks leave with identical parameters
+            for grad, contributors in self.engine.finalize_interval():
+                self._apply_update(grad, contributors)
         if writer is not None:
             writer.close()
         if self.is_chief:

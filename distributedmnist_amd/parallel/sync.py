@@ -6,8 +6,13 @@ four DP synchronization flavors:
              count-renormalize (slowest N-K ranks' grads dropped per step,
              decided from an all-gathered per-rank compute time)
   interval   DP-3: wall-clock timer updates (sync_replicas_optimizer_
-             modified.py:208-215) -> local grad accumulation, rank-0 timer
-             broadcast, all-reduce at interval boundaries only
+             modified.py:208-215) -> FREE-RUNNING workers: local grad
+             accumulation, fire times derived from a once-broadcast shared
+             clock origin (one node = one clock), aggregation posted as an
+             ASYNC all-reduce on a ring of staging buffers and applied on
+             completion — a slow rank delays when an update lands but never
+             stalls another rank's compute, exactly the decoupling of the
+             reference's free-running workers (distributed_train.py:271-288)
   cdf        DP-2: per-worker token barrier + compute-time CDF collection
              (timeout_manager.py:48-70) -> the all-reduce IS the barrier;
              per-rank step timers all-gathered, 'ELAPSED TIMES'/'ITERATION
@@ -42,7 +47,8 @@ class SyncEngine:
                  interval_ms: float | None = None,
                  straggler_timeout_ms: float | None = None,
                  group=None, rank: int = 0, world_size: int = 1,
-                 cdf_log_every: int = 50, cdf_start_tracking: int = 20):
+                 cdf_log_every: int = 50, cdf_start_tracking: int = 20,
+                 wire_dtype: torch.dtype | None = None):
         if mode not in MODES:
             raise ValueError(f"mode {mode!r} not in {MODES}")
         self.mode = mode
@@ -55,13 +61,31 @@ class SyncEngine:
         self.interval_s = (interval_ms or 0.0) / 1000.0
         self.timeout_s = (straggler_timeout_ms / 1000.0) if straggler_timeout_ms else None
         dev = flat_grad.device
+        # wire dtype: bf16 halves the all-reduce payload over xGMI
+        # (6.65 -> 3.33 MB); contributor counts <= world are
+        # exactly representable in bf16 so the flag element rides along fine
+        self.wire_dtype = wire_dtype if wire_dtype != flat_grad.dtype else None
         # flag element rides in the same buffer: [grad..., flag]
-        self._buf = torch.empty(flat_grad.numel() + 1, dtype=flat_grad.dtype, device=dev)
+        buf_dtype = self.wire_dtype or flat_grad.dtype
+        self._buf = torch.empty(flat_grad.numel() + 1, dtype=buf_dtype, device=dev)
         self._accum = None
         self._accum_count = 0
         if mode == "interval":
             self._accum = torch.zeros_like(flat_grad)
-            self._interval_t0 = None
+            self._next_fire = None       # shared-clock fire schedule origin
+            self._ring = [torch.empty_like(self._buf)
+                          for _ in range(self.INTERVAL_RING)]
+            self._free = list(range(self.INTERVAL_RING))
+            self._pending = []           # [(gen, work, ring_idx), ...] FIFO
+            self._deferred = []          # [(grad_clone, contributors), ...]
+            self._gen_posted = 0
+            self.generation = 0          # aggregations APPLIED (global-step
+            #                              equivalent; drives the LR decay)
+            self._gloo = None
+            if world_size > 1 and dist.is_initialized():
+                # CPU side-group for the out-of-band shutdown agreement —
+                # never used on the per-step path
+                self._gloo = dist.new_group(backend="gloo")
         self._ctrl = torch.zeros(1, dtype=torch.float32, device=dev)
         # cdf instrumentation
         self.cdf_log_every = cdf_log_every
@@ -72,6 +96,22 @@ class SyncEngine:
     @property
     def distributed(self) -> bool:
         return self.world > 1 and dist.is_initialized()
+
+    def wire_allreduce(self, tensor: torch.Tensor, start: int = 0):
+        """SUM all-reduce of `tensor` (flat_grad or a slice of it at offset
+        `start`) in the wire dtype.  With a bf16 wire the matching slice of
+        the preallocated staging buffer is used, so disjoint slices can
+        reduce concurrently on different streams (the graphed two-bucket
+        fc/conv split) without aliasing."""
+        if not self.distributed:
+            return
+        if self.wire_dtype is None:
+            dist.all_reduce(tensor, op=dist.ReduceOp.SUM, group=self.group)
+        else:
+            w = self._buf[start:start + tensor.numel()]
+            w.copy_(tensor)
+            dist.all_reduce(w, op=dist.ReduceOp.SUM, group=self.group)
+            tensor.copy_(w)
 
     # ------------------------------------------------------------------
     def step_begin(self, step: int):
@@ -112,8 +152,16 @@ class SyncEngine:
         # host round-trips — the collective IS the barrier
         if self.mode == "full_sync" and self.timeout_s is None:
             if self.distributed:
-                dist.all_reduce(self.flat_grad, op=dist.ReduceOp.SUM,
-                                group=self.group)
+                if self.wire_dtype is None:
+                    dist.all_reduce(self.flat_grad, op=dist.ReduceOp.SUM,
+                                    group=self.group)
+                else:
+                    n = self.flat_grad.numel()
+                    wire = self._buf[:n]
+                    wire.copy_(self.flat_grad)       # fp32 -> bf16 cast
+                    dist.all_reduce(wire, op=dist.ReduceOp.SUM,
+                                    group=self.group)
+                    self.flat_grad.copy_(wire)       # bf16 -> fp32 cast
             return True, self.flat_grad, self.world
 
         contribute = True
@@ -152,38 +200,132 @@ class SyncEngine:
         return True, self.flat_grad, contributors
 
     # ------------------------------------------------------------------
+    INTERVAL_RING = 8   # max in-flight aggregations before backpressure
+
     def _reduce_interval(self, step: int):
-        """DP-3: accumulate locally; all-reduce + apply only when rank 0's
-        wall-clock timer says the interval elapsed (broadcast each step)."""
+        """DP-3 free-running: accumulate locally every step; when the shared
+        fire schedule says an interval boundary passed, POST an async
+        all-reduce of the accumulator and keep stepping.  The update is
+        applied whenever the collective completes — there is no per-step
+        collective, broadcast, or host sync coupling the ranks, so a slow
+        rank cannot stall a fast one between firings (the coupling the
+        reference's interval method exists to avoid,
+        sync_replicas_optimizer_modified.py:208-215)."""
         self._accum += self.flat_grad
         self._accum_count += 1
-        if self._interval_t0 is None:
-            self._interval_t0 = time.time()
-        fire = 0.0
-        if self.rank == 0 and (time.time() - self._interval_t0) >= self.interval_s:
-            fire = 1.0
-        if self.distributed:
-            self._ctrl[0] = fire
-            dist.broadcast(self._ctrl, src=0, group=self.group)
-            fire = float(self._ctrl.item())
-        if fire < 0.5:
-            return False, None, 0
-        # interval fired: average everything accumulated everywhere
+        if self._next_fire is None:
+            # one-time agreement on the fire-schedule origin: all ranks of
+            # one node share a wall clock, so after this single broadcast
+            # every rank computes every future fire time locally
+            t = torch.tensor([time.time() + self.interval_s],
+                             dtype=torch.float64, device=self._ctrl.device)
+            if self.distributed:
+                dist.broadcast(t, src=0, group=self.group)
+            self._next_fire = float(t.item())
+        now = time.time()
+        if self.interval_s <= 0:
+            # degenerate interval: fire-every-step, applied synchronously
+            # (the per-step-averaging semantics the zero-interval cfg means)
+            self._post_generation(step)
+            return self._poll_apply(step, force=True)
+        while now >= self._next_fire:
+            # catch-up loop: a rank that slept through k boundaries owes k
+            # generations (collectives match by posting order on the
+            # communicator, so every rank must post every generation)
+            self._post_generation(step)
+            self._next_fire += self.interval_s
+        return self._poll_apply(step)
+
+    def _post_generation(self, step: int):
         n = self.flat_grad.numel()
-        buf = self._buf
+        if not self._free:
+            # backpressure: ring exhausted (this rank is > RING intervals
+            # ahead of the slowest) — block on the oldest in-flight, move
+            # its result to the deferred-apply queue, recycle its slot
+            gen, work, idx = self._pending.pop(0)
+            if work is not None:
+                work.wait()
+            buf_ = self._ring[idx]
+            contributors = max(1, int(round(float(buf_[n].item()))))
+            self._deferred.append((buf_[:n].clone(), contributors))
+            self._free.append(idx)
+        idx = self._free.pop()
+        buf = self._ring[idx]
         buf[:n].copy_(self._accum)
         buf[n] = float(self._accum_count)
+        work = None
         if self.distributed:
-            dist.all_reduce(buf, op=dist.ReduceOp.SUM, group=self.group)
-        contributors = max(1, int(round(float(buf[n].item()))))
-        self.flat_grad.copy_(buf[:n])
+            work = dist.all_reduce(buf, op=dist.ReduceOp.SUM,
+                                   group=self.group, async_op=True)
+        self._pending.append((self._gen_posted, work, idx))
+        self._gen_posted += 1
         self._accum.zero_()
         self._accum_count = 0
-        self._interval_t0 = time.time()
+
+    def _poll_apply(self, step: int, force: bool = False):
+        """Apply the oldest completed aggregation, if any (at most one per
+        call; later completions surface on subsequent steps)."""
+        if self._deferred:
+            grad, contributors = self._deferred.pop(0)
+            self.flat_grad.copy_(grad)
+            self.generation += 1
+            return True, self.flat_grad, contributors
+        if not self._pending:
+            return False, None, 0
+        gen, work, idx = self._pending[0]
+        if work is not None:
+            if force:
+                work.wait()
+            elif not work.is_completed():
+                return False, None, 0
+        self._pending.pop(0)
+        n = self.flat_grad.numel()
+        buf = self._ring[idx]
+        contributors = max(1, int(round(float(buf[n].item()))))
+        self.flat_grad.copy_(buf[:n])
+        self._free.append(idx)
+        self.generation += 1
         if self.rank == 0:
-            log.info("Interval update fired at step %d (averaged %d grads)",
-                     step, contributors)
+            log.info("Interval update %d applied at step %d (averaged %d "
+                     "grads)", gen, step, contributors)
         return True, self.flat_grad, contributors
+
+    def finalize_interval(self):
+        """Shutdown drain (two-phase, over the CPU side-group, OUT of band
+        of the training communicator):
+
+        Phase 1 — while waiting for every rank to reach finalize (async
+        gloo barrier), KEEP SERVICING THE FIRE SCHEDULE: a rank that
+        stopped stepping must go on posting generations on time, or a
+        slower rank still in its loop would block in ring backpressure
+        waiting for posts that never come.
+
+        Phase 2 — everyone is here and nobody is blocked on the training
+        communicator, so a blocking MAX of the posted-generation counts is
+        safe; post anything still owed, then yield every remaining
+        aggregated gradient in order for the caller to apply."""
+        if self.mode != "interval":
+            return
+        if self.distributed and self._gloo is not None:
+            flag = torch.zeros(1, dtype=torch.int64)
+            barrier = dist.all_reduce(flag, group=self._gloo, async_op=True)
+            while not barrier.is_completed():
+                now = time.time()
+                while (self._next_fire is not None and self.interval_s > 0
+                       and now >= self._next_fire):
+                    self._post_generation(step=-1)
+                    self._next_fire += self.interval_s
+                time.sleep(0.002)
+            g = torch.tensor([self._gen_posted], dtype=torch.int64)
+            dist.all_reduce(g, op=dist.ReduceOp.MAX, group=self._gloo)
+            target = int(g.item())
+            while self._gen_posted < target:
+                self._post_generation(step=-1)
+        while self._pending or self._deferred:
+            applied, grad, contributors = self._poll_apply(step=-1, force=True)
+            if not applied:
+                break
+            yield grad, contributors
 
     # ------------------------------------------------------------------
     def _maybe_log_cdf(self, step: int):
