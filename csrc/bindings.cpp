@@ -593,6 +593,21 @@ void step_advance(torch::Tensor step_dev, torch::Tensor lr_scale_dev,
                       cur_stream());
 }
 
+void grad_mask(torch::Tensor g, double keep, int64_t seed, int64_t step,
+               int64_t rank, int64_t base,
+               c10::optional<torch::Tensor> step_dev) {
+  CHECK_CUDA(g); CHECK_F32(g); CHECK_CONTIG(g);
+  TORCH_CHECK(base % 4 == 0,
+              "grad_mask: slice base must be 16B-aligned so slice-wise "
+              "masks compose to the whole-buffer mask");
+  const long* sd = nullptr;
+  if (step_dev.has_value() && step_dev->defined())
+    sd = step_dev->data_ptr<long>();
+  launch_grad_mask(g.data_ptr<float>(), g.numel(), base, (float)keep,
+                   (uint64_t)seed, (uint64_t)step, (uint64_t)rank, sd,
+                   cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -631,6 +646,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("momentum") = c10::nullopt, py::arg("mu") = 0.0);
   m.def("step_advance", &step_advance,
         "device-side staircase LR + step increment (inside the graph)");
+  m.def("grad_mask", &grad_mask,
+        "per-rank pre-aggregation drop-connect mask (in-place, philox; "
+        "reference distributed_train.py:194-203)",
+        py::arg("g"), py::arg("keep"), py::arg("seed"), py::arg("step"),
+        py::arg("rank"), py::arg("base") = 0,
+        py::arg("step_dev") = c10::nullopt);
   m.def("linear_act_fwd_dev", &linear_act_fwd_dev,
         "linear fwd with device-side dropout offset",
         py::arg("x"), py::arg("w"), py::arg("b"), py::arg("relu"),

@@ -60,6 +60,9 @@ void launch_sgd_step_dev(float* master, const float* grad,
                          const float* lr_scale_dev, float dc_keep,
                          uint64_t seed, const long* offset_dev,
                          float* momentum, float mu, hipStream_t);
+void launch_grad_mask(float* g, long n, long base, float keep, uint64_t seed,
+                      uint64_t step, uint64_t rank, const long* step_dev,
+                      hipStream_t s);
 void launch_step_advance(long* step_dev, float* lr_scale_dev, float lr0,
                          float decay, int decay_steps, float inv_contrib,
                          hipStream_t);

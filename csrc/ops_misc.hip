@@ -355,6 +355,60 @@ void launch_sgd_step_dev(float* master, const float* grad,
                      lr_scale_dev, offset_dev, momentum, mu);
 }
 
+// ---------------------------------------------------------------------------
+// Per-rank PRE-AGGREGATION drop-connect mask (reference
+// distributed_train.py:194-203: each worker masks its OWN gradient before
+// the aggregation; masks are rank-distinct).  In-place Bernoulli(keep)
+// philox mask of a flat fp32 gradient (or a slice of it at element offset
+// `base`, so masking the fc/conv bucket slices separately reproduces
+// exactly the whole-buffer mask).  Stream = (seed ^ SALT, step*2^20+rank):
+// disjoint from both the dropout stream (offset=step) and the
+// post-aggregation drop-connect stream in sgd_step.  `step_dev` (optional)
+// makes it hipGraph-capturable.  Mask, NO rescale, per reference :414-416.
+// ---------------------------------------------------------------------------
+#define DMNIST_DC_SALT 0x9D5AD0C5u
+
+extern "C" __global__ __launch_bounds__(256)
+void grad_mask_kernel(float* g, long n, long base, float keep,
+                      uint64_t seed, uint64_t step, uint64_t rank,
+                      const long* step_dev) {
+  if (step_dev) step = (uint64_t)*step_dev;
+  uint64_t offset = (step << 20) + rank + 1;
+  seed ^= (uint64_t)DMNIST_DC_SALT;
+  long i = (long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  const float s = 1.0f / 4294967296.0f;
+  for (; i * 4 < n; i += stride) {
+    long b = i * 4;
+    // philox counter indexed by ABSOLUTE flat position so slice-wise
+    // launches compose to the same mask as one whole-buffer launch;
+    // requires base % 4 == 0 (checked host-side)
+    Philox4 ph = philox4x32(seed, offset, (uint64_t)((base + b) >> 2));
+    if (b + 3 < n) {
+      float4 v = *reinterpret_cast<const float4*>(g + b);
+      v.x *= (ph.x * s < keep) ? 1.f : 0.f;
+      v.y *= (ph.y * s < keep) ? 1.f : 0.f;
+      v.z *= (ph.z * s < keep) ? 1.f : 0.f;
+      v.w *= (ph.w * s < keep) ? 1.f : 0.f;
+      *reinterpret_cast<float4*>(g + b) = v;
+    } else {
+      float u[4] = {ph.x * s, ph.y * s, ph.z * s, ph.w * s};
+      for (long j = b; j < n; ++j)
+        g[j] *= (u[j - b] < keep) ? 1.f : 0.f;
+    }
+  }
+}
+
+void launch_grad_mask(float* g, long n, long base, float keep, uint64_t seed,
+                      uint64_t step, uint64_t rank, const long* step_dev,
+                      hipStream_t s) {
+  long groups = (n + 3) / 4;
+  int blocks = (int)min((long)2048, (groups + 255) / 256);
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(grad_mask_kernel, dim3(blocks), dim3(256), 0, s, g, n,
+                     base, keep, seed, step, rank, step_dev);
+}
+
 void launch_step_advance(long* step_dev, float* lr_scale_dev, float lr0,
                          float decay, int decay_steps, float inv_contrib,
                          hipStream_t s) {

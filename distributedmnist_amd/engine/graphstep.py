@@ -57,8 +57,13 @@ class GraphedStep:
         self.decay_steps = max(1, int(num_batches * flags.num_epochs_per_decay
                                       / max(1, t.engine.K)))
         self.inv_contrib = 1.0 / max(1, t.world)
-        self.dc_keep = (flags.drop_connect_probability if flags.drop_connect
-                        else -1.0)
+        # pre-aggregation masks run inside the capture (per-rank, keyed on
+        # step_dev); only the legacy --drop_connect_post form rides in the
+        # SGD tail kernel
+        self.dc_pre = t._dc_pre
+        self.dc_prob = flags.drop_connect_probability
+        self.dc_keep = (flags.drop_connect_probability
+                        if (flags.drop_connect and not t._dc_pre) else -1.0)
         self._ext = ext
         # hand-scheduled two-stream step for LeNet (dW off the dX chain);
         # autograd body otherwise
@@ -79,6 +84,12 @@ class GraphedStep:
         if self.split and self._fused is not None:
             # split scope: the bucketed all-reduce must stay OUT of the
             # captured region (run eagerly after replay instead)
+            self._fused.overlap_allreduce = False
+        if self.dc_pre and self._fused is not None:
+            # pre-aggregation masks must precede the reduce: the fused
+            # step's in-body fc all-reduce would fire before the mask, so
+            # fall back to the post-body whole-bucket reduce in _body
+            # (only reachable via --hip_graph full at world > 1)
             self._fused.overlap_allreduce = False
 
         self._prime(t.step)
@@ -128,6 +139,15 @@ class GraphedStep:
                     t.fp.flat_grad.zero_()
                     loss, correct = self._fused.stage_fc(
                         self.static_x, self.static_y, self.step_dev)
+                    if self.dc_pre:
+                        # fc grads final at graph A's boundary: mask them
+                        # BEFORE the eager fc-slice all-reduce between the
+                        # graphs (base = slice offset so the slice-wise
+                        # masks equal one whole-buffer mask)
+                        ext.grad_mask(self.fc_slice, self.dc_prob,
+                                      flags.seed, 0, t.rank,
+                                      base=self.fc_offset,
+                                      step_dev=self.step_dev)
                     self.static_loss = loss.detach()
                     acc = correct / self.static_x.shape[0]
                     self.static_acc = acc.detach()
@@ -135,6 +155,10 @@ class GraphedStep:
                 # graph A allocated (the stage_fc stash)
                 with torch.cuda.graph(self.graph_b, pool=self.graph.pool()):
                     self._fused.stage_conv()
+                    if self.dc_pre:
+                        ext.grad_mask(self.conv_slice, self.dc_prob,
+                                      flags.seed, 0, t.rank, base=0,
+                                      step_dev=self.step_dev)
             else:
                 with torch.cuda.graph(self.graph):
                     if self.split:
@@ -201,6 +225,10 @@ class GraphedStep:
             logits = t.model(self.static_x, train=True)
             loss, acc = t.model.loss_and_accuracy(logits, self.static_y)
             loss.backward()
+        if self.dc_pre:
+            # grads complete, reduce not yet issued: per-rank mask here
+            self._ext.grad_mask(fp.flat_grad, self.dc_prob, t.flags.seed,
+                                0, t.rank, base=0, step_dev=self.step_dev)
         self.static_loss = loss.detach()
         self.static_acc = acc.detach()
         return fused_reduced

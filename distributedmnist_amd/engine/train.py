@@ -109,6 +109,11 @@ class Trainer:
                         if getattr(flags, "grad_dtype", "fp32") == "bf16"
                         else None))
         self.timer = StepTimer(self.device)
+        # drop-connect placement: pre-aggregation per-rank masks are the
+        # reference's estimator (distributed_train.py:194-203);
+        # --drop_connect_post keeps the round-1 shared post-aggregation form
+        self._dc_pre = (flags.drop_connect
+                        and not getattr(flags, "drop_connect_post", False))
         self.is_chief = rank == 0
         self.step = 0
         self.num_contributors = world
@@ -219,6 +224,9 @@ class Trainer:
         loss, acc = self.model.loss_and_accuracy(logits, labels)
         loss.backward()
         self.fp.fix_grad_views()
+        if self._dc_pre:
+            Fx.grad_mask(self.fp.flat_grad, flags.drop_connect_probability,
+                         flags.seed, self.step, self.rank)
         compute_time = self.timer.stop() if timed else 0.0
         applied, grad, contributors = self.engine.reduce(self.step, compute_time)
         if applied:
@@ -241,7 +249,8 @@ class Trainer:
         Fx.sgd_step(self.fp.flat_master, grad, lr,
                     grad_scale=1.0 / max(1, contributors),
                     drop_connect_keep=(flags.drop_connect_probability
-                                       if flags.drop_connect else None),
+                                       if (flags.drop_connect
+                                           and not self._dc_pre) else None),
                     seed=flags.seed, offset=lr_step,
                     shadow=self.fp.flat_shadow,
                     momentum=self.flat_momentum, mu=flags.momentum)

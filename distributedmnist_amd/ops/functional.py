@@ -171,6 +171,24 @@ def softmax_xent(logits, labels):
     return SoftmaxXentFn.apply(logits, labels)
 
 
+def grad_mask(grad: torch.Tensor, keep: float, seed: int, step: int,
+              rank: int, step_dev=None, base: int = 0):
+    """Per-rank PRE-AGGREGATION drop-connect: in-place Bernoulli(keep) mask
+    of this rank's local gradient BEFORE the all-reduce, seeded by
+    (seed, step, rank) so every worker's mask is distinct — the reference's
+    estimator (distributed_train.py:194-203: masks applied per worker, then
+    averaged).  Mask, NO rescale (:414-416)."""
+    if grad.is_cuda:
+        _C.ext().grad_mask(grad, float(keep), int(seed), int(step),
+                           int(rank), base=int(base), step_dev=step_dev)
+    else:
+        gen = torch.Generator(device="cpu")
+        gen.manual_seed(((seed ^ 0x9D5AD0C5) * 0x9E3779B97F4A7C15
+                         + (step << 20) + rank + 1) % (2**63))
+        mask = (torch.rand(grad.shape, generator=gen) < keep).float()
+        grad.mul_(mask)
+
+
 def sgd_step(master: torch.Tensor, grad: torch.Tensor, lr: float,
              grad_scale: float = 1.0, drop_connect_keep=None,
              seed: int = 0, offset: int = 0, shadow=None,

@@ -187,3 +187,32 @@ def test_trainer_momentum_flag(tmp_path):
     ds = make_dataset(flags, 0, 1, t.device, t.compute_dtype)
     t.train(ds)
     assert float(t.flat_momentum.abs().sum()) > 0
+
+
+def test_grad_mask_cpu_deterministic_and_rank_distinct():
+    """CPU reference of the pre-aggregation drop-connect mask: keep-rate,
+    determinism in (seed, step, rank), distinct streams across ranks."""
+    import torch
+    from distributedmnist_amd.ops import functional as Fx
+    g = torch.ones(100_000)
+    Fx.grad_mask(g, 0.9, seed=66478, step=3, rank=0)
+    kept = float((g != 0).float().mean())
+    assert 0.89 < kept < 0.91
+    g2 = torch.ones(100_000)
+    Fx.grad_mask(g2, 0.9, seed=66478, step=3, rank=0)
+    assert torch.equal(g, g2)
+    g3 = torch.ones(100_000)
+    Fx.grad_mask(g3, 0.9, seed=66478, step=3, rank=1)
+    assert not torch.equal(g, g3)
+
+
+def test_trainer_drop_connect_pre_is_default(tmp_path):
+    from distributedmnist_amd.engine.train import Trainer
+    from distributedmnist_amd.utils.flags import build_train_parser
+    mk = lambda *extra: build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", str(tmp_path / "t"),
+         "--batch_size", "8", "--model", "mlp", "--device", "cpu",
+         "--save_interval_secs", "100000"] + list(extra))
+    assert Trainer(mk("--drop_connect"))._dc_pre
+    assert not Trainer(mk("--drop_connect", "--drop_connect_post"))._dc_pre
+    assert not Trainer(mk())._dc_pre

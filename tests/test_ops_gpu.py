@@ -197,6 +197,37 @@ def test_softmax_xent_gpu(ext):
     assert_close_bf16(dl, dl_ref, atol=2e-4)
 
 
+def test_grad_mask_gpu(ext):
+    """Per-rank pre-aggregation drop-connect kernel: keep-rate, determinism,
+    rank/step-distinct streams, and slice-wise composability (masking the
+    conv/fc bucket slices separately must equal one whole-buffer mask)."""
+    n = 1_000_000
+    g = torch.ones(n).cuda()
+    ext.grad_mask(g, 0.9, 66478, 5, 0)
+    kept = float((g != 0).float().mean())
+    assert 0.895 < kept < 0.905, kept
+    # deterministic for the same (seed, step, rank)
+    g2 = torch.ones(n).cuda()
+    ext.grad_mask(g2, 0.9, 66478, 5, 0)
+    assert torch.equal(g, g2)
+    # distinct across ranks and across steps
+    for step, rank in ((5, 1), (6, 0)):
+        g3 = torch.ones(n).cuda()
+        ext.grad_mask(g3, 0.9, 66478, step, rank)
+        assert not torch.equal(g, g3), (step, rank)
+    # slice composability at a 16B-aligned split (the fc bucket offset)
+    split = 52096
+    g4 = torch.ones(n).cuda()
+    ext.grad_mask(g4[:split], 0.9, 66478, 5, 0, base=0)
+    ext.grad_mask(g4[split:], 0.9, 66478, 5, 0, base=split)
+    assert torch.equal(g, g4)
+    # step_dev variant matches the host-arg variant
+    g5 = torch.ones(n).cuda()
+    sd = torch.tensor([5], dtype=torch.int64).cuda()
+    ext.grad_mask(g5, 0.9, 66478, 0, 0, step_dev=sd)
+    assert torch.equal(g, g5)
+
+
 def test_sgd_step_gpu(ext):
     torch.manual_seed(6)
     n = 1_000_003
