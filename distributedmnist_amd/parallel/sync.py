@@ -274,10 +274,13 @@ class SyncEngine:
             return False, None, 0
         gen, work, idx = self._pending[0]
         if work is not None:
-            if force:
-                work.wait()
-            elif not work.is_completed():
+            if not force and not work.is_completed():
                 return False, None, 0
+            # NCCL: wait() inserts the stream dependency for the reads
+            # below (host-blocking only for gloo); the collective itself
+            # ran on the backend's internal stream, so pending aggregations
+            # never serialize this rank's compute stream
+            work.wait()
         self._pending.pop(0)
         n = self.flat_grad.numel()
         buf = self._ring[idx]

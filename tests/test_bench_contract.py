@@ -37,14 +37,12 @@ def test_bench_single_process():
 @pytest.mark.timeout(300)
 def test_bench_mode_is_real_cdf():
     """VERDICT round-1 task 1: --mode must map onto the trainer flags, not
-    just label the JSON.  A cdf bench demonstrably runs the cdf engine:
-    the scraper-contract 'ELAPSED TIMES' lines appear and the JSON reports
-    the mode that ran."""
-    env = dict(os.environ, DMNIST_LOG_STDERR="1")
+    just label the JSON.  A cdf bench demonstrably runs the cdf engine and
+    the JSON reports the mode that ran (trainer.mode, not the CLI arg)."""
     out = subprocess.run(
         [sys.executable, os.path.join(ROOT, "bench.py"), "--steps", "60",
          "--warmup", "1", "--batch_size", "32", "--mode", "cdf"],
-        capture_output=True, text=True, timeout=240, cwd=ROOT, env=env)
+        capture_output=True, text=True, timeout=240, cwd=ROOT)
     assert out.returncode == 0, out.stderr[-2000:]
     j = _last_json_line(out.stdout)
     assert j["config"]["mode"] == "cdf"
