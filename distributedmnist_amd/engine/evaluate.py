@@ -61,10 +61,13 @@ def evaluate(dataset, flags, writer=None):
                     for name in model.PARAM_ORDER:
                         getattr(model, name).copy_(
                             payload["model_state"][name].to(device))
-                    if dtype != torch.float32:
-                        model.shadows = {
-                            n: getattr(model, n).detach().to(dtype)
-                            for n in model.PARAM_ORDER if n.endswith("_w")}
+                # first-class GPU eval: FlatParams builds the bf16 shadows
+                # AND the pre-transposed shadow_T copies, so evaluation
+                # runs the SAME transposed-GEMM kernel path as training
+                # (round-1 built shadows ad hoc without shadows_T and fell
+                # onto the non-transposed GEMMs)
+                from ..parallel import FlatParams
+                FlatParams(model, device=device, compute_dtype=dtype)
                 print('Succesfully loaded model from %s at step=%s.' %
                       (Supervisor.latest_checkpoint(flags.checkpoint_dir)[1], step),
                       flush=True)
