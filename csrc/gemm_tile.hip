@@ -32,7 +32,16 @@ enum BMode {
   B_CONV_DX_W = 2, // Bmat[k=(khkw,co)][n=ci] = w[((khkw)*Cin+ci)*Cout + co]
 };
 enum Epi { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_RELU = 2, EPI_BIAS_RELU_DROP = 3,
-           EPI_POOL = 4 };
+           EPI_POOL = 4, EPI_UNPOOL = 5 };
+// EPI_UNPOOL: the output row m is a batch index and column n a pooled
+// feature (ho, wo, c) of a [CHo][CWo][Cout] activation; instead of writing
+// C[m][n], the epilogue routes the value through the maxpool-2x2 backward
+// (relu-mask by ypool > 0, scatter to the argmax position, 3 zeros) straight
+// into the pre-pool gradient dact [CB][2*CHo][2*CWo][Cout], and accumulates
+// the conv bias grad db[c] — this FUSES the standalone pool_bwd_scatter
+// kernel into the producing dX GEMM (reference backward chain
+// mnist.py:115-127), removing one critical-path kernel + the pooled-grad
+// round trip.
 enum OutKind { OUT_BF16 = 0, OUT_F32_ATOMIC = 1, OUT_F32_SLICES = 2 };
 
 #define NTHREADS 256
@@ -255,6 +264,9 @@ void gemm_tile_kernel(GemmParams p) {
   static_assert(CHA >= 1 && CHB >= 1, "tile too small for 256 threads");
   __shared__ __align__(16) ushort_t As[DB][BM][LDK];
   __shared__ __align__(16) ushort_t Bs[DB][BN][LDK];
+  // EPI_UNPOOL: per-column bias-grad partials, flushed once per block
+  __shared__ float dbred[EPI == EPI_UNPOOL ? BN : 1];
+  if (EPI == EPI_UNPOOL && threadIdx.x < BN) dbred[threadIdx.x] = 0.f;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -402,12 +414,46 @@ void gemm_tile_kernel(GemmParams p) {
 epilogue:
   const int frow = (lane >> 4) * 4;  // C/D: row=(lane>>4)*4+reg, col=lane&15
   const int fcol = lane & 15;
+  float dbloc[EPI == EPI_UNPOOL ? NI : 1];
+  if (EPI == EPI_UNPOOL)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) dbloc[ni] = 0.f;
 #pragma unroll
   for (int mi = 0; mi < MI; ++mi) {
 #pragma unroll
     for (int ni = 0; ni < NI; ++ni) {
       int gc = n0 + wc * WN + ni * 16 + fcol;
       if (gc >= p.N) continue;
+      if (EPI == EPI_UNPOOL) {
+        // n = (ho*CWo + wo)*Cout + c in exactly the pooled tensor layout,
+        // so ypool/amax share the [m][n] offset; dact is the 2x up-scaled
+        // grid
+        int c = gc % p.Cout;
+        int pix = gc / p.Cout;
+        int wo = pix % p.CWo, ho = pix / p.CWo;
+        ushort_t* dact = reinterpret_cast<ushort_t*>(p.C);
+        const int H2 = p.CHo * 2, W2 = p.CWo * 2;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int gr = m0 + wr * WM + mi * 16 + frow + r;
+          if (gr >= p.M) continue;
+          size_t qoff = (size_t)gr * p.N + gc;
+          float g = acc[mi][ni][r];
+          if (!(bf2f(p.ypool[qoff]) > 0.f)) g = 0.f;
+          int pos = p.amax[qoff];
+          dbloc[ni] += g;
+          ushort_t gb = f2bf(g);
+#pragma unroll
+          for (int rr = 0; rr < 2; ++rr)
+#pragma unroll
+            for (int cc2 = 0; cc2 < 2; ++cc2) {
+              size_t o = (((size_t)gr * H2 + ho * 2 + rr) * W2 + wo * 2 +
+                          cc2) * p.Cout + c;
+              dact[o] = (pos == rr * 2 + cc2) ? gb : (ushort_t)0;
+            }
+        }
+        continue;
+      }
       float bias_v = (EPI != EPI_NONE && p.bias) ? p.bias[gc] : 0.f;
       if (EPI == EPI_POOL) {
         // rows (q*4 .. q*4+3) of this lane group are one pool window
@@ -454,6 +500,21 @@ epilogue:
       }
     }
   }
+  if (EPI == EPI_UNPOOL) {
+    // bias-grad flush: lane partials -> LDS per-column -> one global
+    // atomicAdd per block column (channel = col % Cout)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) {
+      int colb = wc * WN + ni * 16 + fcol;
+      if (dbloc[ni] != 0.f) atomicAdd(&dbred[colb], dbloc[ni]);
+    }
+    __syncthreads();
+    if (p.db && (int)threadIdx.x < BN) {
+      int gc = n0 + (int)threadIdx.x;
+      float v = dbred[threadIdx.x];
+      if (gc < p.N && v != 0.f) atomicAdd(&p.db[gc % p.Cout], v);
+    }
+  }
 }
 
 // ---- host-side launch table ----------------------------------------------
@@ -485,6 +546,9 @@ GEMM_ENTRY(gemm_fwd_drop_128, 128, 128, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF1
 GEMM_ENTRY(gemm_fwd_drop_64, 64, 64, A_N, B_KMAJ, EPI_BIAS_RELU_DROP, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dx_128, 128, 128, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dx_64, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
+// dX fused with maxpool-2x2 backward (fc1 dX -> dact2 directly + conv2 db)
+GEMM_ENTRY(gemm_dx_unpool_128, 128, 128, A_N, B_NMAJ, EPI_UNPOOL, OUT_BF16, 1)
+GEMM_ENTRY(gemm_dx_unpool_64, 64, 64, A_N, B_NMAJ, EPI_UNPOOL, OUT_BF16, 1)
 GEMM_ENTRY_SWZ(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 2)
 GEMM_ENTRY_SWZ(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 2)
 // B-transposed (pre-transposed weight) fwd variants: ldb = K, vector staging

@@ -17,6 +17,8 @@ struct GemmParams {
   uint64_t seed, offset;
   const long* offset_dev;  // when set, RNG offset is read from device memory
                            // (hipGraph replay: host args are frozen)
+  const unsigned short* ypool;  // EPI_UNPOOL: pooled activations (relu mask)
+  float* db;                    // EPI_UNPOOL: conv bias-grad accumulator
 };
 
 // gemm_tile.hip — implicit-GEMM MFMA entry points
@@ -28,6 +30,8 @@ void gemm_fwd_drop_128(const GemmParams&, hipStream_t);
 void gemm_fwd_drop_64(const GemmParams&, hipStream_t);
 void gemm_dx_128(const GemmParams&, hipStream_t);
 void gemm_dx_64(const GemmParams&, hipStream_t);
+void gemm_dx_unpool_128(const GemmParams&, hipStream_t);
+void gemm_dx_unpool_64(const GemmParams&, hipStream_t);
 void gemm_dw_128(const GemmParams&, hipStream_t);
 void gemm_dw_64(const GemmParams&, hipStream_t);
 void conv_fwd_pool(const GemmParams&, hipStream_t);
@@ -60,6 +64,10 @@ void launch_sgd_step_dev(float* master, const float* grad,
                          const float* lr_scale_dev, float dc_keep,
                          uint64_t seed, const long* offset_dev,
                          float* momentum, float mu, hipStream_t);
+void launch_conv1_dw_pooled(const unsigned short* x, const unsigned short* dyp,
+                            const unsigned short* yp, const uint8_t* am,
+                            float* dw, float* db, int NB, int H, int W,
+                            int Cout, hipStream_t s);
 void launch_grad_mask(float* g, long n, long base, float keep, uint64_t seed,
                       uint64_t step, uint64_t rank, const long* step_dev,
                       hipStream_t s);

@@ -697,6 +697,130 @@ void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
               red[i / Cout][i % Cout]);
 }
 
+// ---------------------------------------------------------------------------
+// conv1 dW + db consuming the POOLED gradient directly (pool backward fused
+// at the consumer).  dact1 = scatter(dxc, am1) is 75% structural zeros and
+// was the largest intermediate of the backward (B*28*28*32); this kernel
+// never materializes it: each lane decodes (dy_pooled, y_pooled, argmax)
+// into the two dense-row dot2 pair streams the direct kernel walked, so
+// the pool_bwd_scatter kernel AND the dact1 write+read round trip
+// (51 MB @ B=1024, 410 MB @ 8192) disappear from the critical path.
+// db[c] = sum of masked pooled dy (what the scatter kernel used to flush).
+// Same slab/window/dot2 math as conv1_dw_direct_kernel above.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256)
+void conv1_dw_pooled_kernel(const ushort_t* x, const ushort_t* dyp,
+                            const ushort_t* yp, const uint8_t* am,
+                            float* dw, float* db, int NB, int H, int W,
+                            int Cout, int G) {
+  __shared__ ushort_t slab[32][32];
+  __shared__ float red[26][32];  // 25 taps + db row
+  const int tid = threadIdx.x;
+  const int co = tid & 31;
+  const int hw = tid >> 5;  // half-wave 0..7: pooled rows hw, hw+8
+  const int Hp = H / 2, Wp = W / 2;
+  float acc[25];
+#pragma unroll
+  for (int t = 0; t < 25; ++t) acc[t] = 0.f;
+  float dbacc = 0.f;
+
+  const int img0 = blockIdx.x * G;
+  for (int g = 0; g < G; ++g) {
+    const int img = img0 + g;
+    if (img >= NB) break;  // uniform across the block
+    const ushort_t* xi = x + (size_t)img * H * W;
+    for (int i = tid; i < 32 * 32; i += 256) {
+      int xx = i % 32, yy = i / 32;
+      int sy = yy - 2, sx = xx - 2;
+      ushort_t v = 0;
+      if (sy >= 0 && sy < H && sx >= 0 && sx < W) v = xi[(size_t)sy * W + sx];
+      slab[yy][xx] = v;
+    }
+    __syncthreads();
+    if (co < Cout) {
+      const size_t ib = ((size_t)img * Hp) * Wp * Cout + co;
+      for (int rq = hw; rq < Hp; rq += 8) {
+        // decode one pooled row: packed (bf16 value | argmax<<16) per qx —
+        // 3 small strided streams instead of the dense dact row walk
+        unsigned pk[14];
+#pragma unroll
+        for (int qx = 0; qx < 14; ++qx) {
+          size_t o = ib + ((size_t)rq * Wp + qx) * Cout;
+          float gv = bf2f(dyp[o]);
+          if (!(bf2f(yp[o]) > 0.f)) gv = 0.f;
+          pk[qx] = (unsigned)f2bf(gv) | ((unsigned)am[o] << 16);
+        }
+#pragma unroll
+        for (int ry = 0; ry < 2; ++ry) {
+          const int r = 2 * rq + ry;
+          unsigned win[5][3];
+#pragma unroll
+          for (int kr = 0; kr < 5; ++kr) {
+            win[kr][0] = *reinterpret_cast<const unsigned*>(&slab[r + kr][0]);
+            win[kr][1] = *reinterpret_cast<const unsigned*>(&slab[r + kr][2]);
+            win[kr][2] = *reinterpret_cast<const unsigned*>(&slab[r + kr][4]);
+          }
+          for (int j = 0; j < 14; ++j) {  // dense pair c = 2j
+            unsigned pkj = pk[j];
+            unsigned gb = pkj & 0xffffu;
+            int pos = (int)(pkj >> 16);
+            unsigned gp = ((pos >> 1) == ry)
+                              ? ((pos & 1) ? (gb << 16) : gb)
+                              : 0u;
+#pragma unroll
+            for (int kh = 0; kh < 5; ++kh) {
+              unsigned o0 = __builtin_amdgcn_alignbit(win[kh][1], win[kh][0], 16);
+              unsigned o1 = __builtin_amdgcn_alignbit(win[kh][2], win[kh][1], 16);
+              acc[kh * 5 + 0] = dot2bf(win[kh][0], gp, acc[kh * 5 + 0]);
+              acc[kh * 5 + 1] = dot2bf(o0, gp, acc[kh * 5 + 1]);
+              acc[kh * 5 + 2] = dot2bf(win[kh][1], gp, acc[kh * 5 + 2]);
+              acc[kh * 5 + 3] = dot2bf(o1, gp, acc[kh * 5 + 3]);
+              acc[kh * 5 + 4] = dot2bf(win[kh][2], gp, acc[kh * 5 + 4]);
+            }
+            int nc = 2 * j + 6 <= 30 ? 2 * j + 6 : 30;
+#pragma unroll
+            for (int kr = 0; kr < 5; ++kr) {
+              win[kr][0] = win[kr][1];
+              win[kr][1] = win[kr][2];
+              win[kr][2] = *reinterpret_cast<const unsigned*>(&slab[r + kr][nc]);
+            }
+          }
+          if (ry == 0) {
+            // db once per pooled row (value independent of ry)
+#pragma unroll
+            for (int j = 0; j < 14; ++j) dbacc += bf2f((ushort_t)(pk[j] & 0xffffu));
+          }
+        }
+      }
+    }
+    __syncthreads();  // slab re-staged next image
+  }
+
+  // block reduction: 8 half-waves -> LDS, then one global flush
+  for (int i = tid; i < 26 * 32; i += 256) red[i / 32][i % 32] = 0.f;
+  __syncthreads();
+  if (co < Cout) {
+#pragma unroll
+    for (int t = 0; t < 25; ++t) atomicAdd(&red[t][co], acc[t]);
+    atomicAdd(&red[25][co], dbacc);
+  }
+  __syncthreads();
+  for (int i = tid; i < 25 * Cout; i += 256)
+    atomicAdd(&dw[(size_t)(i / Cout) * Cout + (i % Cout)],
+              red[i / Cout][i % Cout]);
+  if (db && tid < Cout) atomicAdd(&db[tid], red[25][tid]);
+}
+
+void launch_conv1_dw_pooled(const unsigned short* x, const unsigned short* dyp,
+                            const unsigned short* yp, const uint8_t* am,
+                            float* dw, float* db, int NB, int H, int W,
+                            int Cout, hipStream_t s) {
+  int G = NB >= 8192 ? 8 : (NB >= 2048 ? NB / 1024 : 1);
+  int blocks = (NB + G - 1) / G;
+  hipLaunchKernelGGL(conv1_dw_pooled_kernel, dim3(blocks), dim3(256), 0, s,
+                     x, dyp, yp, am, dw, db, NB, H, W, Cout, G);
+}
+
 void launch_conv1_dw_direct(const unsigned short* x,
                             const unsigned short* dact, float* dw, int NB,
                             int H, int W, int Cout, hipStream_t s) {

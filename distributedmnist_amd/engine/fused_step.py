@@ -90,16 +90,18 @@ class FusedLeNetStep:
                 # ordered by wait_stream) is final here
                 self.t.engine.wire_allreduce(self.fc_slice,
                                              start=self.fc_offset)
-        dx1 = ext.linear_dx(dyeff1, sh["fc1_w"]).view(B, 7, 7, 64)
-
-        dact2 = ext.pool_scatter(dx1, y2, am2, gv("conv2_b"), 14, 14)
+        # fc1 dX fused with pool2 backward: dact2 + conv2 db come straight
+        # out of the GEMM epilogue (no pool_scatter kernel on the chain)
+        dact2 = ext.linear_dx_unpool(dyeff1, sh["fc1_w"], y2, am2,
+                                     gv("conv2_b"), 7, 7, 64)
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
             ext.conv_dw_into(y1, dact2, gv("conv2_w"))
         dxc = ext.conv_dx(dact2, sh["conv2_w"], 32)
 
-        dact1 = ext.pool_scatter(dxc, y1, am1, gv("conv1_b"), 28, 28)
-        ext.conv_dw_into(x, dact1, gv("conv1_w"))
+        # conv1 dW+db consume the POOLED gradient (pool1 backward fused in
+        # the consumer; the 4x-size dact1 is never materialized)
+        ext.conv1_dw_pooled(x, dxc, y1, am1, gv("conv1_w"), gv("conv1_b"))
 
         s0.wait_stream(s1)
         if self.overlap_allreduce:
@@ -147,9 +149,12 @@ class FusedLeNetStep:
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
-        dx1 = ext.linear_dx(dyeff1, sh["fc1_w"]).view(B, 7, 7, 64)
+        # fused dX+pool2-backward: conv2_b lands here (conv slice — reduced
+        # after graph B, so computing it early is safe)
+        dact2 = ext.linear_dx_unpool(dyeff1, sh["fc1_w"], y2, am2,
+                                     gv("conv2_b"), 7, 7, 64)
         s0.wait_stream(s1)  # graph A boundary: fc grads complete
-        self._stash = (x, y1, y2, am1, am2, dx1)
+        self._stash = (x, y1, am1, dact2)
         self._keep = (a1, h2, dyeff1, dyeff2)
         return loss, correct
 
@@ -160,15 +165,13 @@ class FusedLeNetStep:
         def gv(name):
             return getattr(m, name).grad
 
-        x, y1, y2, am1, am2, dx1 = self._stash
+        x, y1, am1, dact2 = self._stash
         s0 = torch.cuda.current_stream()
         s1 = self.side
-        dact2 = ext.pool_scatter(dx1, y2, am2, gv("conv2_b"), 14, 14)
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
             ext.conv_dw_into(y1, dact2, gv("conv2_w"))
         dxc = ext.conv_dx(dact2, m.shadows["conv2_w"], 32)
-        dact1 = ext.pool_scatter(dxc, y1, am1, gv("conv1_b"), 28, 28)
-        ext.conv_dw_into(x, dact1, gv("conv1_w"))
+        ext.conv1_dw_pooled(x, dxc, y1, am1, gv("conv1_w"), gv("conv1_b"))
         s0.wait_stream(s1)
         self._keep2 = (dact2,)

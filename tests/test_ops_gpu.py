@@ -197,6 +197,47 @@ def test_softmax_xent_gpu(ext):
     assert_close_bf16(dl, dl_ref, atol=2e-4)
 
 
+def test_linear_dx_unpool_matches_composition(ext):
+    """Fused dX+pool-backward epilogue == linear_dx followed by
+    pool_scatter: identical GEMM tile math, so dact must match BITWISE;
+    db differs only in summation point (fp32 acc vs rounded bf16)."""
+    torch.manual_seed(11)
+    B = 256
+    dyeff = (torch.randn(B, 512) * 0.1).to(bf16).cuda()
+    w = (torch.randn(3136, 512) * 0.05).to(bf16).cuda()
+    ypool = (torch.randn(B, 7, 7, 64)).to(bf16).cuda()
+    amax = torch.randint(0, 4, (B, 7, 7, 64), dtype=torch.uint8).cuda()
+    db_ref = torch.zeros(64).cuda()
+    dx1 = ext.linear_dx(dyeff, w).view(B, 7, 7, 64)
+    dact_ref = ext.pool_scatter(dx1, ypool, amax, db_ref, 14, 14)
+    db = torch.zeros(64).cuda()
+    dact = ext.linear_dx_unpool(dyeff, w, ypool, amax, db, 7, 7, 64)
+    assert dact.shape == (B, 14, 14, 64)
+    assert torch.equal(dact, dact_ref)
+    torch.testing.assert_close(db, db_ref, rtol=2e-2, atol=1e-2)
+
+
+def test_conv1_dw_pooled_matches_composition(ext):
+    """Pooled-consumer conv1 dW+db == pool_scatter followed by
+    conv_dw_into, within fp32-atomic reordering tolerance."""
+    torch.manual_seed(12)
+    B = 64
+    x = (torch.randn(B, 28, 28, 1) * 0.5).to(bf16).cuda()
+    dyp = (torch.randn(B, 14, 14, 32) * 0.1).to(bf16).cuda()
+    yp = torch.randn(B, 14, 14, 32).to(bf16).cuda()
+    am = torch.randint(0, 4, (B, 14, 14, 32), dtype=torch.uint8).cuda()
+    db_ref = torch.zeros(32).cuda()
+    dact1 = ext.pool_scatter(dyp, yp, am, db_ref, 28, 28)
+    dw_ref = torch.zeros(800).cuda()
+    ext.conv_dw_into(x, dact1, dw_ref)
+    dw = torch.zeros(800).cuda()
+    db = torch.zeros(32).cuda()
+    ext.conv1_dw_pooled(x, dyp, yp, am, dw, db)
+    torch.testing.assert_close(db, db_ref, rtol=1e-3, atol=1e-3)
+    scale = float(dw_ref.abs().max())
+    torch.testing.assert_close(dw, dw_ref, rtol=1e-2, atol=1e-3 * max(scale, 1.0))
+
+
 def test_grad_mask_gpu(ext):
     """Per-rank pre-aggregation drop-connect kernel: keep-rate, determinism,
     rank/step-distinct streams, and slice-wise composability (masking the
