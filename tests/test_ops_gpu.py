@@ -214,7 +214,12 @@ def test_linear_dx_unpool_matches_composition(ext):
     dact = ext.linear_dx_unpool(dyeff, w, ypool, amax, db, 7, 7, 64)
     assert dact.shape == (B, 14, 14, 64)
     assert torch.equal(dact, dact_ref)
-    torch.testing.assert_close(db, db_ref, rtol=2e-2, atol=1e-2)
+    # db sums fp32 GEMM accs (fused) vs rounded-bf16 dx1 (scatter): compare
+    # both against the exact fp32 reduction, scale-relative
+    db_torch = (dx1.float() * (ypool.float() > 0)).sum(dim=(0, 1, 2))
+    scale = float(db_torch.abs().max().clamp(min=1.0))
+    torch.testing.assert_close(db, db_torch, rtol=2e-2, atol=0.02 * scale)
+    torch.testing.assert_close(db_ref, db_torch, rtol=2e-2, atol=0.02 * scale)
 
 
 def test_conv1_dw_pooled_matches_composition(ext):
@@ -327,11 +332,16 @@ def test_lenet_full_model_gpu_vs_cpu(ext):
     # tolerance (scaled by each slice's own magnitude)
     ratio = float(gg.norm() / gc.norm().clamp(min=1e-12))
     assert 0.95 < ratio < 1.05, f"grad norm ratio {ratio}"
+    # per-parameter element-wise bound, relative to each slice's own max
+    # magnitude (bf16 deep reductions put ~5% of slice scale of noise on
+    # conv1_w's B*784-term dot2 chains; the norm-ratio above catches any
+    # uniform scale error the cosine alone would pass)
     for name, off, sz in zip(fpc.names, fpc.offsets, fpc.numels):
         ref_sl = gc[off:off + sz]
         got_sl = gg[off:off + sz]
-        assert_close_bf16(got_sl, ref_sl, rtol=0.08,
-                          scale=float(ref_sl.abs().max().clamp(min=1e-6)))
+        scale = float(ref_sl.abs().max().clamp(min=1e-6))
+        err = float((got_sl - ref_sl).abs().max())
+        assert err <= 0.08 * scale, f"{name}: maxerr {err} vs scale {scale}"
 
 
 def test_train_steps_reduce_loss_gpu(ext):
