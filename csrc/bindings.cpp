@@ -429,16 +429,15 @@ torch::Tensor linear_dx(torch::Tensor dyeff, torch::Tensor w) {
 }
 
 torch::Tensor linear_dx_unpool(torch::Tensor dyeff, torch::Tensor w,
-                               torch::Tensor ypool, torch::Tensor amax,
-                               torch::Tensor db_out, int64_t Ho, int64_t Wo,
-                               int64_t C) {
+                               torch::Tensor amax, torch::Tensor db_out,
+                               int64_t Ho, int64_t Wo, int64_t C) {
   // dx = dyeff @ W^T fused with maxpool2x2-backward: the dX rows are pooled
-  // [Ho][Wo][C] features; the epilogue relu-masks by ypool, scatters each
-  // value to its argmax position of the 2x2 window and accumulates the conv
-  // bias grad — dact [B][2Ho][2Wo][C] comes straight out of the GEMM
-  // (replaces linear_dx + pool_scatter on the backward critical path).
+  // [Ho][Wo][C] features; the epilogue scatters each value to its argmax
+  // position of the 2x2 window (amax byte: 0..3 live, 7 dead — liveness
+  // encoded at forward time so no y read is needed here) and accumulates
+  // the conv bias grad — dact [B][2Ho][2Wo][C] comes straight out of the
+  // GEMM (replaces linear_dx + pool_scatter on the backward critical path).
   CHECK_BF16(dyeff); CHECK_CONTIG(dyeff); CHECK_BF16(w); CHECK_CONTIG(w);
-  CHECK_BF16(ypool); CHECK_CONTIG(ypool);
   int B = dyeff.size(0), N = dyeff.size(1), K = w.size(0);
   TORCH_CHECK(K == Ho * Wo * C, "linear_dx_unpool: K != Ho*Wo*C");
   auto dact = torch::empty({B, 2 * (int)Ho, 2 * (int)Wo, (int)C},
@@ -450,7 +449,6 @@ torch::Tensor linear_dx_unpool(torch::Tensor dyeff, torch::Tensor w,
   p.lda = N; p.ldb = N; p.ldc = K;
   p.splitk = 1;
   p.CHo = Ho; p.CWo = Wo; p.Cout = C;
-  p.ypool = bf16_ptr(ypool);
   p.amax = amax.data_ptr<uint8_t>();
   p.db = db_out.defined() ? db_out.data_ptr<float>() : nullptr;
   bool big = cdiv(B, 128) * cdiv(K, 128) >= 128;
@@ -511,20 +509,19 @@ void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
   }
 }
 
-void conv1_dw_pooled(torch::Tensor x, torch::Tensor dyp, torch::Tensor yp,
-                     torch::Tensor am, torch::Tensor dw_out,
-                     torch::Tensor db_out) {
+void conv1_dw_pooled(torch::Tensor x, torch::Tensor dyp, torch::Tensor am,
+                     torch::Tensor dw_out, torch::Tensor db_out) {
   // conv1 dW+db straight from the pooled gradient (pool backward fused in
-  // the consumer; dact1 never materialized) — see conv1_dw_pooled_kernel.
+  // the consumer; dact1 never materialized; liveness in the amax byte) —
+  // see conv1_dw_pooled_kernel.
   CHECK_BF16(x); CHECK_CONTIG(x);
   CHECK_BF16(dyp); CHECK_CONTIG(dyp);
-  CHECK_BF16(yp); CHECK_CONTIG(yp);
   int NB = x.size(0), H = x.size(1), W = x.size(2);
   int Cout = dyp.size(3);
   TORCH_CHECK(x.size(3) == 1 && H == 28 && W == 28 && Cout == 32 &&
                   dyp.size(1) == H / 2 && dyp.size(2) == W / 2,
               "conv1_dw_pooled: LeNet conv1 shapes only");
-  launch_conv1_dw_pooled(bf16_ptr(x), bf16_ptr(dyp), bf16_ptr(yp),
+  launch_conv1_dw_pooled(bf16_ptr(x), bf16_ptr(dyp),
                          am.data_ptr<uint8_t>(), dw_out.data_ptr<float>(),
                          db_out.defined() ? db_out.data_ptr<float>() : nullptr,
                          NB, H, W, Cout, cur_stream());
@@ -672,13 +669,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_dx", &linear_dx, "dx = dyeff @ W^T");
   m.def("linear_dx_unpool", &linear_dx_unpool,
         "dx GEMM fused with maxpool2x2 backward scatter + conv db",
-        py::arg("dyeff"), py::arg("w"), py::arg("ypool"), py::arg("amax"),
+        py::arg("dyeff"), py::arg("w"), py::arg("amax"),
         py::arg("db_out"), py::arg("Ho"), py::arg("Wo"), py::arg("C"));
   m.def("pool_scatter", &pool_scatter, "maxpool bwd scatter + conv db");
   m.def("conv_dw_into", &conv_dw_into, "conv dW into bucket view");
   m.def("conv1_dw_pooled", &conv1_dw_pooled,
         "conv1 dW+db from the pooled gradient (pool bwd fused, no dact1)",
-        py::arg("x"), py::arg("dyp"), py::arg("yp"), py::arg("am"),
+        py::arg("x"), py::arg("dyp"), py::arg("am"),
         py::arg("dw_out"), py::arg("db_out"));
   m.def("conv_dx", &conv_dx, "conv dX");
   m.def("linear_act_bwd", &linear_act_bwd, "linear backward (dx, dw, db)");

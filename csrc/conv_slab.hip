@@ -212,7 +212,9 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
         if (v > best) { best = v; barg = r; }
       }
       yi[(size_t)gq * COUT + gc] = f2bf(best);
-      ai[(size_t)gq * COUT + gc] = (uint8_t)barg;
+      // liveness rides in the argmax byte: 7 = dead window (relu zeroed
+      // all 4) so the pooled-consumer backward kernels need no y read
+      ai[(size_t)gq * COUT + gc] = (uint8_t)(best > 0.f ? barg : 7);
     }
   }
 }

@@ -438,9 +438,8 @@ epilogue:
           int gr = m0 + wr * WM + mi * 16 + frow + r;
           if (gr >= p.M) continue;
           size_t qoff = (size_t)gr * p.N + gc;
-          float g = acc[mi][ni][r];
-          if (!(bf2f(p.ypool[qoff]) > 0.f)) g = 0.f;
-          int pos = p.amax[qoff];
+          int pos = p.amax[qoff];  // 0..3 live, 7 dead (no y read needed)
+          float g = pos < 4 ? acc[mi][ni][r] : 0.f;
           dbloc[ni] += g;
           ushort_t gb = f2bf(g);
 #pragma unroll
@@ -469,7 +468,8 @@ epilogue:
         }
         ushort_t* Cb = reinterpret_cast<ushort_t*>(p.C);
         Cb[(size_t)gq * p.ldc + gc] = f2bf(best);
-        if (p.amax) p.amax[(size_t)gq * p.ldc + gc] = (uint8_t)barg;
+        if (p.amax)  // 7 = dead window (liveness for pooled-consumer bwd)
+          p.amax[(size_t)gq * p.ldc + gc] = (uint8_t)(best > 0.f ? barg : 7);
       } else {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {

@@ -17,8 +17,7 @@ struct GemmParams {
   uint64_t seed, offset;
   const long* offset_dev;  // when set, RNG offset is read from device memory
                            // (hipGraph replay: host args are frozen)
-  const unsigned short* ypool;  // EPI_UNPOOL: pooled activations (relu mask)
-  float* db;                    // EPI_UNPOOL: conv bias-grad accumulator
+  float* db;  // EPI_UNPOOL: conv bias-grad accumulator
 };
 
 // gemm_tile.hip — implicit-GEMM MFMA entry points
@@ -65,9 +64,8 @@ void launch_sgd_step_dev(float* master, const float* grad,
                          uint64_t seed, const long* offset_dev,
                          float* momentum, float mu, hipStream_t);
 void launch_conv1_dw_pooled(const unsigned short* x, const unsigned short* dyp,
-                            const unsigned short* yp, const uint8_t* am,
-                            float* dw, float* db, int NB, int H, int W,
-                            int Cout, hipStream_t s);
+                            const uint8_t* am, float* dw, float* db, int NB,
+                            int H, int W, int Cout, hipStream_t s);
 void launch_grad_mask(float* g, long n, long base, float keep, uint64_t seed,
                       uint64_t step, uint64_t rank, const long* step_dev,
                       hipStream_t s);

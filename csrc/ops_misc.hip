@@ -548,7 +548,8 @@ void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
       if (v > best) { best = v; barg = pz; }
     }
     yi[(size_t)q * Cout + co] = f2bf(best);
-    ai[(size_t)q * Cout + co] = (uint8_t)barg;
+    // 7 = dead window (liveness for the pooled-consumer backward)
+    ai[(size_t)q * Cout + co] = (uint8_t)(best > 0.f ? barg : 7);
   }
 }
 
@@ -710,9 +711,8 @@ void conv1_dw_direct_kernel(const ushort_t* x, const ushort_t* dact,
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256)
 void conv1_dw_pooled_kernel(const ushort_t* x, const ushort_t* dyp,
-                            const ushort_t* yp, const uint8_t* am,
-                            float* dw, float* db, int NB, int H, int W,
-                            int Cout, int G) {
+                            const uint8_t* am, float* dw, float* db,
+                            int NB, int H, int W, int Cout, int G) {
   __shared__ ushort_t slab[32][32];
   __shared__ float red[26][32];  // 25 taps + db row
   const int tid = threadIdx.x;
@@ -746,9 +746,9 @@ void conv1_dw_pooled_kernel(const ushort_t* x, const ushort_t* dyp,
 #pragma unroll
         for (int qx = 0; qx < 14; ++qx) {
           size_t o = ib + ((size_t)rq * Wp + qx) * Cout;
-          float gv = bf2f(dyp[o]);
-          if (!(bf2f(yp[o]) > 0.f)) gv = 0.f;
-          pk[qx] = (unsigned)f2bf(gv) | ((unsigned)am[o] << 16);
+          int pos = am[o];  // 0..3 live, 7 dead (liveness in the byte)
+          float gv = pos < 4 ? bf2f(dyp[o]) : 0.f;
+          pk[qx] = (unsigned)f2bf(gv) | ((unsigned)pos << 16);
         }
 #pragma unroll
         for (int ry = 0; ry < 2; ++ry) {
@@ -812,13 +812,12 @@ void conv1_dw_pooled_kernel(const ushort_t* x, const ushort_t* dyp,
 }
 
 void launch_conv1_dw_pooled(const unsigned short* x, const unsigned short* dyp,
-                            const unsigned short* yp, const uint8_t* am,
-                            float* dw, float* db, int NB, int H, int W,
-                            int Cout, hipStream_t s) {
+                            const uint8_t* am, float* dw, float* db, int NB,
+                            int H, int W, int Cout, hipStream_t s) {
   int G = NB >= 8192 ? 8 : (NB >= 2048 ? NB / 1024 : 1);
   int blocks = (NB + G - 1) / G;
   hipLaunchKernelGGL(conv1_dw_pooled_kernel, dim3(blocks), dim3(256), 0, s,
-                     x, dyp, yp, am, dw, db, NB, H, W, Cout, G);
+                     x, dyp, am, dw, db, NB, H, W, Cout, G);
 }
 
 void launch_conv1_dw_direct(const unsigned short* x,

@@ -91,8 +91,9 @@ class FusedLeNetStep:
                 self.t.engine.wire_allreduce(self.fc_slice,
                                              start=self.fc_offset)
         # fc1 dX fused with pool2 backward: dact2 + conv2 db come straight
-        # out of the GEMM epilogue (no pool_scatter kernel on the chain)
-        dact2 = ext.linear_dx_unpool(dyeff1, sh["fc1_w"], y2, am2,
+        # out of the GEMM epilogue (no pool_scatter kernel on the chain;
+        # liveness rides in the amax byte, no y2 re-read)
+        dact2 = ext.linear_dx_unpool(dyeff1, sh["fc1_w"], am2,
                                      gv("conv2_b"), 7, 7, 64)
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
@@ -101,7 +102,7 @@ class FusedLeNetStep:
 
         # conv1 dW+db consume the POOLED gradient (pool1 backward fused in
         # the consumer; the 4x-size dact1 is never materialized)
-        ext.conv1_dw_pooled(x, dxc, y1, am1, gv("conv1_w"), gv("conv1_b"))
+        ext.conv1_dw_pooled(x, dxc, am1, gv("conv1_w"), gv("conv1_b"))
 
         s0.wait_stream(s1)
         if self.overlap_allreduce:
@@ -151,7 +152,7 @@ class FusedLeNetStep:
             ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
         # fused dX+pool2-backward: conv2_b lands here (conv slice — reduced
         # after graph B, so computing it early is safe)
-        dact2 = ext.linear_dx_unpool(dyeff1, sh["fc1_w"], y2, am2,
+        dact2 = ext.linear_dx_unpool(dyeff1, sh["fc1_w"], am2,
                                      gv("conv2_b"), 7, 7, 64)
         s0.wait_stream(s1)  # graph A boundary: fc grads complete
         self._stash = (x, y1, am1, dact2)
@@ -172,6 +173,6 @@ class FusedLeNetStep:
         with torch.cuda.stream(s1):
             ext.conv_dw_into(y1, dact2, gv("conv2_w"))
         dxc = ext.conv_dx(dact2, m.shadows["conv2_w"], 32)
-        ext.conv1_dw_pooled(x, dxc, y1, am1, gv("conv1_w"), gv("conv1_b"))
+        ext.conv1_dw_pooled(x, dxc, am1, gv("conv1_w"), gv("conv1_b"))
         s0.wait_stream(s1)
         self._keep2 = (dact2,)

@@ -43,6 +43,11 @@ def conv_pool_fwd(x, w, b):
     r = idx // W - ar * 2
     c = idx % W - ac * 2
     amax = (r * 2 + c).to(torch.uint8)               # [N,Cout,Ho,Wo]
+    # liveness rides in the argmax byte (matches the HIP kernels): 7 marks
+    # a dead window (relu zeroed all four) so pooled-consumer backward
+    # kernels need no y read
+    amax = torch.where(pooled > 0, amax,
+                       torch.tensor(7, dtype=torch.uint8))
     y = pooled.permute(0, 2, 3, 1).contiguous()      # NHWC
     amax = amax.permute(0, 2, 3, 1).contiguous()
     return y, amax
@@ -60,7 +65,10 @@ def conv_pool_bwd(dy, x, w, y, amax):
     Cin = x.shape[3]
     # route pooled grad (masked by relu) back to the argmax position
     g = (dy * (y > 0)).permute(0, 3, 1, 2).float()   # [N,Cout,Ho,Wo]
-    am = amax.permute(0, 3, 1, 2).long()
+    # clamp the dead-window marker (7) to a valid position: its g is zero
+    # by the relu mask, so the scatter target is irrelevant but must be
+    # in-bounds
+    am = amax.permute(0, 3, 1, 2).long().clamp(max=3)
     dact = x.new_zeros((N, Cout, H, W), dtype=torch.float32)
     ar = torch.arange(Ho, device=x.device).view(1, 1, Ho, 1)
     ac = torch.arange(Wo, device=x.device).view(1, 1, 1, Wo)
