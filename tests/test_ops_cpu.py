@@ -32,7 +32,8 @@ def test_conv_pool_fwd(N, H, W, Cin, Cout):
     assert y.shape == (N, H // 2, W // 2, Cout)
     torch.testing.assert_close(y, ref, rtol=1e-4, atol=1e-4)
     assert amax.dtype == torch.uint8
-    assert int(amax.max()) <= 3
+    vals = set(torch.unique(amax).tolist())
+    assert vals <= {0, 1, 2, 3, 7}, vals  # 7 = dead-window liveness marker
 
 
 def test_conv_pool_bwd_matches_autograd():
