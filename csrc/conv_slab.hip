@@ -247,8 +247,10 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
   // MFMAs, one barrier/K-step — but the extra 4 KB costs a resident block
   // (2/CU) at this slab size.  DB=1: serial glds drain, two barriers, 3
   // blocks/CU — the TLP wins at large NB (same threshold the hi-occ
-  // scatter variant sat on).
-  __shared__ __align__(16) u16 Bs[DB][BN][BK];
+  // scatter variant sat on).  DB=3: paired K-steps (two buffers staged per
+  // barrier pair — half the barriers at DB=1's footprint).
+  constexpr int NBUF = (DB >= 2) ? 2 : 1;
+  __shared__ __align__(16) u16 Bs[NBUF][BN][BK];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -303,6 +305,51 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
 
   constexpr int NT_K = (K + BK - 1) / BK;
   int cur = 0;
+  // DB=3: PAIRED K-steps — both 64-wide buffers staged per barrier pair,
+  // halving the barrier count vs DB=1 (13 pairs instead of 25 steps for
+  // K=1600) at DB=1's occupancy; the two glds streams drain together.
+  if (DB == 3) {
+    for (int t = 0; t < NT_K; t += 2) {
+      issueB(0, t * BK);
+      if (t + 1 < NT_K) issueB(1, (t + 1) * BK);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        if (t + half >= NT_K) break;
+        int kt = (t + half) * BK;
+        const u16* bp = &Bs[half][0][0];
+#pragma unroll
+        for (int kh2 = 0; kh2 < 2; ++kh2) {
+          int kbase = kt + kh2 * 32 + kq;
+          int khkw = kbase / COUT, co = kbase % COUT;
+          int kh = khkw / 5, kw = khkw % 5;
+          int shift = (-(kh - 2) * WP - (kw - 2)) * PST + co;
+          short8 af[MI], bf[NI];
+#pragma unroll
+          for (int mi = 0; mi < MI; ++mi)
+            af[mi] = *reinterpret_cast<const short8*>(
+                &slab[arow_off[mi] + shift]);
+#pragma unroll
+          for (int ni = 0; ni < NI; ++ni) {
+            int n = wc * WN + ni * 16 + (lane & 15);
+            int xs = (n ^ (n >> 3)) & 7;
+            int chunk = (kh2 * 4 + (lane >> 4)) ^ xs;
+            bf[ni] = *reinterpret_cast<const short8*>(bp + (size_t)n * BK +
+                                                      chunk * 8);
+          }
+#pragma unroll
+          for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < NI; ++ni)
+              acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+    goto epilogue_dx;
+  }
   for (int t = 0; t < NT_K; ++t) {
     int kt = t * BK;
     if (DB == 2) {
@@ -344,6 +391,7 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
     if (DB == 2) cur ^= 1;
   }
 
+epilogue_dx:
   const int frow = (lane >> 4) * 4;
   const int fcol = lane & 15;
   u16* xo = dx + (size_t)img * M * CIN;
@@ -503,7 +551,10 @@ void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
     // large batches feel — there the SINGLE-buffered glds form keeps
     // 3/CU.  DMNIST_DX_HIOCC=1 reverts large batches to the original
     // scatter-staged kernel for A/B.
-    if (NB < 2048)
+    if (getenv("DMNIST_DX_PAIR"))  // paired-K ablation (half the barriers)
+      hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 3>), dim3(NB),
+                         dim3(NTHREADS), 0, s, dact, w, dx, NB);
+    else if (NB < 2048)
       hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 2>), dim3(NB),
                          dim3(NTHREADS), 0, s, dact, w, dx, NB);
     else if (getenv("DMNIST_DX_HIOCC"))

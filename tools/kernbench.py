@@ -92,10 +92,12 @@ def bench_batch(ext, B):
         ("fc1 dW (A_T split-K)", lambda: ext.linear_dw_into(h2, dyeff1, dw1f)),
         ("fc1 dX", lambda: ext.linear_dx(dyeff1, f1)),
         ("pool2 bwd scatter", lambda: ext.pool_scatter(dx1, y2, am2, dbc2, 14, 14)),
+        ("fc1 dX+unpool fused", lambda: ext.linear_dx_unpool(dyeff1, f1, am2, dbc2, 7, 7, 64)),
         ("conv2 dW (slab/gemm tiered)", lambda: ext.conv_dw_into(y1, dact2, dwc2)),
         ("conv2 dX (image slab)", lambda: ext.conv_dx(dact2, w2, 32)),
         ("pool1 bwd scatter", lambda: ext.pool_scatter(dxc, y1, am1, dbc1, 28, 28)),
         ("conv1 dW (slab)", lambda: ext.conv_dw_into(x1, dact1, dwc1)),
+        ("conv1 dW+db pooled fused", lambda: ext.conv1_dw_pooled(x1, dxc, am1, dwc1.view(-1), dbc1)),
         ("fused SGD apply (1.66M params)", lambda: ext.sgd_step(master, grad, shadow, True, 0.01, 1.0, -1.0, 0, 0)),
     ]
     print(f"\n## B = {B} (solo, median of 30, us)\n")
