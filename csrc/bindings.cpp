@@ -456,6 +456,32 @@ torch::Tensor linear_dx_unpool(torch::Tensor dyeff, torch::Tensor w,
   return dact;
 }
 
+torch::Tensor linear_dx_mask(torch::Tensor dyeff, torch::Tensor w,
+                             torch::Tensor actm, torch::Tensor db_out,
+                             double p_keep) {
+  // dx = dyeff @ W^T with the downstream relu+dropout mask folded into the
+  // epilogue (mask recovered from sign(actm): kept positions are scaled by
+  // 1/p_keep, dropped/clipped are 0) + bias-grad column sums into db_out —
+  // replaces linear_dx + the standalone mask_db pass (fc2-dX -> dyeff1).
+  CHECK_BF16(dyeff); CHECK_CONTIG(dyeff); CHECK_BF16(w); CHECK_CONTIG(w);
+  CHECK_BF16(actm); CHECK_CONTIG(actm);
+  int B = dyeff.size(0), N = dyeff.size(1), K = w.size(0);
+  TORCH_CHECK(actm.size(0) == B && actm.size(1) == K, "actm shape mismatch");
+  auto dx = torch::empty({B, K}, dyeff.options());
+  GemmParams p{};
+  p.A = bf16_ptr(dyeff); p.B = bf16_ptr(w);
+  p.C = dx.data_ptr();
+  p.M = B; p.N = K; p.K = N;
+  p.lda = N; p.ldb = N; p.ldc = K;
+  p.splitk = 1;
+  p.p_keep = (float)(1.0 / p_keep);  // kernel multiplies by the inverse
+  p.actm = bf16_ptr(actm);
+  p.db = db_out.defined() ? db_out.data_ptr<float>() : nullptr;
+  bool big = cdiv(B, 128) * cdiv(K, 128) >= 128;
+  (big ? gemm_dx_mask_128 : gemm_dx_mask_64)(p, cur_stream());
+  return dx;
+}
+
 torch::Tensor pool_scatter(torch::Tensor dy, torch::Tensor y,
                            torch::Tensor amax, torch::Tensor db_out,
                            int64_t H, int64_t W) {
@@ -552,7 +578,10 @@ torch::Tensor conv_dx(torch::Tensor dact, torch::Tensor w, int64_t Cin) {
 
 // --------------------------------------------------------------------------
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
-                                            torch::Tensor labels) {
+                                            torch::Tensor labels,
+                                            c10::optional<torch::Tensor> db_out) {
+  // db_out (optional): fc2 bias grad accumulated in the same pass (column
+  // sums of dlogits) — removes the standalone mask_db launch
   CHECK_CUDA(logits); CHECK_BF16(logits); CHECK_CONTIG(logits);
   TORCH_CHECK(labels.scalar_type() == at::kLong, "labels must be int64");
   CHECK_CONTIG(labels);
@@ -560,8 +589,11 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
   TORCH_CHECK(C <= 16, "softmax_xent kernel supports C<=16");
   auto dl = torch::empty_like(logits);
   auto out = torch::zeros({2}, logits.options().dtype(at::kFloat));
+  float* dbp = nullptr;
+  if (db_out.has_value() && db_out->defined())
+    dbp = db_out->data_ptr<float>();
   launch_softmax_xent(bf16_ptr(logits), labels.data_ptr<long>(),
-                      bf16_mut(dl), out.data_ptr<float>(), B, C,
+                      bf16_mut(dl), out.data_ptr<float>(), B, C, dbp,
                       cur_stream());
   auto loss = out.select(0, 0);
   auto correct = out.select(0, 1);
@@ -590,7 +622,8 @@ void sgd_step_dev(torch::Tensor master, torch::Tensor grad,
                   torch::Tensor shadow, bool has_shadow,
                   torch::Tensor lr_scale_dev, double dc_keep, int64_t seed,
                   torch::Tensor offset_dev,
-                  c10::optional<torch::Tensor> momentum, double mu) {
+                  c10::optional<torch::Tensor> momentum, double mu,
+                  bool zero_grad) {
   CHECK_CUDA(master); CHECK_F32(master); CHECK_CONTIG(master);
   CHECK_F32(grad); CHECK_CONTIG(grad);
   CHECK_F32(lr_scale_dev);
@@ -604,7 +637,7 @@ void sgd_step_dev(torch::Tensor master, torch::Tensor grad,
                       has_shadow ? 1 : 0, master.numel(),
                       lr_scale_dev.data_ptr<float>(), (float)dc_keep,
                       (uint64_t)seed, offset_dev.data_ptr<long>(),
-                      mom, (float)mu, cur_stream());
+                      mom, (float)mu, zero_grad ? 1 : 0, cur_stream());
 }
 
 void transpose_bf16_batch(std::vector<torch::Tensor> srcs,
@@ -667,6 +700,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mask_db", &mask_db, "relu/dropout grad mask + bias-grad column sums");
   m.def("linear_dw_into", &linear_dw_into, "dW = x^T dyeff into bucket view");
   m.def("linear_dx", &linear_dx, "dx = dyeff @ W^T");
+  m.def("linear_dx_mask", &linear_dx_mask,
+        "dx GEMM with fused relu/dropout mask + bias-grad column sums",
+        py::arg("dyeff"), py::arg("w"), py::arg("actm"), py::arg("db_out"),
+        py::arg("p_keep"));
   m.def("linear_dx_unpool", &linear_dx_unpool,
         "dx GEMM fused with maxpool2x2 backward scatter + conv db",
         py::arg("dyeff"), py::arg("w"), py::arg("amax"),
@@ -686,18 +723,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_pool_bwd", &conv_pool_bwd, "conv+pool backward (dx, dw, db)");
   m.def("conv_pool_bwd_into", &conv_pool_bwd_into,
         "conv backward accumulating dw/db into bucket views");
-  m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax-CE (+grad)");
+  m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax-CE (+grad, +optional fc2 db)",
+        py::arg("logits"), py::arg("labels"),
+        py::arg("db_out") = c10::nullopt);
   m.def("sgd_step", &sgd_step, "fused flat SGD(+momentum) apply",
         py::arg("master"), py::arg("grad"), py::arg("shadow"),
         py::arg("has_shadow"), py::arg("lr"), py::arg("scale"),
         py::arg("dc_keep"), py::arg("seed"), py::arg("offset"),
         py::arg("momentum") = c10::nullopt, py::arg("mu") = 0.0);
   m.def("sgd_step_dev", &sgd_step_dev,
-        "SGD apply with device-side lr/offset (hipGraph-capturable)",
+        "SGD apply with device-side lr/offset (hipGraph-capturable); "
+        "zero_grad clears the bucket in the same pass",
         py::arg("master"), py::arg("grad"), py::arg("shadow"),
         py::arg("has_shadow"), py::arg("lr_scale_dev"), py::arg("dc_keep"),
         py::arg("seed"), py::arg("offset_dev"),
-        py::arg("momentum") = c10::nullopt, py::arg("mu") = 0.0);
+        py::arg("momentum") = c10::nullopt, py::arg("mu") = 0.0,
+        py::arg("zero_grad") = false);
   m.def("step_advance", &step_advance,
         "device-side staircase LR + step increment (inside the graph)");
   m.def("grad_mask", &grad_mask,

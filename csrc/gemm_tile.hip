@@ -32,7 +32,11 @@ enum BMode {
   B_CONV_DX_W = 2, // Bmat[k=(khkw,co)][n=ci] = w[((khkw)*Cin+ci)*Cout + co]
 };
 enum Epi { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_RELU = 2, EPI_BIAS_RELU_DROP = 3,
-           EPI_POOL = 4, EPI_UNPOOL = 5 };
+           EPI_POOL = 4, EPI_UNPOOL = 5, EPI_MASK_DB = 6 };
+// EPI_MASK_DB: dX epilogue that recovers the downstream relu+dropout mask
+// from the sign of the saved activation (actm) — v = a>0 ? v/p_keep : 0 —
+// and accumulates the bias grad db[n] (column sums).  Folds the standalone
+// relu_drop_bwd (mask_db) pass into the producing dX GEMM.
 // EPI_UNPOOL: the output row m is a batch index and column n a pooled
 // feature (ho, wo, c) of a [CHo][CWo][Cout] activation; instead of writing
 // C[m][n], the epilogue routes the value through the maxpool-2x2 backward
@@ -264,9 +268,10 @@ void gemm_tile_kernel(GemmParams p) {
   static_assert(CHA >= 1 && CHB >= 1, "tile too small for 256 threads");
   __shared__ __align__(16) ushort_t As[DB][BM][LDK];
   __shared__ __align__(16) ushort_t Bs[DB][BN][LDK];
-  // EPI_UNPOOL: per-column bias-grad partials, flushed once per block
-  __shared__ float dbred[EPI == EPI_UNPOOL ? BN : 1];
-  if (EPI == EPI_UNPOOL && threadIdx.x < BN) dbred[threadIdx.x] = 0.f;
+  // EPI_UNPOOL / EPI_MASK_DB: per-column bias-grad partials, flushed once
+  constexpr bool HAS_DB = (EPI == EPI_UNPOOL || EPI == EPI_MASK_DB);
+  __shared__ float dbred[HAS_DB ? BN : 1];
+  if (HAS_DB && threadIdx.x < BN) dbred[threadIdx.x] = 0.f;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -414,8 +419,8 @@ void gemm_tile_kernel(GemmParams p) {
 epilogue:
   const int frow = (lane >> 4) * 4;  // C/D: row=(lane>>4)*4+reg, col=lane&15
   const int fcol = lane & 15;
-  float dbloc[EPI == EPI_UNPOOL ? NI : 1];
-  if (EPI == EPI_UNPOOL)
+  float dbloc[(EPI == EPI_UNPOOL || EPI == EPI_MASK_DB) ? NI : 1];
+  if (EPI == EPI_UNPOOL || EPI == EPI_MASK_DB)
 #pragma unroll
     for (int ni = 0; ni < NI; ++ni) dbloc[ni] = 0.f;
 #pragma unroll
@@ -476,7 +481,12 @@ epilogue:
           int gr = m0 + wr * WM + mi * 16 + frow + r;
           if (gr >= p.M) continue;
           float v = acc[mi][ni][r];
-          if (EPI != EPI_NONE) v += bias_v;
+          if (EPI == EPI_MASK_DB) {
+            float a = bf2f(p.actm[(size_t)gr * p.ldc + gc]);
+            v = (a > 0.f) ? v * p.p_keep : 0.f;  // p_keep holds 1/keep here
+            dbloc[ni] += v;
+          }
+          if (EPI != EPI_NONE && EPI != EPI_MASK_DB) v += bias_v;
           if (EPI == EPI_BIAS_RELU || EPI == EPI_BIAS_RELU_DROP)
             v = v > 0.f ? v : 0.f;
           if (EPI == EPI_BIAS_RELU_DROP) {
@@ -500,9 +510,9 @@ epilogue:
       }
     }
   }
-  if (EPI == EPI_UNPOOL) {
+  if (EPI == EPI_UNPOOL || EPI == EPI_MASK_DB) {
     // bias-grad flush: lane partials -> LDS per-column -> one global
-    // atomicAdd per block column (channel = col % Cout)
+    // atomicAdd per block column (UNPOOL: channel = col % Cout)
 #pragma unroll
     for (int ni = 0; ni < NI; ++ni) {
       int colb = wc * WN + ni * 16 + fcol;
@@ -512,7 +522,8 @@ epilogue:
     if (p.db && (int)threadIdx.x < BN) {
       int gc = n0 + (int)threadIdx.x;
       float v = dbred[threadIdx.x];
-      if (gc < p.N && v != 0.f) atomicAdd(&p.db[gc % p.Cout], v);
+      int idx = (EPI == EPI_UNPOOL) ? gc % p.Cout : gc;
+      if (gc < p.N && v != 0.f) atomicAdd(&p.db[idx], v);
     }
   }
 }
@@ -549,6 +560,9 @@ GEMM_ENTRY(gemm_dx_64, 64, 64, A_N, B_NMAJ, EPI_NONE, OUT_BF16, 1)
 // dX fused with maxpool-2x2 backward (fc1 dX -> dact2 directly + conv2 db)
 GEMM_ENTRY(gemm_dx_unpool_128, 128, 128, A_N, B_NMAJ, EPI_UNPOOL, OUT_BF16, 1)
 GEMM_ENTRY(gemm_dx_unpool_64, 64, 64, A_N, B_NMAJ, EPI_UNPOOL, OUT_BF16, 1)
+// dX fused with the downstream relu/dropout mask + bias-grad column sums
+GEMM_ENTRY(gemm_dx_mask_128, 128, 128, A_N, B_NMAJ, EPI_MASK_DB, OUT_BF16, 1)
+GEMM_ENTRY(gemm_dx_mask_64, 64, 64, A_N, B_NMAJ, EPI_MASK_DB, OUT_BF16, 1)
 GEMM_ENTRY_SWZ(gemm_dw_128, 128, 128, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 2)
 GEMM_ENTRY_SWZ(gemm_dw_64, 64, 64, A_T, B_KMAJ, EPI_NONE, OUT_F32_ATOMIC, 2)
 // B-transposed (pre-transposed weight) fwd variants: ldb = K, vector staging

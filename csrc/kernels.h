@@ -17,7 +17,8 @@ struct GemmParams {
   uint64_t seed, offset;
   const long* offset_dev;  // when set, RNG offset is read from device memory
                            // (hipGraph replay: host args are frozen)
-  float* db;  // EPI_UNPOOL: conv bias-grad accumulator
+  float* db;  // EPI_UNPOOL / EPI_MASK_DB: bias-grad accumulator
+  const unsigned short* actm;  // EPI_MASK_DB: saved activation (mask source)
 };
 
 // gemm_tile.hip — implicit-GEMM MFMA entry points
@@ -31,6 +32,8 @@ void gemm_dx_128(const GemmParams&, hipStream_t);
 void gemm_dx_64(const GemmParams&, hipStream_t);
 void gemm_dx_unpool_128(const GemmParams&, hipStream_t);
 void gemm_dx_unpool_64(const GemmParams&, hipStream_t);
+void gemm_dx_mask_128(const GemmParams&, hipStream_t);
+void gemm_dx_mask_64(const GemmParams&, hipStream_t);
 void gemm_dw_128(const GemmParams&, hipStream_t);
 void gemm_dw_64(const GemmParams&, hipStream_t);
 void conv_fwd_pool(const GemmParams&, hipStream_t);
@@ -52,17 +55,18 @@ void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
                              hipStream_t);
 void launch_softmax_xent(const unsigned short* logits, const long* labels,
                          unsigned short* dlogits, float* out, int B, int C,
-                         hipStream_t);
-void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
+                         float* db, hipStream_t);
+void launch_sgd_step(float* master, float* grad, unsigned short* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
                      uint64_t seed, uint64_t offset, float* momentum, float mu,
                      hipStream_t);
 // graph-capturable variants: lr_scale / RNG offset read from device memory
-void launch_sgd_step_dev(float* master, const float* grad,
+void launch_sgd_step_dev(float* master, float* grad,
                          unsigned short* shadow, int has_shadow, long n,
                          const float* lr_scale_dev, float dc_keep,
                          uint64_t seed, const long* offset_dev,
-                         float* momentum, float mu, hipStream_t);
+                         float* momentum, float mu, int zero_grad,
+                         hipStream_t);
 void launch_conv1_dw_pooled(const unsigned short* x, const unsigned short* dyp,
                             const uint8_t* am, float* dw, float* db, int NB,
                             int H, int W, int Cout, hipStream_t s);

@@ -130,12 +130,18 @@ void pool_bwd_scatter_kernel(const ushort_t* dy, const ushort_t* y,
 // ---------------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(256)
 void softmax_xent_kernel(const ushort_t* logits, const long* labels,
-                         ushort_t* dlogits, float* out, int B, int C) {
+                         ushort_t* dlogits, float* out, int B, int C,
+                         float* db) {
+  // db (optional): column sums of dlogits = the fc2 bias grad — folding it
+  // here removes the standalone mask_db pass on the backward chain
   __shared__ float red[2];
+  __shared__ float dbred[16];
   if (threadIdx.x == 0) { red[0] = 0.f; red[1] = 0.f; }
+  if (db && threadIdx.x < 16) dbred[threadIdx.x] = 0.f;
   __syncthreads();
   int b = blockIdx.x * blockDim.x + threadIdx.x;
   float loss = 0.f, correct = 0.f;
+  float dloc[16];
   if (b < B) {
     float v[16];
     float mx = -1e30f;
@@ -153,16 +159,24 @@ void softmax_xent_kernel(const ushort_t* logits, const long* labels,
       float p = v[c] * inv_se;
       float d = (p - (c == (int)lab ? 1.f : 0.f)) * invB;
       dlogits[(size_t)b * C + c] = f2bf(d);
+      dloc[c] = d;
     }
     loss = -(__logf(v[(int)lab] * inv_se)) * invB;
     correct = (arg == (int)lab) ? 1.f : 0.f;
   }
   atomicAdd(&red[0], loss);
   atomicAdd(&red[1], correct);
+  if (db && b < B)
+    for (int c = 0; c < C; ++c)
+      if (dloc[c] != 0.f) atomicAdd(&dbred[c], dloc[c]);
   __syncthreads();
   if (threadIdx.x == 0) {
     atomicAdd(&out[0], red[0]);
     atomicAdd(&out[1], red[1]);
+  }
+  if (db && (int)threadIdx.x < C) {
+    float v = dbred[threadIdx.x];
+    if (v != 0.f) atomicAdd(&db[threadIdx.x], v);
   }
 }
 
@@ -172,11 +186,14 @@ void softmax_xent_kernel(const ushort_t* logits, const long* labels,
 // (mask, NO rescale).  Vectorized float4 path + scalar tail.
 // ---------------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(256)
-void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
+void sgd_step_kernel(float* master, float* grad, ushort_t* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
                      uint64_t seed, uint64_t offset,
                      const float* lr_scale_dev, const long* offset_dev,
-                     float* momentum, float mu) {
+                     float* momentum, float mu, int zero_grad) {
+  // zero_grad: clear the flat gradient bucket in the SAME pass that
+  // consumes it (the captured graph then needs no per-step fill kernel;
+  // all dW/db writers accumulate with atomics into a zeroed bucket)
   // momentum (optional): v = mu*v + g; w -= lr*v  (v=nullptr => plain SGD,
   // the reference's GradientDescentOptimizer semantics)
   if (lr_scale_dev) lr_scale = *lr_scale_dev;
@@ -206,6 +223,8 @@ void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
       m.x -= lr_scale * g.x; m.y -= lr_scale * g.y;
       m.z -= lr_scale * g.z; m.w -= lr_scale * g.w;
       *reinterpret_cast<float4*>(master + base) = m;
+      if (zero_grad)
+        *reinterpret_cast<float4*>(grad + base) = float4{0.f, 0.f, 0.f, 0.f};
       if (has_shadow) {
         shadow[base] = f2bf(m.x); shadow[base + 1] = f2bf(m.y);
         shadow[base + 2] = f2bf(m.z); shadow[base + 3] = f2bf(m.w);
@@ -224,6 +243,7 @@ void sgd_step_kernel(float* master, const float* grad, ushort_t* shadow,
         }
         float m = master[j] - lr_scale * g;
         master[j] = m;
+        if (zero_grad) grad[j] = 0.f;
         if (has_shadow) shadow[j] = f2bf(m);
       }
     }
@@ -319,17 +339,17 @@ void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
 
 void launch_softmax_xent(const unsigned short* logits, const long* labels,
                          unsigned short* dlogits, float* out, int B, int C,
-                         hipStream_t s) {
+                         float* db, hipStream_t s) {
   // one row per thread: B=1024 in 256-thread blocks was only 4 blocks
   // (256-CU chip ~idle) — 64-thread blocks spread it; at large B the
   // extra per-block out[] atomics cost more than the spread gains
   int bt = B >= 2048 ? 256 : 64;
   dim3 grid(cdivh(B, bt));
   hipLaunchKernelGGL(softmax_xent_kernel, grid, dim3(bt), 0, s, logits,
-                     labels, dlogits, out, B, C);
+                     labels, dlogits, out, B, C, db);
 }
 
-void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
+void launch_sgd_step(float* master, float* grad, unsigned short* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
                      uint64_t seed, uint64_t offset, float* momentum, float mu,
                      hipStream_t s) {
@@ -339,20 +359,21 @@ void launch_sgd_step(float* master, const float* grad, unsigned short* shadow,
   hipLaunchKernelGGL(sgd_step_kernel, dim3(blocks), dim3(256), 0, s, master,
                      grad, shadow, has_shadow, n, lr_scale, dc_keep, seed,
                      offset, (const float*)nullptr, (const long*)nullptr,
-                     momentum, mu);
+                     momentum, mu, 0);
 }
 
-void launch_sgd_step_dev(float* master, const float* grad,
+void launch_sgd_step_dev(float* master, float* grad,
                          unsigned short* shadow, int has_shadow, long n,
                          const float* lr_scale_dev, float dc_keep,
                          uint64_t seed, const long* offset_dev,
-                         float* momentum, float mu, hipStream_t s) {
+                         float* momentum, float mu, int zero_grad,
+                         hipStream_t s) {
   long groups = (n + 3) / 4;
   int blocks = (int)min((long)2048, (groups + 255) / 256);
   if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(sgd_step_kernel, dim3(blocks), dim3(256), 0, s, master,
                      grad, shadow, has_shadow, n, 0.f, dc_keep, seed, 0,
-                     lr_scale_dev, offset_dev, momentum, mu);
+                     lr_scale_dev, offset_dev, momentum, mu, zero_grad);
 }
 
 // ---------------------------------------------------------------------------

@@ -72,16 +72,19 @@ class FusedLeNetStep:
                                     shT["fc1_w"])
         logits = ext.linear_act_fwd(a1, sh["fc2_w"], m.fc2_b, False, 1.0,
                                     0, 0, shT["fc2_w"])
-        loss, correct, dl = ext.softmax_xent_fwd(logits, labels)
+        # softmax computes the fc2 bias grad in the same pass (column sums
+        # of dlogits) — no standalone mask_db launch for fc2
+        loss, correct, dl = ext.softmax_xent_fwd(logits, labels,
+                                                 db_out=gv("fc2_b"))
 
         # ---- backward: dX chain on s0, dW GEMMs on s1 ----
-        dyeff2 = ext.mask_db(dl, dl, False, 1.0, gv("fc2_b"))
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
-            ext.linear_dw_into(a1, dyeff2, gv("fc2_w"))
-        dx2 = ext.linear_dx(dyeff2, sh["fc2_w"])
-
-        dyeff1 = ext.mask_db(dx2, a1, True, self.p_keep, gv("fc1_b"))
+            ext.linear_dw_into(a1, dl, gv("fc2_w"))
+        # fc2 dX with the fc1 relu+dropout mask folded into the epilogue
+        # (+ fc1 bias grad) — replaces linear_dx + mask_db
+        dyeff1 = ext.linear_dx_mask(dl, sh["fc2_w"], a1, gv("fc1_b"),
+                                    self.p_keep)
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
@@ -108,7 +111,7 @@ class FusedLeNetStep:
         if self.overlap_allreduce:
             self.t.engine.wire_allreduce(self.conv_slice, start=0)
         # keep the side-stream consumers alive until the join (capture-safe)
-        self._keep = (a1, h2, dyeff1, dyeff2, y1, dact2)
+        self._keep = (a1, h2, dyeff1, dl, y1, dact2)
         return loss, correct
 
     # ------------------------------------------------------------------
@@ -140,13 +143,13 @@ class FusedLeNetStep:
                                     shT["fc1_w"])
         logits = ext.linear_act_fwd(a1, sh["fc2_w"], m.fc2_b, False, 1.0,
                                     0, 0, shT["fc2_w"])
-        loss, correct, dl = ext.softmax_xent_fwd(logits, labels)
-        dyeff2 = ext.mask_db(dl, dl, False, 1.0, gv("fc2_b"))
+        loss, correct, dl = ext.softmax_xent_fwd(logits, labels,
+                                                 db_out=gv("fc2_b"))
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
-            ext.linear_dw_into(a1, dyeff2, gv("fc2_w"))
-        dx2 = ext.linear_dx(dyeff2, sh["fc2_w"])
-        dyeff1 = ext.mask_db(dx2, a1, True, self.p_keep, gv("fc1_b"))
+            ext.linear_dw_into(a1, dl, gv("fc2_w"))
+        dyeff1 = ext.linear_dx_mask(dl, sh["fc2_w"], a1, gv("fc1_b"),
+                                    self.p_keep)
         s1.wait_stream(s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
@@ -156,7 +159,7 @@ class FusedLeNetStep:
                                      gv("conv2_b"), 7, 7, 64)
         s0.wait_stream(s1)  # graph A boundary: fc grads complete
         self._stash = (x, y1, am1, dact2)
-        self._keep = (a1, h2, dyeff1, dyeff2)
+        self._keep = (a1, h2, dyeff1, dl)
         return loss, correct
 
     def stage_conv(self):

@@ -75,6 +75,10 @@ class GraphedStep:
             from .fused_step import FusedLeNetStep
             self._fused = FusedLeNetStep(t)
 
+        # entering from eager steps the bucket may hold stale grads (the
+        # eager path zeroes at step START); the graphed protocol zeroes at
+        # step END (SGD tail), so establish its invariant once here
+        t.fp.flat_grad.zero_()
         # snapshot state: the warmup iterations below really train
         master0 = t.fp.flat_master.clone()
         mom0 = (t.flat_momentum.clone()
@@ -136,7 +140,8 @@ class GraphedStep:
         try:
             if self.two_graph:
                 with torch.cuda.graph(self.graph):
-                    t.fp.flat_grad.zero_()
+                    # no grad-bucket fill: the previous replay's SGD tail
+                    # zeroed it (zero_grad=True), and warmup left it zeroed
                     loss, correct = self._fused.stage_fc(
                         self.static_x, self.static_y, self.step_dev)
                     if self.dc_pre:
@@ -214,7 +219,7 @@ class GraphedStep:
         the fused step already issued the bucketed all-reduce."""
         t = self.t
         fp = t.fp
-        fp.flat_grad.zero_()
+        # grad bucket already zero: the SGD tail clears it after consuming
         fused_reduced = False
         if self._fused is not None:
             loss, correct = self._fused(self.static_x, self.static_y,
@@ -238,6 +243,8 @@ class GraphedStep:
         this is capturable AND replay-equivalent when run eagerly)."""
         t = self.t
         fp = t.fp
+        # zero_grad=True: the bucket is cleared in the same pass that
+        # consumes it — no per-step fill kernel in the replayed graph
         self._ext.sgd_step_dev(fp.flat_master, fp.flat_grad,
                                fp.flat_shadow if fp.flat_shadow is not None
                                else fp.flat_master,
@@ -245,7 +252,7 @@ class GraphedStep:
                                self.lr_scale_dev, self.dc_keep,
                                t.flags.seed, self.step_dev,
                                momentum=t.flat_momentum,
-                               mu=t.flags.momentum)
+                               mu=t.flags.momentum, zero_grad=True)
         fp.refresh_transposes()
         # advance step + LR on-device for the next body execution
         self._ext.step_advance(self.step_dev, self.lr_scale_dev,

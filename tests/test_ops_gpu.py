@@ -251,6 +251,59 @@ def test_conv1_dw_pooled_matches_composition(ext):
     torch.testing.assert_close(dw, dw_ref, rtol=1e-2, atol=1e-3 * max(scale, 1.0))
 
 
+def test_softmax_db_matches_mask_db(ext):
+    """softmax_xent_fwd(db_out=...) must equal the old mask_db column sums
+    of dlogits (the fc2 bias grad folded into the softmax pass)."""
+    torch.manual_seed(13)
+    B = 1024
+    logits = (torch.randn(B, 10)).to(bf16).cuda()
+    labels = torch.randint(0, 10, (B,)).cuda()
+    l0, c0, dl0 = ext.softmax_xent_fwd(logits, labels)
+    db_ref = torch.zeros(10).cuda()
+    ext.mask_db(dl0, dl0, False, 1.0, db_ref)
+    db = torch.zeros(10).cuda()
+    l1, c1, dl1 = ext.softmax_xent_fwd(logits, labels, db_out=db)
+    assert torch.equal(dl0, dl1)
+    assert float(l0) == float(l1) and float(c0) == float(c1)
+    torch.testing.assert_close(db, db_ref, rtol=1e-3, atol=1e-5)
+
+
+def test_linear_dx_mask_matches_composition(ext):
+    """Fused dX+mask epilogue == linear_dx followed by mask_db (relu+
+    dropout mask from sign(actm), 1/p_keep scale, fc1 bias column sums)."""
+    torch.manual_seed(14)
+    B, N, K = 512, 10, 512
+    dl = (torch.randn(B, N) * 0.1).to(bf16).cuda()
+    w = (torch.randn(K, N) * 0.1).to(bf16).cuda()
+    a1 = torch.randn(B, K).to(bf16).cuda().relu()  # ~half zeros
+    db_ref = torch.zeros(K).cuda()
+    dx2 = ext.linear_dx(dl, w)
+    dyeff_ref = ext.mask_db(dx2, a1, True, 0.5, db_ref)
+    db = torch.zeros(K).cuda()
+    dyeff = ext.linear_dx_mask(dl, w, a1, db, 0.5)
+    # same GEMM tile -> same acc; mask/scale applied at fp32 acc (fused) vs
+    # rounded bf16 (composition): one extra rounding, bf16-tolerance match
+    assert_close_bf16(dyeff, dyeff_ref, rtol=0.02,
+                      scale=float(dyeff_ref.float().abs().max()))
+    scale = float(db_ref.abs().max().clamp(min=1.0))
+    torch.testing.assert_close(db, db_ref, rtol=2e-2, atol=0.02 * scale)
+
+
+def test_sgd_zero_grad_after_consume(ext):
+    n = 100_003
+    master = torch.randn(n).cuda()
+    orig = master.clone()
+    grad = torch.randn(n).cuda()
+    gcopy = grad.clone()
+    shadow = torch.zeros(n, dtype=bf16).cuda()
+    lr = torch.tensor([0.05], dtype=torch.float32).cuda()
+    off = torch.tensor([0], dtype=torch.int64).cuda()
+    ext.sgd_step_dev(master, grad, shadow, True, lr, -1.0, 0, off,
+                     zero_grad=True)
+    torch.testing.assert_close(master, orig - 0.05 * gcopy)
+    assert float(grad.abs().max()) == 0.0, "grad bucket not zeroed"
+
+
 def test_grad_mask_gpu(ext):
     """Per-rank pre-aggregation drop-connect kernel: keep-rate, determinism,
     rank/step-distinct streams, and slice-wise composability (masking the

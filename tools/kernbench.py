@@ -85,9 +85,11 @@ def bench_batch(ext, B):
         ("fc1 fwd+relu+dropout (MFMA, wT)", lambda: ext.linear_act_fwd(h2, f1, fb1, True, 0.5, 1, 2, wT=f1T)),
         ("fc2 fwd (MFMA, wT)", lambda: ext.linear_act_fwd(a1, f2, fb2, False, 1.0, 0, 0, wT=f2T)),
         ("softmax-CE fused fwd+grad", lambda: ext.softmax_xent_fwd(logits, labels)),
+        ("softmax-CE +fc2 db fused", lambda: ext.softmax_xent_fwd(logits, labels, db_out=db2f)),
         ("mask+db (fc2)", lambda: ext.mask_db(dl, dl, False, 1.0, db2f)),
         ("fc2 dW", lambda: ext.linear_dw_into(a1, dyeff2, dw2f)),
         ("fc2 dX", lambda: ext.linear_dx(dyeff2, f2)),
+        ("fc2 dX+mask+db fused", lambda: ext.linear_dx_mask(dl, f2, a1, db1f, 0.5)),
         ("mask+db (fc1)", lambda: ext.mask_db(dx2, a1, True, 0.5, db1f)),
         ("fc1 dW (A_T split-K)", lambda: ext.linear_dw_into(h2, dyeff1, dw1f)),
         ("fc1 dX", lambda: ext.linear_dx(dyeff1, f1)),
