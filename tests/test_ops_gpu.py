@@ -265,7 +265,12 @@ def test_softmax_db_matches_mask_db(ext):
     l1, c1, dl1 = ext.softmax_xent_fwd(logits, labels, db_out=db)
     assert torch.equal(dl0, dl1)
     assert float(l0) == float(l1) and float(c0) == float(c1)
-    torch.testing.assert_close(db, db_ref, rtol=1e-3, atol=1e-5)
+    # fused db sums fp32 d-values; mask_db sums the bf16-ROUNDED dlogits —
+    # compare both to the exact fp32 column sum at bf16-rounding tolerance
+    db_exact = dl0.float().sum(dim=0)
+    scale = float(db_exact.abs().max().clamp(min=1e-3))
+    torch.testing.assert_close(db, db_exact, rtol=2e-2, atol=0.02 * scale)
+    torch.testing.assert_close(db_ref, db_exact, rtol=2e-2, atol=0.02 * scale)
 
 
 def test_linear_dx_mask_matches_composition(ext):
