@@ -223,7 +223,7 @@ void conv_fwd_slab_kernel(const u16* __restrict__ x,
 // conv dX: dx[hi,wi,ci] = sum_{kh,kw,co} dact[hi-kh+2, wi-kw+2, co] *
 //          w[kh,kw,ci,co] — dact slab per image, W read from global per tile.
 // ---------------------------------------------------------------------------
-template <int H, int W, int CIN, int COUT, int DB = 2>
+template <int H, int W, int CIN, int COUT, int DB = 2, int PAD = 8>
 __global__ __launch_bounds__(NTHREADS)
 void conv_dx_slab_kernel(const u16* __restrict__ dact,
                          const u16* __restrict__ w,  // [25*CIN][COUT]
@@ -235,7 +235,12 @@ void conv_dx_slab_kernel(const u16* __restrict__ dact,
   constexpr int WM = 112, WN = BN / 2; // 16
   constexpr int MI = WM / 16, NI = WN / 16;  // 7, 1
   constexpr int K = 25 * COUT;
-  constexpr int PST = COUT + 8;  // 144B pixel stride: conflict-free b128
+  // pixel stride: +8 pad = 144B (the original conflict-free choice); +4 =
+  // 136B keeps the b128 phases on distinct banks too (34-bank lane step,
+  // verified distinct across each 8-lane phase) and shrinks the slab
+  // 44.1 vs 46.7 KB — with DB=2's 8 KB of B buffers that is the difference
+  // between 2 and 3 resident blocks/CU
+  constexpr int PST = COUT + PAD;
   __shared__ __align__(16) u16 slab[HP * WP * PST];
   // weight tile: n-major rows of BK (w rows are already k-contiguous per
   // ci), double-buffered and staged by glds — one barrier per K-step, the
@@ -560,9 +565,14 @@ void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
     else if (getenv("DMNIST_DX_HIOCC"))
       hipLaunchKernelGGL((conv_dx_slab_hi_occ_kernel<14, 14, 32, 64>),
                          dim3(NB), dim3(NTHREADS), 0, s, dact, w, dx, NB);
-    else
+    else if (getenv("DMNIST_DX_DB1"))  // round-1 default for A/B
       hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 1>), dim3(NB),
                          dim3(NTHREADS), 0, s, dact, w, dx, NB);
+    else
+      // large NB: glds-pipelined DB=2 at the slim PAD=4 slab -> 3
+      // blocks/CU (PMC: the DB=1 serial drain parked waves 67% of cycles)
+      hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 2, 4>),
+                         dim3(NB), dim3(NTHREADS), 0, s, dact, w, dx, NB);
   }
 }
 

@@ -540,27 +540,34 @@ void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
         wv[r][2] = *reinterpret_cast<const unsigned*>(&slab[oy + r][ox + 4]);
       }
     }
-    float acc0 = b, acc1 = b, acc2 = b, acc3 = b;
+    // two partials per output (even/odd kh) -> 8 independent dot2 chains:
+    // a single chain per output serialized 15 dependent v_dot2c (~5-cycle
+    // RAW each — PMC showed 71% SQ_WAIT_INST_ANY); 8 chains interleave
+    // past the latency
+    float pe[4] = {b, 0.f, b, 0.f}, po[4] = {0.f, b, 0.f, b};
 #pragma unroll
     for (int kh = 0; kh < 5; ++kh) {
       unsigned o0 = __builtin_amdgcn_alignbit(wv[kh][1], wv[kh][0], 16);
       unsigned o1 = __builtin_amdgcn_alignbit(wv[kh][2], wv[kh][1], 16);
       unsigned p0 = __builtin_amdgcn_alignbit(wv[kh + 1][1], wv[kh + 1][0], 16);
       unsigned p1 = __builtin_amdgcn_alignbit(wv[kh + 1][2], wv[kh + 1][1], 16);
-      acc0 = dot2bf(wv[kh][0], wpk[kh][0],
-             dot2bf(wv[kh][1], wpk[kh][1],
-             dot2bf(wv[kh][2], wt4e[kh], acc0)));
-      acc1 = dot2bf(o0, wpk[kh][0],
-             dot2bf(o1, wpk[kh][1],
-             dot2bf(wv[kh][2], wt4o[kh], acc1)));
-      acc2 = dot2bf(wv[kh + 1][0], wpk[kh][0],
-             dot2bf(wv[kh + 1][1], wpk[kh][1],
-             dot2bf(wv[kh + 1][2], wt4e[kh], acc2)));
-      acc3 = dot2bf(p0, wpk[kh][0],
-             dot2bf(p1, wpk[kh][1],
-             dot2bf(wv[kh + 1][2], wt4o[kh], acc3)));
+      float* h0 = (kh & 1) ? po : pe;
+      h0[0] = dot2bf(wv[kh][0], wpk[kh][0],
+              dot2bf(wv[kh][1], wpk[kh][1],
+              dot2bf(wv[kh][2], wt4e[kh], h0[0])));
+      h0[1] = dot2bf(o0, wpk[kh][0],
+              dot2bf(o1, wpk[kh][1],
+              dot2bf(wv[kh][2], wt4o[kh], h0[1])));
+      float* h1 = (kh & 1) ? pe : po;
+      h1[2] = dot2bf(wv[kh + 1][0], wpk[kh][0],
+              dot2bf(wv[kh + 1][1], wpk[kh][1],
+              dot2bf(wv[kh + 1][2], wt4e[kh], h1[2])));
+      h1[3] = dot2bf(p0, wpk[kh][0],
+              dot2bf(p1, wpk[kh][1],
+              dot2bf(wv[kh + 1][2], wt4o[kh], h1[3])));
     }
-    float vals[4] = {acc0, acc1, acc2, acc3};
+    float vals[4] = {pe[0] + po[0], pe[1] + po[1], pe[2] + po[2],
+                     pe[3] + po[3]};
     float best = -1.0f / 0.0f;
     int barg = 0;
 #pragma unroll
@@ -760,15 +767,31 @@ void conv1_dw_pooled_kernel(const ushort_t* x, const ushort_t* dyp,
     __syncthreads();
     if (co < Cout) {
       const size_t ib = ((size_t)img * Hp) * Wp * Cout + co;
-      for (int rq = hw; rq < Hp; rq += 8) {
-        // decode one pooled row: packed (bf16 value | argmax<<16) per qx —
-        // 3 small strided streams instead of the dense dact row walk
-        unsigned pk[14];
+      // issue ALL of this lane's pooled loads up front (rows hw and hw+8):
+      // 56 independent loads in flight -> ONE latency wall per image
+      // instead of one per row (PMC: 55% of wave cycles were parked)
+      ushort_t rdy[2][14];
+      uint8_t ram[2][14];
+#pragma unroll
+      for (int rr = 0; rr < 2; ++rr) {
+        int rq = hw + rr * 8;
+        if (rq >= Hp) break;
 #pragma unroll
         for (int qx = 0; qx < 14; ++qx) {
           size_t o = ib + ((size_t)rq * Wp + qx) * Cout;
-          int pos = am[o];  // 0..3 live, 7 dead (liveness in the byte)
-          float gv = pos < 4 ? bf2f(dyp[o]) : 0.f;
+          rdy[rr][qx] = dyp[o];
+          ram[rr][qx] = am[o];
+        }
+      }
+      for (int rr = 0; rr < 2; ++rr) {
+        int rq = hw + rr * 8;
+        if (rq >= Hp) break;
+        // packed (bf16 value | argmax<<16) per qx, from the prefetched raws
+        unsigned pk[14];
+#pragma unroll
+        for (int qx = 0; qx < 14; ++qx) {
+          int pos = ram[rr][qx];  // 0..3 live, 7 dead (liveness in the byte)
+          float gv = pos < 4 ? bf2f(rdy[rr][qx]) : 0.f;
           pk[qx] = (unsigned)f2bf(gv) | ((unsigned)pos << 16);
         }
 #pragma unroll
