@@ -565,14 +565,15 @@ void launch_conv_dx_slab(const unsigned short* dact, const unsigned short* w,
     else if (getenv("DMNIST_DX_HIOCC"))
       hipLaunchKernelGGL((conv_dx_slab_hi_occ_kernel<14, 14, 32, 64>),
                          dim3(NB), dim3(NTHREADS), 0, s, dact, w, dx, NB);
-    else if (getenv("DMNIST_DX_DB1"))  // round-1 default for A/B
-      hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 1>), dim3(NB),
-                         dim3(NTHREADS), 0, s, dact, w, dx, NB);
-    else
-      // large NB: glds-pipelined DB=2 at the slim PAD=4 slab -> 3
-      // blocks/CU (PMC: the DB=1 serial drain parked waves 67% of cycles)
+    else if (getenv("DMNIST_DX_DB2SLIM"))
+      // glds-pipelined DB=2 at the slim PAD=4 slab (3 blocks/CU) —
+      // measured SLOWER than DB=1 at 8192 (357 vs 320 us solo) despite the
+      // occupancy win; kept for A/B
       hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 2, 4>),
                          dim3(NB), dim3(NTHREADS), 0, s, dact, w, dx, NB);
+    else
+      hipLaunchKernelGGL((conv_dx_slab_kernel<14, 14, 32, 64, 1>), dim3(NB),
+                         dim3(NTHREADS), 0, s, dact, w, dx, NB);
   }
 }
 

@@ -544,30 +544,43 @@ void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
     // a single chain per output serialized 15 dependent v_dot2c (~5-cycle
     // RAW each — PMC showed 71% SQ_WAIT_INST_ANY); 8 chains interleave
     // past the latency
-    float pe[4] = {b, 0.f, b, 0.f}, po[4] = {0.f, b, 0.f, b};
+    float a0e = b, a1e = 0.f, a2e = b, a3e = 0.f;
+    float a0o = 0.f, a1o = b, a2o = 0.f, a3o = b;
 #pragma unroll
     for (int kh = 0; kh < 5; ++kh) {
       unsigned o0 = __builtin_amdgcn_alignbit(wv[kh][1], wv[kh][0], 16);
       unsigned o1 = __builtin_amdgcn_alignbit(wv[kh][2], wv[kh][1], 16);
       unsigned p0 = __builtin_amdgcn_alignbit(wv[kh + 1][1], wv[kh + 1][0], 16);
       unsigned p1 = __builtin_amdgcn_alignbit(wv[kh + 1][2], wv[kh + 1][1], 16);
-      float* h0 = (kh & 1) ? po : pe;
-      h0[0] = dot2bf(wv[kh][0], wpk[kh][0],
+      if ((kh & 1) == 0) {
+        a0e = dot2bf(wv[kh][0], wpk[kh][0],
               dot2bf(wv[kh][1], wpk[kh][1],
-              dot2bf(wv[kh][2], wt4e[kh], h0[0])));
-      h0[1] = dot2bf(o0, wpk[kh][0],
+              dot2bf(wv[kh][2], wt4e[kh], a0e)));
+        a1e = dot2bf(o0, wpk[kh][0],
               dot2bf(o1, wpk[kh][1],
-              dot2bf(wv[kh][2], wt4o[kh], h0[1])));
-      float* h1 = (kh & 1) ? pe : po;
-      h1[2] = dot2bf(wv[kh + 1][0], wpk[kh][0],
+              dot2bf(wv[kh][2], wt4o[kh], a1e)));
+        a2o = dot2bf(wv[kh + 1][0], wpk[kh][0],
               dot2bf(wv[kh + 1][1], wpk[kh][1],
-              dot2bf(wv[kh + 1][2], wt4e[kh], h1[2])));
-      h1[3] = dot2bf(p0, wpk[kh][0],
+              dot2bf(wv[kh + 1][2], wt4e[kh], a2o)));
+        a3o = dot2bf(p0, wpk[kh][0],
               dot2bf(p1, wpk[kh][1],
-              dot2bf(wv[kh + 1][2], wt4o[kh], h1[3])));
+              dot2bf(wv[kh + 1][2], wt4o[kh], a3o)));
+      } else {
+        a0o = dot2bf(wv[kh][0], wpk[kh][0],
+              dot2bf(wv[kh][1], wpk[kh][1],
+              dot2bf(wv[kh][2], wt4e[kh], a0o)));
+        a1o = dot2bf(o0, wpk[kh][0],
+              dot2bf(o1, wpk[kh][1],
+              dot2bf(wv[kh][2], wt4o[kh], a1o)));
+        a2e = dot2bf(wv[kh + 1][0], wpk[kh][0],
+              dot2bf(wv[kh + 1][1], wpk[kh][1],
+              dot2bf(wv[kh + 1][2], wt4e[kh], a2e)));
+        a3e = dot2bf(p0, wpk[kh][0],
+              dot2bf(p1, wpk[kh][1],
+              dot2bf(wv[kh + 1][2], wt4o[kh], a3e)));
+      }
     }
-    float vals[4] = {pe[0] + po[0], pe[1] + po[1], pe[2] + po[2],
-                     pe[3] + po[3]};
+    float vals[4] = {a0e + a0o, a1e + a1o, a2e + a2o, a3e + a3o};
     float best = -1.0f / 0.0f;
     int barg = 0;
 #pragma unroll
@@ -783,6 +796,7 @@ void conv1_dw_pooled_kernel(const ushort_t* x, const ushort_t* dyp,
           ram[rr][qx] = am[o];
         }
       }
+#pragma unroll
       for (int rr = 0; rr < 2; ++rr) {
         int rq = hw + rr * 8;
         if (rq >= Hp) break;

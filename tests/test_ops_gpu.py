@@ -264,7 +264,9 @@ def test_softmax_db_matches_mask_db(ext):
     db = torch.zeros(10).cuda()
     l1, c1, dl1 = ext.softmax_xent_fwd(logits, labels, db_out=db)
     assert torch.equal(dl0, dl1)
-    assert float(l0) == float(l1) and float(c0) == float(c1)
+    # loss is an fp32 atomic sum over blocks: ordering varies run-to-run
+    assert abs(float(l0) - float(l1)) < 1e-5 * max(1.0, abs(float(l0)))
+    assert float(c0) == float(c1)  # whole-number sum: exact
     # fused db sums fp32 d-values; mask_db sums the bf16-ROUNDED dlogits —
     # compare both to the exact fp32 column sum at bf16-rounding tolerance
     db_exact = dl0.float().sum(dim=0)
