@@ -544,43 +544,31 @@ void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
     // a single chain per output serialized 15 dependent v_dot2c (~5-cycle
     // RAW each — PMC showed 71% SQ_WAIT_INST_ANY); 8 chains interleave
     // past the latency
-    float a0e = b, a1e = 0.f, a2e = b, a3e = 0.f;
-    float a0o = 0.f, a1o = b, a2o = 0.f, a3o = b;
+    // NOTE: an 8-chain even/odd-kh accumulator split was tried against the
+    // 71% SQ_WAIT_INST_ANY reading and measured SLOWER (38/232 us vs
+    // 34/206 at B=1024/8192) — the stall is not the acc RAW chain; the
+    // 4-chain nested-dot2 form below is the measured optimum
+    float acc0 = b, acc1 = b, acc2 = b, acc3 = b;
 #pragma unroll
     for (int kh = 0; kh < 5; ++kh) {
       unsigned o0 = __builtin_amdgcn_alignbit(wv[kh][1], wv[kh][0], 16);
       unsigned o1 = __builtin_amdgcn_alignbit(wv[kh][2], wv[kh][1], 16);
       unsigned p0 = __builtin_amdgcn_alignbit(wv[kh + 1][1], wv[kh + 1][0], 16);
       unsigned p1 = __builtin_amdgcn_alignbit(wv[kh + 1][2], wv[kh + 1][1], 16);
-      if ((kh & 1) == 0) {
-        a0e = dot2bf(wv[kh][0], wpk[kh][0],
-              dot2bf(wv[kh][1], wpk[kh][1],
-              dot2bf(wv[kh][2], wt4e[kh], a0e)));
-        a1e = dot2bf(o0, wpk[kh][0],
-              dot2bf(o1, wpk[kh][1],
-              dot2bf(wv[kh][2], wt4o[kh], a1e)));
-        a2o = dot2bf(wv[kh + 1][0], wpk[kh][0],
-              dot2bf(wv[kh + 1][1], wpk[kh][1],
-              dot2bf(wv[kh + 1][2], wt4e[kh], a2o)));
-        a3o = dot2bf(p0, wpk[kh][0],
-              dot2bf(p1, wpk[kh][1],
-              dot2bf(wv[kh + 1][2], wt4o[kh], a3o)));
-      } else {
-        a0o = dot2bf(wv[kh][0], wpk[kh][0],
-              dot2bf(wv[kh][1], wpk[kh][1],
-              dot2bf(wv[kh][2], wt4e[kh], a0o)));
-        a1o = dot2bf(o0, wpk[kh][0],
-              dot2bf(o1, wpk[kh][1],
-              dot2bf(wv[kh][2], wt4o[kh], a1o)));
-        a2e = dot2bf(wv[kh + 1][0], wpk[kh][0],
-              dot2bf(wv[kh + 1][1], wpk[kh][1],
-              dot2bf(wv[kh + 1][2], wt4e[kh], a2e)));
-        a3e = dot2bf(p0, wpk[kh][0],
-              dot2bf(p1, wpk[kh][1],
-              dot2bf(wv[kh + 1][2], wt4o[kh], a3e)));
-      }
+      acc0 = dot2bf(wv[kh][0], wpk[kh][0],
+             dot2bf(wv[kh][1], wpk[kh][1],
+             dot2bf(wv[kh][2], wt4e[kh], acc0)));
+      acc1 = dot2bf(o0, wpk[kh][0],
+             dot2bf(o1, wpk[kh][1],
+             dot2bf(wv[kh][2], wt4o[kh], acc1)));
+      acc2 = dot2bf(wv[kh + 1][0], wpk[kh][0],
+             dot2bf(wv[kh + 1][1], wpk[kh][1],
+             dot2bf(wv[kh + 1][2], wt4e[kh], acc2)));
+      acc3 = dot2bf(p0, wpk[kh][0],
+             dot2bf(p1, wpk[kh][1],
+             dot2bf(wv[kh + 1][2], wt4o[kh], acc3)));
     }
-    float vals[4] = {a0e + a0o, a1e + a1o, a2e + a2o, a3e + a3o};
+    float vals[4] = {acc0, acc1, acc2, acc3};
     float best = -1.0f / 0.0f;
     int barg = 0;
 #pragma unroll
