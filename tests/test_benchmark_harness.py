@@ -72,3 +72,33 @@ def test_run_cfg_cpu_2worker(tmp_path):
     assert res["max_step"] >= 3
     assert "examples_per_sec_per_worker_p50" in res
     assert os.path.exists(tmp_path / "out" / "results.json")
+
+
+@pytest.mark.timeout(300)
+def test_run_cfg_with_evaluator_and_plots(tmp_path):
+    """VERDICT round-1 task 7: run_cfg launches the evaluator per sweep
+    point, scrapes its 'Precision @ 1' lines into results, and plot_figs
+    emits the reference's full plot set (time->precision + time-CDF)."""
+    cfg = {
+        "name": "test_eval_w1",
+        "workers": 1,
+        "timeout_s": 240,
+        "evaluator": True,
+        "eval_interval_secs": 1,
+        "flags": {"model": "mlp", "batch_size": 16, "max_steps": 25,
+                  "synthetic_data": True, "device": "cpu",
+                  "worker_times_cdf_method": True,
+                  "train_dir": str(tmp_path / "td"),
+                  "save_interval_secs": 1},
+    }
+    cfg_path = tmp_path / "cfg.json"
+    cfg_path.write_text(json.dumps(cfg))
+    out = tmp_path / "out"
+    res = bm.run_cfg(str(cfg_path), outdir=str(out))
+    assert res["returncode"] == 0
+    assert os.path.exists(out / "out_evaluator")
+    assert "eval" in res, open(out / "out_evaluator").read()[-1500:]
+    assert res["eval"]["precisions"], res["eval"]
+    assert 0.0 <= res["eval"]["final_precision"] <= 1.0
+    bm.plot_figs([str(out)], str(tmp_path / "plots"))
+    assert os.path.exists(tmp_path / "plots" / "sweep.png")

@@ -152,11 +152,45 @@ def run_cfg(cfg_path, outdir=None):
     else:
         cmd = [sys.executable, entry] + argv
     master_log = os.path.join(outdir, "out_master")
+    # concurrent evaluator (reference run_tf launches one per sweep point,
+    # tf_ec2.py:445, and benchmark.py:54-58 downloads/scrapes out_evaluator)
+    eval_proc = None
+    eval_log_f = None
+    eval_log = os.path.join(outdir, "out_evaluator")
+    if cfg.get("evaluator"):
+        eargv = [sys.executable, os.path.join(ROOT, "src", "mnist_eval.py"),
+                 "--checkpoint_dir", flags["train_dir"],
+                 "--eval_dir", os.path.join(outdir, "eval_dir"),
+                 "--eval_interval_secs",
+                 str(cfg.get("eval_interval_secs", 2))]
+        for k in ("synthetic_data", "fake_data"):
+            if flags.get(k):
+                eargv.append(f"--{k}")
+        if "data_dir" in flags:
+            eargv += ["--data_dir", str(flags["data_dir"])]
+        if "model" in flags:
+            eargv += ["--model", str(flags["model"])]
+        eval_log_f = open(eval_log, "w")
+        eval_proc = subprocess.Popen(eargv, stdout=eval_log_f,
+                                     stderr=subprocess.STDOUT, cwd=ROOT)
     t0 = time.time()
     with open(master_log, "w") as lf:
         proc = subprocess.run(cmd, stdout=lf, stderr=subprocess.STDOUT,
                               timeout=cfg.get("timeout_s", 3600), cwd=ROOT)
     run_time = time.time() - t0
+    if eval_proc is not None:
+        # let the poller catch the final checkpoint, then stop THIS exact
+        # child (never pattern-kills)
+        deadline = time.time() + 2 * cfg.get("eval_interval_secs", 2) + 8
+        while time.time() < deadline and eval_proc.poll() is None:
+            time.sleep(0.5)
+        if eval_proc.poll() is None:
+            eval_proc.terminate()
+            try:
+                eval_proc.wait(15)
+            except subprocess.TimeoutExpired:
+                eval_proc.kill()
+        eval_log_f.close()
     results = {
         "name": name,
         "cfg": cfg,
@@ -181,6 +215,13 @@ def run_cfg(cfg_path, outdir=None):
         results["iteration_times_mean"] = (
             float(np.mean(extract_iteration_times(master_log)))
             if extract_iteration_times(master_log) else None)
+    if os.path.exists(eval_log):
+        times, losses, precs, steps = extract_times_losses_precision(eval_log)
+        if times:
+            results["eval"] = {
+                "times": times, "losses": losses, "precisions": precs,
+                "steps": steps, "final_precision": precs[-1],
+            }
     with open(os.path.join(outdir, "results.json"), "w") as f:
         json.dump(results, f, indent=2)
     with open(os.path.join(outdir, "results.txt"), "a") as f:
@@ -190,8 +231,10 @@ def run_cfg(cfg_path, outdir=None):
 
 
 def plot_figs(outdirs, dest):
-    """time->loss / step->loss / time->precision plots (benchmark.py:165-263)
-    when matplotlib is available."""
+    """The reference's full plot set (benchmark.py:165-263): step->loss,
+    step->train-acc, step->sec/batch, time->validation-precision,
+    time->validation-loss (both from the scraped out_evaluator lines) and
+    the per-worker compute-time CDF (from the ELAPSED TIMES report)."""
     try:
         import matplotlib
         matplotlib.use("Agg")
@@ -199,27 +242,38 @@ def plot_figs(outdirs, dest):
     except ImportError:
         print("matplotlib unavailable; skipping plots")
         return
-    fig, axes = plt.subplots(1, 3, figsize=(15, 4))
+    fig, axes = plt.subplots(2, 3, figsize=(16, 9))
+    (ax_sl, ax_sa, ax_sb), (ax_tp, ax_tl, ax_cdf) = axes
     for d in outdirs:
         master = os.path.join(d, "out_master")
         if not os.path.exists(master):
             continue
         recs = extract_step_records(master)
-        if not recs:
-            continue
         name = os.path.basename(d)
-        steps = [r[0] for r in recs]
-        losses = [r[1] for r in recs]
-        accs = [r[2] for r in recs]
-        axes[0].plot(steps, losses, label=name)
-        axes[1].plot(steps, accs, label=name)
-        spb = [r[4] for r in recs]
-        axes[2].plot(steps, spb, label=name)
-    axes[0].set_xlabel("step"); axes[0].set_ylabel("loss")
-    axes[1].set_xlabel("step"); axes[1].set_ylabel("train acc")
-    axes[2].set_xlabel("step"); axes[2].set_ylabel("sec/batch")
-    for ax in axes:
-        ax.legend(fontsize=6)
+        if recs:
+            steps = [r[0] for r in recs]
+            ax_sl.plot(steps, [r[1] for r in recs], label=name)
+            ax_sa.plot(steps, [r[2] for r in recs], label=name)
+            ax_sb.plot(steps, [r[4] for r in recs], label=name)
+        ev = os.path.join(d, "out_evaluator")
+        if os.path.exists(ev):
+            times, losses, precs, _steps = extract_times_losses_precision(ev)
+            if times:
+                ax_tp.plot(times, precs, marker="o", label=name)
+                ax_tl.plot(times, losses, marker="o", label=name)
+        ct = extract_compute_times(master)
+        if ct:
+            ts = np.sort(np.array([t for (t, _, _) in ct]))
+            ax_cdf.plot(ts, np.arange(1, len(ts) + 1) / len(ts), label=name)
+    ax_sl.set_xlabel("step"); ax_sl.set_ylabel("loss")
+    ax_sa.set_xlabel("step"); ax_sa.set_ylabel("train acc")
+    ax_sb.set_xlabel("step"); ax_sb.set_ylabel("sec/batch")
+    ax_tp.set_xlabel("time (s)"); ax_tp.set_ylabel("validation precision @ 1")
+    ax_tl.set_xlabel("time (s)"); ax_tl.set_ylabel("validation loss")
+    ax_cdf.set_xlabel("per-worker compute time (s)"); ax_cdf.set_ylabel("CDF")
+    for row in axes:
+        for ax in row:
+            ax.legend(fontsize=6)
     os.makedirs(dest, exist_ok=True)
     fig.savefig(os.path.join(dest, "sweep.png"), dpi=120)
     print(f"wrote {dest}/sweep.png")
