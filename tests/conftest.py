@@ -10,6 +10,9 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires an MI355X GPU (run on a gpurun box)")
+    config.addinivalue_line(
+        "markers", "realdata: needs the real MNIST idx-gz files dropped in "
+                   "data/ (skips cleanly otherwise; tests/test_realdata.py)")
 
 
 def pytest_collection_modifyitems(config, items):
