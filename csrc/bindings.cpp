@@ -451,7 +451,9 @@ torch::Tensor linear_dx_unpool(torch::Tensor dyeff, torch::Tensor w,
   p.CHo = Ho; p.CWo = Wo; p.Cout = C;
   p.amax = amax.data_ptr<uint8_t>();
   p.db = db_out.defined() ? db_out.data_ptr<float>() : nullptr;
-  bool big = cdiv(B, 128) * cdiv(K, 128) >= 128;
+  // 128-tile under ~2 blocks/CU is grid-starved (PMC: 57% parked at
+  // B=1024's 200-block grid); 64-tiles quadruple the grid
+  bool big = cdiv(B, 128) * cdiv(K, 128) >= 512;
   (big ? gemm_dx_unpool_128 : gemm_dx_unpool_64)(p, cur_stream());
   return dact;
 }
@@ -654,6 +656,29 @@ void transpose_bf16_batch(std::vector<torch::Tensor> srcs,
   launch_transpose_bf16_batch(d, (int)srcs.size(), cur_stream());
 }
 
+void transpose_bf16_batch_adv(std::vector<torch::Tensor> srcs,
+                              std::vector<torch::Tensor> dsts,
+                              torch::Tensor step_dev,
+                              torch::Tensor lr_scale_dev, double lr0,
+                              double decay, int64_t decay_steps,
+                              double inv_contrib) {
+  // graph-tail variant: the batched transposes + the device step/LR
+  // advance in ONE dispatch (step_advance folded into the last launch)
+  TORCH_CHECK(srcs.size() == dsts.size() && srcs.size() <= 4,
+              "up to 4 transpose pairs");
+  TransposeDesc d[4];
+  for (size_t i = 0; i < srcs.size(); ++i) {
+    CHECK_BF16(srcs[i]); CHECK_CONTIG(srcs[i]);
+    CHECK_BF16(dsts[i]); CHECK_CONTIG(dsts[i]);
+    d[i] = {bf16_ptr(srcs[i]), bf16_mut(dsts[i]),
+            (int)srcs[i].size(0), (int)srcs[i].size(1)};
+  }
+  launch_transpose_bf16_batch_adv(
+      d, (int)srcs.size(), step_dev.data_ptr<long>(),
+      lr_scale_dev.data_ptr<float>(), (float)lr0, (float)decay,
+      (int)decay_steps, (float)inv_contrib, cur_stream());
+}
+
 void transpose_bf16(torch::Tensor src, torch::Tensor dst) {
   CHECK_CUDA(src); CHECK_BF16(src); CHECK_CONTIG(src);
   CHECK_BF16(dst); CHECK_CONTIG(dst);
@@ -695,6 +720,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("p_keep"), py::arg("seed"), py::arg("offset"),
         py::arg("wT") = c10::nullopt);
   m.def("transpose_bf16", &transpose_bf16, "bf16 2-D transpose (wT refresh)");
+  m.def("transpose_bf16_batch_adv", &transpose_bf16_batch_adv,
+        "batched transposes + fused device step/LR advance (graph tail)");
   m.def("transpose_bf16_batch", &transpose_bf16_batch,
         "up to 4 transposes in one launch");
   m.def("mask_db", &mask_db, "relu/dropout grad mask + bias-grad column sums");

@@ -87,9 +87,19 @@ struct TransposeBatchArgs {
   TransposeDesc d[4];
   int tile0[4];
   int n, total;
+  // optional fused step/LR advance (graph tail: saves one dispatch)
+  int do_advance;
+  long* step_dev;
+  float* lr_scale_dev;
+  float lr0, decay, inv_contrib;
+  int decay_steps;
 };
 void launch_transpose_bf16_batch(const TransposeDesc* descs, int n,
                                  hipStream_t);
+void launch_transpose_bf16_batch_adv(const TransposeDesc* descs, int n,
+                                     long* step_dev, float* lr_scale_dev,
+                                     float lr0, float decay, int decay_steps,
+                                     float inv_contrib, hipStream_t);
 void launch_conv1_direct_fwd(const unsigned short* x, const unsigned short* w,
                              const float* bias, unsigned short* y,
                              uint8_t* amax, int NB, int H, int W, int Cout,

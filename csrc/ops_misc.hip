@@ -258,10 +258,12 @@ extern "C" __global__ void step_advance_kernel(long* step_dev,
                                                float decay, int decay_steps,
                                                float inv_contrib) {
   if (threadIdx.x == 0 && blockIdx.x == 0) {
-    long s = *step_dev;
+    // prime step s+1: the staircase exponent must be the NEXT step's
+    // (eager lr_at(step) parity at decay boundaries)
+    long s = *step_dev + 1;
     float lr = lr0 * __powf(decay, (float)(s / decay_steps));
     *lr_scale_dev = lr * inv_contrib;
-    *step_dev = s + 1;
+    *step_dev = s;
   }
 }
 
@@ -271,6 +273,15 @@ extern "C" __global__ __launch_bounds__(256)
 void transpose_bf16_batch_kernel(TransposeBatchArgs a) {
   __shared__ ushort_t tile[32][33];
   int t = blockIdx.x;
+  if (a.do_advance && t == 0 && threadIdx.x == 0) {
+    // fold the step/LR advance into this (last-in-tail) launch: one fewer
+    // dispatch per replayed graph; nothing later in THIS replay reads
+    // step_dev/lr_scale_dev, so ordering vs the transposes is free
+    long sp = *a.step_dev + 1;
+    float lr = a.lr0 * __powf(a.decay, (float)(sp / a.decay_steps));
+    *a.lr_scale_dev = lr * a.inv_contrib;
+    *a.step_dev = sp;
+  }
   int which = 0;
   while (which + 1 < a.n && t >= a.tile0[which + 1]) ++which;
   int local = t - a.tile0[which];
@@ -589,8 +600,7 @@ void launch_transpose_bf16(const unsigned short* src, unsigned short* dst,
                      R, C);
 }
 
-void launch_transpose_bf16_batch(const TransposeDesc* descs, int n,
-                                 hipStream_t s) {
+static TransposeBatchArgs make_batch_args(const TransposeDesc* descs, int n) {
   TransposeBatchArgs a{};
   int total = 0;
   for (int i = 0; i < n && i < 4; ++i) {
@@ -600,7 +610,29 @@ void launch_transpose_bf16_batch(const TransposeDesc* descs, int n,
   }
   a.n = n < 4 ? n : 4;
   a.total = total;
-  hipLaunchKernelGGL(transpose_bf16_batch_kernel, dim3(total), dim3(256), 0,
+  return a;
+}
+
+void launch_transpose_bf16_batch(const TransposeDesc* descs, int n,
+                                 hipStream_t s) {
+  TransposeBatchArgs a = make_batch_args(descs, n);
+  hipLaunchKernelGGL(transpose_bf16_batch_kernel, dim3(a.total), dim3(256), 0,
+                     s, a);
+}
+
+void launch_transpose_bf16_batch_adv(const TransposeDesc* descs, int n,
+                                     long* step_dev, float* lr_scale_dev,
+                                     float lr0, float decay, int decay_steps,
+                                     float inv_contrib, hipStream_t s) {
+  TransposeBatchArgs a = make_batch_args(descs, n);
+  a.do_advance = 1;
+  a.step_dev = step_dev;
+  a.lr_scale_dev = lr_scale_dev;
+  a.lr0 = lr0;
+  a.decay = decay;
+  a.decay_steps = decay_steps < 1 ? 1 : decay_steps;
+  a.inv_contrib = inv_contrib;
+  hipLaunchKernelGGL(transpose_bf16_batch_kernel, dim3(a.total), dim3(256), 0,
                      s, a);
 }
 

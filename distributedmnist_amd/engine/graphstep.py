@@ -253,12 +253,22 @@ class GraphedStep:
                                t.flags.seed, self.step_dev,
                                momentum=t.flat_momentum,
                                mu=t.flags.momentum, zero_grad=True)
-        fp.refresh_transposes()
-        # advance step + LR on-device for the next body execution
-        self._ext.step_advance(self.step_dev, self.lr_scale_dev,
-                               t.flags.initial_learning_rate,
-                               t.flags.learning_rate_decay_factor,
-                               self.decay_steps, self.inv_contrib)
+        if fp._t_pairs:
+            # transposes + step/LR advance in ONE dispatch
+            self._ext.transpose_bf16_batch_adv(
+                [v2 for v2, _ in fp._t_pairs],
+                [tt for _, tt in fp._t_pairs],
+                self.step_dev, self.lr_scale_dev,
+                t.flags.initial_learning_rate,
+                t.flags.learning_rate_decay_factor,
+                self.decay_steps, self.inv_contrib)
+        else:
+            fp.refresh_transposes()
+            # advance step + LR on-device for the next body execution
+            self._ext.step_advance(self.step_dev, self.lr_scale_dev,
+                                   t.flags.initial_learning_rate,
+                                   t.flags.learning_rate_decay_factor,
+                                   self.decay_steps, self.inv_contrib)
 
     def _body(self):
         t = self.t
