@@ -52,9 +52,11 @@ def main():
     args = ap.parse_args()
     on_gpu_probe = torch.cuda.is_available()
     if args.steps is None:
-        args.steps = 2000 if on_gpu_probe else 40
+        # multi-second timed region by default so SMI utilization sampling
+        # sees the run (round-1 BENCH carried gpu_busy=0.0 off one sample)
+        args.steps = 5000 if on_gpu_probe else 40
     if args.warmup is None:
-        args.warmup = 200 if on_gpu_probe else 10
+        args.warmup = 500 if on_gpu_probe else 10
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))

@@ -164,3 +164,37 @@ def test_two_graph_split_matches_eager_numerics(tmp_path):
     loss_e, _ = run(29393, ["--hip_graph", "off"])
     assert abs(loss_g - loss_e) < 0.05 * max(1.0, abs(loss_e)), \
         (loss_g, loss_e)
+
+
+@pytest.mark.gpu
+def test_graph_drop_connect_matches_eager(tmp_path):
+    """Per-rank pre-aggregation drop-connect INSIDE the two-graph capture
+    (grad_mask keyed on step_dev) must track the eager path's trajectory:
+    same philox stream (seed, step, rank), so losses match within
+    fp32-atomic noise."""
+    import re
+
+    def run(port, extra):
+        env = dict(os.environ)
+        env["DMNIST_BACKEND"] = "gloo"
+        out = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", str(port),
+             os.path.join(ROOT, "src", "mnist_distributed_train.py"),
+             "--synthetic_data", "--backend", "gloo", "--batch_size", "128",
+             "--max_steps", "25", "--save_interval_secs", "100000",
+             "--drop_connect",
+             "--train_dir", str(tmp_path / f"dc{port}")] + extra,
+            capture_output=True, text=True, timeout=280, env=env, cwd=ROOT)
+        assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+        txt = out.stdout + out.stderr
+        losses = re.findall(r"Worker 0: .*step 24, loss = ([0-9.]+)", txt)
+        assert losses, txt[-1500:]
+        return float(losses[0]), txt
+
+    loss_g, txt_g = run(29394, [])
+    assert "running eager" not in txt_g
+    loss_e, _ = run(29395, ["--hip_graph", "off"])
+    assert abs(loss_g - loss_e) < 0.05 * max(1.0, abs(loss_e)), \
+        (loss_g, loss_e)
