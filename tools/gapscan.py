@@ -26,11 +26,22 @@ def main():
     if not rows:
         print("no dispatches")
         return
-    # restrict to the steady-state middle half of the trace
-    t0 = rows[0][0]
-    t1 = rows[-1][1]
-    lo = t0 + (t1 - t0) // 4
-    hi = t1 - (t1 - t0) // 4
+    # steady state = the tightest contiguous run of steps, located by the
+    # one-per-step SGD apply (startup/capture/sampling phases have huge
+    # host-side gaps that would swamp the replay-gap signal)
+    sgd = [s for (s, e, n) in rows if n.startswith("sgd_step_kernel")]
+    if len(sgd) > 25:
+        run = 20
+        best = min(range(len(sgd) - run),
+                   key=lambda i: sgd[i + run] - sgd[i])
+        lo, hi = sgd[best], sgd[best + run]
+        nsteps = run
+        print(f"steady window: {run} steps, "
+              f"{(hi - lo) / 1e3 / run:.1f} us/step wall")
+    else:
+        t0, t1 = rows[0][0], rows[-1][1]
+        lo = t0 + (t1 - t0) // 4
+        hi = t1 - (t1 - t0) // 4
     rows = [r for r in rows if r[0] >= lo and r[1] <= hi]
     # merge busy intervals
     busy = 0
