@@ -34,7 +34,13 @@ class FusedLeNetStep:
         assert m.shadows and m.shadows_T, "fused step needs bf16 shadows"
         self.t = t
         self.ext = _C.ext()
-        self.side = torch.cuda.Stream()
+        # DMNIST_SINGLE_STREAM=1: run the dW chain on the main stream too —
+        # ablation for the cross-stream fork/join bubbles a captured graph
+        # pays per edge (gapscan: ~57 us/step constant) vs the overlap win
+        import os as _os
+        self.single_stream = bool(_os.environ.get("DMNIST_SINGLE_STREAM"))
+        self.side = (torch.cuda.current_stream() if self.single_stream
+                     else torch.cuda.Stream())
         self.p_keep = 0.5
         self.seed = t.flags.seed
         # bucketed all-reduce: the fc gradients (96.5% of the payload,
@@ -47,6 +53,12 @@ class FusedLeNetStep:
         self.conv_slice = fp.flat_grad[:fc0]
         self.fc_slice = fp.flat_grad[fc0:]
         self.overlap_allreduce = t.world > 1 and dist.is_initialized()
+
+    def _fork(self, a, b):
+        """a.wait_stream(b), elided in single-stream mode (each edge is a
+        semaphore pair in the captured graph)."""
+        if not self.single_stream:
+            a.wait_stream(b)
 
     def __call__(self, x, labels, step_dev):
         """One fwd+bwd; gradients land in the flat bucket.
@@ -78,14 +90,14 @@ class FusedLeNetStep:
                                                  db_out=gv("fc2_b"))
 
         # ---- backward: dX chain on s0, dW GEMMs on s1 ----
-        s1.wait_stream(s0)
+        self._fork(s1, s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(a1, dl, gv("fc2_w"))
         # fc2 dX with the fc1 relu+dropout mask folded into the epilogue
         # (+ fc1 bias grad) — replaces linear_dx + mask_db
         dyeff1 = ext.linear_dx_mask(dl, sh["fc2_w"], a1, gv("fc1_b"),
                                     self.p_keep)
-        s1.wait_stream(s0)
+        self._fork(s1, s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
             if self.overlap_allreduce:
@@ -98,7 +110,7 @@ class FusedLeNetStep:
         # liveness rides in the amax byte, no y2 re-read)
         dact2 = ext.linear_dx_unpool(dyeff1, sh["fc1_w"], am2,
                                      gv("conv2_b"), 7, 7, 64)
-        s1.wait_stream(s0)
+        self._fork(s1, s0)
         with torch.cuda.stream(s1):
             ext.conv_dw_into(y1, dact2, gv("conv2_w"))
         dxc = ext.conv_dx(dact2, sh["conv2_w"], 32)
@@ -107,7 +119,7 @@ class FusedLeNetStep:
         # the consumer; the 4x-size dact1 is never materialized)
         ext.conv1_dw_pooled(x, dxc, am1, gv("conv1_w"), gv("conv1_b"))
 
-        s0.wait_stream(s1)
+        self._fork(s0, s1)
         if self.overlap_allreduce:
             self.t.engine.wire_allreduce(self.conv_slice, start=0)
         # keep the side-stream consumers alive until the join (capture-safe)
@@ -145,12 +157,12 @@ class FusedLeNetStep:
                                     0, 0, shT["fc2_w"])
         loss, correct, dl = ext.softmax_xent_fwd(logits, labels,
                                                  db_out=gv("fc2_b"))
-        s1.wait_stream(s0)
+        self._fork(s1, s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(a1, dl, gv("fc2_w"))
         dyeff1 = ext.linear_dx_mask(dl, sh["fc2_w"], a1, gv("fc1_b"),
                                     self.p_keep)
-        s1.wait_stream(s0)
+        self._fork(s1, s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(h2, dyeff1, gv("fc1_w"))
         # fused dX+pool2-backward: conv2_b lands here (conv slice — reduced
@@ -172,10 +184,10 @@ class FusedLeNetStep:
         x, y1, am1, dact2 = self._stash
         s0 = torch.cuda.current_stream()
         s1 = self.side
-        s1.wait_stream(s0)
+        self._fork(s1, s0)
         with torch.cuda.stream(s1):
             ext.conv_dw_into(y1, dact2, gv("conv2_w"))
         dxc = ext.conv_dx(dact2, m.shadows["conv2_w"], 32)
         ext.conv1_dw_pooled(x, dxc, am1, gv("conv1_w"), gv("conv1_b"))
-        s0.wait_stream(s1)
+        self._fork(s0, s1)
         self._keep2 = (dact2,)
