@@ -1,19 +1,22 @@
 """Hand-scheduled LeNet training step (GPU bf16 path).
 
-Replaces autograd's serial backward with an explicit schedule on two HIP
-streams: the dX chain stays on the main stream while every dW GEMM (which
-only the optimizer needs) runs concurrently on a side stream — ~30% of the
-backward is off the critical path.  Used inside the hipGraph capture
-(graphstep.py); stream forks/joins are recorded as graph dependencies.
+Replaces autograd's serial backward with an explicit schedule; used inside
+the hipGraph capture (graphstep.py).
 
-Backward dataflow (s0 = main, s1 = side):
+Backward dataflow (round 2, pool backward fused away):
 
-  dl ->[mask_db: db2]-> dyeff2 --s1--> dW2
-        |--s0--> dx2 ->[mask_db: db1]-> dyeff1 --s1--> dW1
-                  |--s0--> dx1 ->[pool_scatter: dbc2]-> dact2 --s1--> conv2 dW
-                            |--s0--> conv_dx -> [pool_scatter: dbc1]-> dact1
-                                      |--s0--> conv1 dW
-All dW/db land directly in the flat fp32 all-reduce bucket (pre-zeroed).
+  softmax(+db2) -> dl --> dW2
+     dl --> dx2+mask+db1 = dyeff1 --> dW1
+              dyeff1 --> dX+unpool = dact2 (+db_conv2) --> conv2 dW
+                          dact2 --> conv_dx = dxc --> conv1 dW+db (pooled)
+All dW/db land directly in the flat fp32 all-reduce bucket (zeroed by the
+SGD tail of the previous step).
+
+Round-1 ran the dW chain on a SIDE stream overlapping the dX chain; round-2
+measurement showed the single-stream schedule is 25-40% faster end-to-end
+(cross-stream graph-edge semaphores + CU contention on the dX chain
+outweighed the overlap) — see FusedLeNetStep.__init__.  DMNIST_TWO_STREAM=1
+restores the overlapped schedule.
 """
 
 from __future__ import annotations
@@ -34,11 +37,15 @@ class FusedLeNetStep:
         assert m.shadows and m.shadows_T, "fused step needs bf16 shadows"
         self.t = t
         self.ext = _C.ext()
-        # DMNIST_SINGLE_STREAM=1: run the dW chain on the main stream too —
-        # ablation for the cross-stream fork/join bubbles a captured graph
-        # pays per edge (gapscan: ~57 us/step constant) vs the overlap win
+        # SINGLE-STREAM schedule is the measured default: the two-stream
+        # dX/dW overlap looked right on paper but lost 25-40% end-to-end —
+        # cross-stream fork/join edges in the replayed graph cost ~15 us
+        # each AND the dW kernels' CU contention inflated the dX chain far
+        # past the overlap win (conv_dx 50 us solo vs 94 contended; same-box
+        # A/B: 0.445 -> 0.319 ms @B=1024, 1.81 -> 1.43 ms @8192).
+        # DMNIST_TWO_STREAM=1 restores the overlapped schedule for A/B.
         import os as _os
-        self.single_stream = bool(_os.environ.get("DMNIST_SINGLE_STREAM"))
+        self.single_stream = not _os.environ.get("DMNIST_TWO_STREAM")
         self.side = (torch.cuda.current_stream() if self.single_stream
                      else torch.cuda.Stream())
         self.p_keep = 0.5
