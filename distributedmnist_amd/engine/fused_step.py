@@ -46,8 +46,12 @@ class FusedLeNetStep:
         # DMNIST_TWO_STREAM=1 restores the overlapped schedule for A/B.
         import os as _os
         self.single_stream = not _os.environ.get("DMNIST_TWO_STREAM")
-        self.side = (torch.cuda.current_stream() if self.single_stream
-                     else torch.cuda.Stream())
+        # side stream resolved PER CALL: in single-stream mode it must be
+        # whatever stream the body is running on RIGHT NOW (warmup and
+        # graph capture run on their own streams — freezing the init-time
+        # stream here sent the dW ops to a foreign stream with no
+        # dependency edges: silent gradient races)
+        self.side = None if self.single_stream else torch.cuda.Stream()
         self.p_keep = 0.5
         self.seed = t.flags.seed
         # bucketed all-reduce: the fc gradients (96.5% of the payload,
@@ -79,7 +83,7 @@ class FusedLeNetStep:
             return getattr(m, name).grad
 
         s0 = torch.cuda.current_stream()
-        s1 = self.side
+        s1 = s0 if self.single_stream else self.side
 
         # ---- forward (s0) ----
         y1, am1 = ext.conv_pool_fwd(x, sh["conv1_w"], m.conv1_b, None)
@@ -152,7 +156,7 @@ class FusedLeNetStep:
             return getattr(m, name).grad
 
         s0 = torch.cuda.current_stream()
-        s1 = self.side
+        s1 = s0 if self.single_stream else self.side
         y1, am1 = ext.conv_pool_fwd(x, sh["conv1_w"], m.conv1_b, None)
         y2, am2 = ext.conv_pool_fwd(y1, sh["conv2_w"], m.conv2_b,
                                     shT["conv2_w"])
@@ -190,7 +194,7 @@ class FusedLeNetStep:
 
         x, y1, am1, dact2 = self._stash
         s0 = torch.cuda.current_stream()
-        s1 = self.side
+        s1 = s0 if self.single_stream else self.side
         self._fork(s1, s0)
         with torch.cuda.stream(s1):
             ext.conv_dw_into(y1, dact2, gv("conv2_w"))
