@@ -37,15 +37,20 @@ class FusedLeNetStep:
         assert m.shadows and m.shadows_T, "fused step needs bf16 shadows"
         self.t = t
         self.ext = _C.ext()
-        # SINGLE-STREAM schedule is the measured default: the two-stream
-        # dX/dW overlap looked right on paper but lost 25-40% end-to-end —
-        # cross-stream fork/join edges in the replayed graph cost ~15 us
-        # each AND the dW kernels' CU contention inflated the dX chain far
-        # past the overlap win (conv_dx 50 us solo vs 94 contended; same-box
-        # A/B: 0.445 -> 0.319 ms @B=1024, 1.81 -> 1.43 ms @8192).
-        # DMNIST_TWO_STREAM=1 restores the overlapped schedule for A/B.
+        # Stream schedule is BATCH-TIERED (same-box A/B): at B=1024 the
+        # cross-stream fork/join edges of the replayed graph (~15 us each)
+        # plus the dW kernels' CU contention on the dX chain cost more than
+        # the overlap buys (0.429 two-stream vs 0.400 ms single); at B=8192
+        # the hidden dW work dominates the edge cost (1.698 vs 1.733) —
+        # single-stream below 4096, overlapped above.  DMNIST_TWO_STREAM /
+        # DMNIST_SINGLE_STREAM force either for A/B.
         import os as _os
-        self.single_stream = not _os.environ.get("DMNIST_TWO_STREAM")
+        if _os.environ.get("DMNIST_TWO_STREAM"):
+            self.single_stream = False
+        elif _os.environ.get("DMNIST_SINGLE_STREAM"):
+            self.single_stream = True
+        else:
+            self.single_stream = t.flags.batch_size < 4096
         # side stream resolved PER CALL: in single-stream mode it must be
         # whatever stream the body is running on RIGHT NOW (warmup and
         # graph capture run on their own streams — freezing the init-time
