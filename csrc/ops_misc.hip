@@ -893,6 +893,8 @@ void launch_conv1_dw_pooled(const unsigned short* x, const unsigned short* dyp,
                             const uint8_t* am, float* dw, float* db, int NB,
                             int H, int W, int Cout, hipStream_t s) {
   int G = NB >= 8192 ? 8 : (NB >= 2048 ? NB / 1024 : 1);
+  if (const char* e = getenv("DMNIST_DW1_G")) G = atoi(e);  // flush sweep
+  if (G < 1) G = 1;
   int blocks = (NB + G - 1) / G;
   hipLaunchKernelGGL(conv1_dw_pooled_kernel, dim3(blocks), dim3(256), 0, s,
                      x, dyp, am, dw, db, NB, H, W, Cout, G);
