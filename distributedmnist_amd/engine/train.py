@@ -120,6 +120,9 @@ class Trainer:
         self._num_examples = 60000  # overwritten by train(); LR-schedule default
         self._graph = None
         self._graph_tried = False
+        self._eager_fused = None
+        self._eager_fused_tried = False
+        self._step_dev = None
         self.flat_momentum = (torch.zeros_like(self.fp.flat_master)
                               if flags.momentum > 0 else None)
         # parameter-init parity across ranks (broadcast once)
@@ -198,6 +201,26 @@ class Trainer:
             return True, loss, acc, 0.0
         return self.train_step(images, labels)
 
+    def _get_eager_fused(self):
+        """FusedLeNetStep for the eager path (non-graphable modes), or None
+        (CPU / non-LeNet / fp32 / DMNIST_EAGER_FUSED=0)."""
+        if self._eager_fused_tried:
+            return self._eager_fused
+        self._eager_fused_tried = True
+        from ..models import LeNet5
+        if (self.device.type == "cuda"
+                and isinstance(self.model, LeNet5)
+                and self.model.shadows and self.model.shadows_T
+                and getattr(self.flags, "fused_step", "auto") != "off"
+                and os.environ.get("DMNIST_EAGER_FUSED", "1") != "0"):
+            from .fused_step import FusedLeNetStep
+            f = FusedLeNetStep(self)
+            f.overlap_allreduce = False  # SyncEngine owns the collective
+            self._eager_fused = f
+            self._step_dev = torch.zeros(1, dtype=torch.int64,
+                                         device=self.device)
+        return self._eager_fused
+
     @property
     def _needs_step_timing(self) -> bool:
         """Host-synchronizing per-step timers are only needed for the
@@ -220,10 +243,20 @@ class Trainer:
         if flags.inject_slow_rank == self.rank and flags.inject_slow_ms > 0:
             time.sleep(flags.inject_slow_ms / 1000.0)
         self.fp.zero_grad()
-        logits = self.model(images, train=True)
-        loss, acc = self.model.loss_and_accuracy(logits, labels)
-        loss.backward()
-        self.fp.fix_grad_views()
+        fused = self._get_eager_fused()
+        if fused is not None:
+            # hand-scheduled fused step in EAGER mode too: the instrumented
+            # modes (cdf / k_of_n / interval / straggler) get the same
+            # kernel path as the captured graph; the collective stays with
+            # SyncEngine.reduce below (fused.overlap_allreduce disabled)
+            self._step_dev.fill_(self.step)
+            loss, correct = fused(images, labels, self._step_dev)
+            acc = correct / images.shape[0]
+        else:
+            logits = self.model(images, train=True)
+            loss, acc = self.model.loss_and_accuracy(logits, labels)
+            loss.backward()
+            self.fp.fix_grad_views()
         if self._dc_pre:
             Fx.grad_mask(self.fp.flat_grad, flags.drop_connect_probability,
                          flags.seed, self.step, self.rank)
