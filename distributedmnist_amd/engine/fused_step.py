@@ -12,11 +12,11 @@ Backward dataflow (round 2, pool backward fused away):
 All dW/db land directly in the flat fp32 all-reduce bucket (zeroed by the
 SGD tail of the previous step).
 
-Round-1 ran the dW chain on a SIDE stream overlapping the dX chain; round-2
-measurement showed the single-stream schedule is 25-40% faster end-to-end
-(cross-stream graph-edge semaphores + CU contention on the dX chain
-outweighed the overlap) — see FusedLeNetStep.__init__.  DMNIST_TWO_STREAM=1
-restores the overlapped schedule.
+Round-1 ran the dW chain on a SIDE stream overlapping the dX chain;
+round-2 same-box A/B showed the single-stream schedule is ~7% faster
+end-to-end at B<=2048 (cross-stream graph-edge semaphores + CU contention
+on the dX chain outweigh the overlap there) while the overlap still wins at
+B>=4096 — the schedule is batch-tiered (FusedLeNetStep.__init__).
 """
 
 from __future__ import annotations
@@ -37,10 +37,11 @@ class FusedLeNetStep:
         assert m.shadows and m.shadows_T, "fused step needs bf16 shadows"
         self.t = t
         self.ext = _C.ext()
-        # Stream schedule is BATCH-TIERED (same-box A/B): at B=1024 the
-        # cross-stream fork/join edges of the replayed graph (~15 us each)
-        # plus the dW kernels' CU contention on the dX chain cost more than
-        # the overlap buys (0.429 two-stream vs 0.400 ms single); at B=8192
+        # Stream schedule is BATCH-TIERED (same-box A/B, after the
+        # per-call side-stream fix): at B=1024 the cross-stream fork/join
+        # edges of the replayed graph plus the dW kernels' CU contention on
+        # the dX chain cost more than the overlap buys (0.429 two-stream vs
+        # 0.400 ms single; 0.384 after the conv1-dW G retune); at B=8192
         # the hidden dW work dominates the edge cost (1.698 vs 1.733) —
         # single-stream below 4096, overlapped above.  DMNIST_TWO_STREAM /
         # DMNIST_SINGLE_STREAM force either for A/B.
