@@ -581,9 +581,12 @@ torch::Tensor conv_dx(torch::Tensor dact, torch::Tensor w, int64_t Cin) {
 // --------------------------------------------------------------------------
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
                                             torch::Tensor labels,
-                                            c10::optional<torch::Tensor> db_out) {
+                                            c10::optional<torch::Tensor> db_out,
+                                            double inv_n) {
   // db_out (optional): fc2 bias grad accumulated in the same pass (column
-  // sums of dlogits) — removes the standalone mask_db launch
+  // sums of dlogits) — removes the standalone mask_db launch.
+  // inv_n: scale on the correct-count output (pass 1/B to get the MEAN
+  // accuracy straight out of the kernel; default 1.0 = raw count)
   CHECK_CUDA(logits); CHECK_BF16(logits); CHECK_CONTIG(logits);
   TORCH_CHECK(labels.scalar_type() == at::kLong, "labels must be int64");
   CHECK_CONTIG(labels);
@@ -596,7 +599,7 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
     dbp = db_out->data_ptr<float>();
   launch_softmax_xent(bf16_ptr(logits), labels.data_ptr<long>(),
                       bf16_mut(dl), out.data_ptr<float>(), B, C, dbp,
-                      cur_stream());
+                      (float)inv_n, cur_stream());
   auto loss = out.select(0, 0);
   auto correct = out.select(0, 1);
   return {loss, correct, dl};
@@ -750,9 +753,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_pool_bwd", &conv_pool_bwd, "conv+pool backward (dx, dw, db)");
   m.def("conv_pool_bwd_into", &conv_pool_bwd_into,
         "conv backward accumulating dw/db into bucket views");
-  m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax-CE (+grad, +optional fc2 db)",
+  m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax-CE (+grad, +optional fc2 db, +optional acc-mean scale)",
         py::arg("logits"), py::arg("labels"),
-        py::arg("db_out") = c10::nullopt);
+        py::arg("db_out") = c10::nullopt, py::arg("inv_n") = 1.0);
   m.def("sgd_step", &sgd_step, "fused flat SGD(+momentum) apply",
         py::arg("master"), py::arg("grad"), py::arg("shadow"),
         py::arg("has_shadow"), py::arg("lr"), py::arg("scale"),

@@ -55,7 +55,7 @@ void launch_pool_bwd_scatter(const unsigned short* dy, const unsigned short* y,
                              hipStream_t);
 void launch_softmax_xent(const unsigned short* logits, const long* labels,
                          unsigned short* dlogits, float* out, int B, int C,
-                         float* db, hipStream_t);
+                         float* db, float inv_n, hipStream_t);
 void launch_sgd_step(float* master, float* grad, unsigned short* shadow,
                      int has_shadow, long n, float lr_scale, float dc_keep,
                      uint64_t seed, uint64_t offset, float* momentum, float mu,

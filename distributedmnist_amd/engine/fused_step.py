@@ -79,7 +79,7 @@ class FusedLeNetStep:
 
     def __call__(self, x, labels, step_dev):
         """One fwd+bwd; gradients land in the flat bucket.
-        Returns (loss, correct_count) device scalars."""
+        Returns (loss, mean_accuracy) device scalars."""
         ext = self.ext
         m = self.t.model
         sh, shT = m.shadows, m.shadows_T
@@ -103,8 +103,11 @@ class FusedLeNetStep:
                                     0, 0, shT["fc2_w"])
         # softmax computes the fc2 bias grad in the same pass (column sums
         # of dlogits) — no standalone mask_db launch for fc2
-        loss, correct, dl = ext.softmax_xent_fwd(logits, labels,
-                                                 db_out=gv("fc2_b"))
+        # inv_n folds the accuracy mean into the kernel: no scalar-divide
+        # launch per step; `acc` below is already the MEAN accuracy
+        loss, acc, dl = ext.softmax_xent_fwd(logits, labels,
+                                             db_out=gv("fc2_b"),
+                                             inv_n=1.0 / B)
 
         # ---- backward: dX chain on s0, dW GEMMs on s1 ----
         self._fork(s1, s0)
@@ -141,7 +144,7 @@ class FusedLeNetStep:
             self.t.engine.wire_allreduce(self.conv_slice, start=0)
         # keep the side-stream consumers alive until the join (capture-safe)
         self._keep = (a1, h2, dyeff1, dl, y1, dact2)
-        return loss, correct
+        return loss, acc
 
     # ------------------------------------------------------------------
     # Two-stage form of the SAME schedule, for the two-graph split capture
@@ -172,8 +175,11 @@ class FusedLeNetStep:
                                     shT["fc1_w"])
         logits = ext.linear_act_fwd(a1, sh["fc2_w"], m.fc2_b, False, 1.0,
                                     0, 0, shT["fc2_w"])
-        loss, correct, dl = ext.softmax_xent_fwd(logits, labels,
-                                                 db_out=gv("fc2_b"))
+        # inv_n folds the accuracy mean into the kernel: no scalar-divide
+        # launch per step; `acc` below is already the MEAN accuracy
+        loss, acc, dl = ext.softmax_xent_fwd(logits, labels,
+                                             db_out=gv("fc2_b"),
+                                             inv_n=1.0 / B)
         self._fork(s1, s0)
         with torch.cuda.stream(s1):
             ext.linear_dw_into(a1, dl, gv("fc2_w"))
@@ -189,7 +195,7 @@ class FusedLeNetStep:
         s0.wait_stream(s1)  # graph A boundary: fc grads complete
         self._stash = (x, y1, am1, dact2)
         self._keep = (a1, h2, dyeff1, dl)
-        return loss, correct
+        return loss, acc
 
     def stage_conv(self):
         ext = self.ext

@@ -142,7 +142,7 @@ class GraphedStep:
                 with torch.cuda.graph(self.graph):
                     # no grad-bucket fill: the previous replay's SGD tail
                     # zeroed it (zero_grad=True), and warmup left it zeroed
-                    loss, correct = self._fused.stage_fc(
+                    loss, acc = self._fused.stage_fc(
                         self.static_x, self.static_y, self.step_dev)
                     if self.dc_pre:
                         # fc grads final at graph A's boundary: mask them
@@ -154,8 +154,7 @@ class GraphedStep:
                                       base=self.fc_offset,
                                       step_dev=self.step_dev)
                     self.static_loss = loss.detach()
-                    acc = correct / self.static_x.shape[0]
-                    self.static_acc = acc.detach()
+                    self.static_acc = acc.detach()  # mean via inv_n
                 # graph B shares graph A's memory pool: it reads tensors
                 # graph A allocated (the stage_fc stash)
                 with torch.cuda.graph(self.graph_b, pool=self.graph.pool()):
@@ -222,9 +221,8 @@ class GraphedStep:
         # grad bucket already zero: the SGD tail clears it after consuming
         fused_reduced = False
         if self._fused is not None:
-            loss, correct = self._fused(self.static_x, self.static_y,
-                                        self.step_dev)
-            acc = correct / self.static_x.shape[0]
+            loss, acc = self._fused(self.static_x, self.static_y,
+                                    self.step_dev)
             fused_reduced = self._fused.overlap_allreduce
         else:
             logits = t.model(self.static_x, train=True)

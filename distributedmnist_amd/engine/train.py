@@ -250,8 +250,7 @@ class Trainer:
             # kernel path as the captured graph; the collective stays with
             # SyncEngine.reduce below (fused.overlap_allreduce disabled)
             self._step_dev.fill_(self.step)
-            loss, correct = fused(images, labels, self._step_dev)
-            acc = correct / images.shape[0]
+            loss, acc = fused(images, labels, self._step_dev)
         else:
             logits = self.model(images, train=True)
             loss, acc = self.model.loss_and_accuracy(logits, labels)
