@@ -523,6 +523,35 @@ def test_conv_dw_slab_image_groups(ext, force_g, monkeypatch):
         del os.environ["DMNIST_DW_G"]
 
 
+def test_graph_path_checkpoint_resume(ext, tmp_path):
+    """Checkpoint/resume THROUGH the captured-graph path: train N graphed
+    steps, save, restore into a fresh Trainer whose graph replays pick up
+    the restored flat_master (load_flat copies in place)."""
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+    td = str(tmp_path / "train")
+    argv = ["--synthetic_data", "--train_dir", td, "--batch_size", "256",
+            "--max_steps", "12", "--model", "lenet",
+            "--save_interval_secs", "0"]
+    flags = build_train_parser().parse_args(argv)
+    t = Trainer(flags, device=torch.device("cuda:0"))
+    t.train(make_dataset(flags, 0, 1, t.device, t.compute_dtype))
+    assert t._graph is not None, "graph path must engage"
+    w_end = t.fp.flat_master.detach().cpu().clone()
+    # resume run: restores at step 12, continues to 20 via graph replays
+    flags2 = build_train_parser().parse_args(
+        argv[:-2] + ["--save_interval_secs", "100000", "--max_steps", "20"])
+    t2 = Trainer(flags2, device=torch.device("cuda:0"))
+    hist = t2.train(make_dataset(flags2, 0, 1, t2.device, t2.compute_dtype))
+    assert t2._graph is not None
+    assert len(hist) == 8, len(hist)  # resumed at 12, ran 12..19
+    # the restore really loaded the previous run's weights
+    assert not torch.equal(t2.fp.flat_master.detach().cpu(), w_end) \
+        or len(hist) == 0
+    losses = [h[3] for h in hist]
+    assert all(np.isfinite(losses)), losses
+
+
 def test_train_loop_and_eval_gpu(ext, tmp_path):
     """Full engine on GPU: train() with checkpointing, then the evaluator
     entry point consumes the checkpoint (BASELINE configs 2 plumbing)."""

@@ -99,6 +99,22 @@ def test_all_modes_2gpu_rccl(mode_args, tmp_path):
 
 
 @pytest.mark.timeout(600)
+def test_hip_graph_full_capture_over_rccl(tmp_path):
+    """--hip_graph full captures the WHOLE step including the RCCL
+    all-reduce (ROUND2 candidate the 1-GPU boxes could never exercise);
+    30 steps at world=2 must complete with the capture engaged."""
+    out = _torchrun(
+        [os.path.join(ROOT, "src", "mnist_distributed_train.py"),
+         "--synthetic_data", "--batch_size", "256", "--max_steps", "30",
+         "--save_interval_secs", "100000", "--hip_graph", "full",
+         "--train_dir", str(tmp_path / "full")],
+        port=29456)
+    txt = out.stdout + out.stderr
+    assert re.search(r"step 29, loss = [0-9.]+", txt), txt[-2000:]
+    assert "running eager" not in txt, txt[-1500:]
+
+
+@pytest.mark.timeout(600)
 def test_param_parity_across_gpus_after_training(tmp_path):
     """Bitwise-policy check per SURVEY section 7.2 slice 2: after N sync
     steps over RCCL both ranks hold (near-)identical parameters — the
