@@ -559,25 +559,30 @@ void conv1_direct_fwd_kernel(const ushort_t* x, const ushort_t* w,
     // NOTE: an 8-chain even/odd-kh accumulator split was tried against the
     // 71% SQ_WAIT_INST_ANY reading and measured SLOWER (38/232 us vs
     // 34/206 at B=1024/8192) — the stall is not the acc RAW chain; the
-    // 4-chain nested-dot2 form below is the measured optimum
+    // 4-chain nested-dot2 form below is the measured optimum.
+    // The alignbit cross-pairs are HOISTED ahead of the dot2 chains
+    // (+12 VGPR at 28-register headroom) so no v_alignbit result is
+    // consumed the cycle after it issues.
+    unsigned oA[6][2];
+#pragma unroll
+    for (int r = 0; r < 6; ++r) {
+      oA[r][0] = __builtin_amdgcn_alignbit(wv[r][1], wv[r][0], 16);
+      oA[r][1] = __builtin_amdgcn_alignbit(wv[r][2], wv[r][1], 16);
+    }
     float acc0 = b, acc1 = b, acc2 = b, acc3 = b;
 #pragma unroll
     for (int kh = 0; kh < 5; ++kh) {
-      unsigned o0 = __builtin_amdgcn_alignbit(wv[kh][1], wv[kh][0], 16);
-      unsigned o1 = __builtin_amdgcn_alignbit(wv[kh][2], wv[kh][1], 16);
-      unsigned p0 = __builtin_amdgcn_alignbit(wv[kh + 1][1], wv[kh + 1][0], 16);
-      unsigned p1 = __builtin_amdgcn_alignbit(wv[kh + 1][2], wv[kh + 1][1], 16);
       acc0 = dot2bf(wv[kh][0], wpk[kh][0],
              dot2bf(wv[kh][1], wpk[kh][1],
              dot2bf(wv[kh][2], wt4e[kh], acc0)));
-      acc1 = dot2bf(o0, wpk[kh][0],
-             dot2bf(o1, wpk[kh][1],
+      acc1 = dot2bf(oA[kh][0], wpk[kh][0],
+             dot2bf(oA[kh][1], wpk[kh][1],
              dot2bf(wv[kh][2], wt4o[kh], acc1)));
       acc2 = dot2bf(wv[kh + 1][0], wpk[kh][0],
              dot2bf(wv[kh + 1][1], wpk[kh][1],
              dot2bf(wv[kh + 1][2], wt4e[kh], acc2)));
-      acc3 = dot2bf(p0, wpk[kh][0],
-             dot2bf(p1, wpk[kh][1],
+      acc3 = dot2bf(oA[kh + 1][0], wpk[kh][0],
+             dot2bf(oA[kh + 1][1], wpk[kh][1],
              dot2bf(wv[kh + 1][2], wt4o[kh], acc3)));
     }
     float vals[4] = {acc0, acc1, acc2, acc3};
