@@ -45,13 +45,20 @@ class FusedLeNetStep:
         # the hidden dW work dominates the edge cost (1.698 vs 1.733) —
         # single-stream below 4096, overlapped above.  DMNIST_TWO_STREAM /
         # DMNIST_SINGLE_STREAM force either for A/B.
+        # overlap level: 0 = single stream, 1 = fc dW only on the side
+        # stream (cheap contention, conv2 dW serial), 2 = every dW on the
+        # side stream (round-1 schedule).  DMNIST_OVERLAP=0/1/2 forces;
+        # legacy DMNIST_TWO_STREAM=1 -> 2, DMNIST_SINGLE_STREAM=1 -> 0.
         import os as _os
-        if _os.environ.get("DMNIST_TWO_STREAM"):
-            self.single_stream = False
+        if _os.environ.get("DMNIST_OVERLAP"):
+            self.overlap_level = int(_os.environ["DMNIST_OVERLAP"])
+        elif _os.environ.get("DMNIST_TWO_STREAM"):
+            self.overlap_level = 2
         elif _os.environ.get("DMNIST_SINGLE_STREAM"):
-            self.single_stream = True
+            self.overlap_level = 0
         else:
-            self.single_stream = t.flags.batch_size < 4096
+            self.overlap_level = 0 if t.flags.batch_size < 4096 else 2
+        self.single_stream = self.overlap_level == 0
         # side stream resolved PER CALL: in single-stream mode it must be
         # whatever stream the body is running on RIGHT NOW (warmup and
         # graph capture run on their own streams — freezing the init-time
@@ -130,8 +137,10 @@ class FusedLeNetStep:
         # liveness rides in the amax byte, no y2 re-read)
         dact2 = ext.linear_dx_unpool(dyeff1, sh["fc1_w"], am2,
                                      gv("conv2_b"), 7, 7, 64)
-        self._fork(s1, s0)
-        with torch.cuda.stream(s1):
+        sW = s1 if self.overlap_level >= 2 else s0  # conv2 dW placement
+        if sW is not s0:
+            self._fork(sW, s0)
+        with torch.cuda.stream(sW):
             ext.conv_dw_into(y1, dact2, gv("conv2_w"))
         dxc = ext.conv_dx(dact2, sh["conv2_w"], 32)
 
@@ -207,8 +216,10 @@ class FusedLeNetStep:
         x, y1, am1, dact2 = self._stash
         s0 = torch.cuda.current_stream()
         s1 = s0 if self.single_stream else self.side
-        self._fork(s1, s0)
-        with torch.cuda.stream(s1):
+        sW = s1 if self.overlap_level >= 2 else s0  # conv2 dW placement
+        if sW is not s0:
+            self._fork(sW, s0)
+        with torch.cuda.stream(sW):
             ext.conv_dw_into(y1, dact2, gv("conv2_w"))
         dxc = ext.conv_dx(dact2, m.shadows["conv2_w"], 32)
         ext.conv1_dw_pooled(x, dxc, am1, gv("conv1_w"), gv("conv1_b"))
