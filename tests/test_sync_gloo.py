@@ -171,6 +171,38 @@ def test_interval_slow_rank_does_not_stall_fast_rank():
     assert res[0][3] == res[1][3]        # same final aggregated grad
 
 
+def _run_interval_3rank(rank, world, port, q):
+    """3 ranks with DIFFERENT speeds + jitter: the generation agreement and
+    ring backpressure must hold for any relative timing."""
+    import random
+    import time
+    _init(rank, world, port)
+    rng = random.Random(1000 + rank)
+    g = torch.ones(512)
+    eng = SyncEngine(g, mode="interval", interval_ms=25.0,
+                     rank=rank, world_size=world)
+    for s in range(24):
+        eng.step_begin(s)
+        time.sleep(rng.uniform(0, 0.012) * (rank + 1))
+        eng.reduce(s, 0.001)
+    for _grad, _c in eng.finalize_interval():
+        pass
+    q.put((rank, eng.generation, eng._gen_posted,
+           float(eng.flat_grad.sum())))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_interval_3rank_jittered_agreement():
+    res = _spawn(_run_interval_3rank, world=3)
+    gens = {res[r][0] for r in range(3)}
+    posted = {res[r][1] for r in range(3)}
+    finals = {res[r][2] for r in range(3)}
+    assert len(gens) == 1 and len(posted) == 1, (gens, posted)
+    assert res[0][0] > 0, "no aggregation ever fired"
+    assert len(finals) == 1, finals  # same aggregated grad everywhere
+
+
 def _run_trainer_interval_e2e(rank, world, port, q):
     _init(rank, world, port)
     from distributedmnist_amd.engine.train import Trainer, make_dataset
