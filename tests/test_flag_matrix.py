@@ -1,0 +1,78 @@
+"""Flag-interaction matrix: every DP mode crossed with drop-connect
+placement and the bf16 gradient wire, world_size=2 over gloo.  Each cell
+trains 6 steps end-to-end through Trainer.train() and checks rank-0/1
+parameter agreement for the synchronous modes — interaction bugs (e.g. a
+mode that skips the mask, a wire buffer aliasing a mode's flag element)
+don't show up in the single-feature tests."""
+
+import os
+import sys
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from test_sync_gloo import free_port  # noqa: E402
+
+CASES = [
+    # (extra argv, params_must_match)
+    ([], True),
+    (["--drop_connect"], True),                        # pre-agg masks
+    (["--drop_connect", "--drop_connect_post"], True),
+    (["--grad_dtype", "bf16"], True),
+    (["--drop_connect", "--grad_dtype", "bf16"], True),
+    (["--num_replicas_to_aggregate", "1"], True),
+    (["--num_replicas_to_aggregate", "1", "--grad_dtype", "bf16"], True),
+    (["--worker_times_cdf_method"], True),
+    (["--worker_times_cdf_method", "--drop_connect"], True),
+    (["--interval_method", "--interval_ms", "0"], True),
+    (["--interval_method", "--interval_ms", "20", "--drop_connect"], True),
+    (["--straggler_timeout_ms", "10000"], True),
+    (["--momentum", "0.9"], True),
+    (["--momentum", "0.9", "--drop_connect", "--grad_dtype", "bf16"], True),
+]
+
+
+def _run(rank, world, port, extra, q):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+    flags = build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", f"/tmp/dmnist_matrix_{port}",
+         "--batch_size", "16", "--max_steps", "6", "--model", "mlp",
+         "--device", "cpu", "--save_interval_secs", "100000"] + extra)
+    t = Trainer(flags, rank=rank, world=world, local_rank=rank)
+    ds = make_dataset(flags, rank, world, t.device, t.compute_dtype)
+    t.train(ds)
+    q.put((rank, float(t.fp.flat_master.sum()),
+           t.fp.flat_master[:4].tolist()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+@pytest.mark.parametrize("extra,must_match",
+                         CASES, ids=[" ".join(c[0]) or "default"
+                                     for c in CASES])
+def test_flag_matrix_world2(extra, must_match):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = free_port()
+    ps = [ctx.Process(target=_run, args=(r, 2, port, extra, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    res = {}
+    for _ in range(2):
+        r, s, head = q.get()
+        res[r] = (s, tuple(head))
+    for p in ps:
+        p.join(120)
+        assert p.exitcode == 0, (extra, p.exitcode)
+    if must_match:
+        assert res[0] == res[1], (extra, res)
