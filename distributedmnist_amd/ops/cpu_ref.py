@@ -114,8 +114,9 @@ def dropout_fwd(x, p_drop: float, gen: torch.Generator | None = None):
     """TF-style dropout (mnist.py:139-140): keep with prob 1-p, scale by
     1/(1-p). Returns (y, mask bool)."""
     keep = 1.0 - p_drop
-    mask = torch.rand(x.shape, device=x.device, dtype=torch.float32,
-                      generator=gen) < keep
+    # generator lives on CPU; sample there and move (device-agnostic ref)
+    mask = (torch.rand(x.shape, device="cpu", dtype=torch.float32,
+                       generator=gen) < keep).to(x.device)
     y = x * mask.to(x.dtype) / keep
     return y.to(x.dtype), mask
 

@@ -21,7 +21,10 @@ from . import cpu_ref
 
 
 def _use_hip(t: torch.Tensor) -> bool:
-    return t.is_cuda
+    # the HIP kernels are the bf16 compute path (bf16 operands, fp32
+    # epilogues); fp32-on-GPU routes to the torch reference ops (MIOpen /
+    # rocBLAS) — a numerics-debug mode, never the bench path
+    return t.is_cuda and t.dtype == torch.bfloat16
 
 
 class ConvPoolFn(torch.autograd.Function):
@@ -97,7 +100,7 @@ class LinearActFn(torch.autograd.Function):
                 gen = torch.Generator(device="cpu")
                 gen.manual_seed((seed * 0x9E3779B97F4A7C15 + offset) % (2**63))
                 mask = (torch.rand(y.shape, generator=gen) < p_keep)
-                y = y * mask.to(y.dtype) / p_keep
+                y = y * mask.to(device=y.device, dtype=y.dtype) / p_keep
         ctx.save_for_backward(x, w_comp, y)
         ctx.relu = relu
         ctx.p_keep = p_keep
@@ -186,7 +189,7 @@ def grad_mask(grad: torch.Tensor, keep: float, seed: int, step: int,
         gen.manual_seed(((seed ^ 0x9D5AD0C5) * 0x9E3779B97F4A7C15
                          + (step << 20) + rank + 1) % (2**63))
         mask = (torch.rand(grad.shape, generator=gen) < keep).float()
-        grad.mul_(mask)
+        grad.mul_(mask.to(grad.device))
 
 
 def sgd_step(master: torch.Tensor, grad: torch.Tensor, lr: float,

@@ -523,6 +523,25 @@ def test_conv_dw_slab_image_groups(ext, force_g, monkeypatch):
         del os.environ["DMNIST_DW_G"]
 
 
+def test_fp32_on_gpu_debug_mode(ext, tmp_path):
+    """--compute_dtype fp32 on GPU is the numerics-debug mode: no bf16
+    shadows, so ops route to the torch reference path (MIOpen/rocBLAS) —
+    it must TRAIN, not crash on the bf16-only HIP bindings."""
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+    flags = build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", str(tmp_path / "t"),
+         "--batch_size", "64", "--max_steps", "4", "--model", "lenet",
+         "--compute_dtype", "fp32", "--save_interval_secs", "100000"])
+    t = Trainer(flags, device=torch.device("cuda:0"))
+    assert t.compute_dtype == torch.float32
+    ds = make_dataset(flags, 0, 1, t.device, t.compute_dtype)
+    for _ in range(4):
+        x, y = ds.next_batch(64)
+        _, loss, acc, _ = t.train_step(x, y)
+    assert np.isfinite(float(loss))
+
+
 @pytest.mark.parametrize("B", [1, 7, 100, 513, 1000])
 def test_odd_batch_sizes_gpu(ext, B):
     """Non-aligned batch sizes through the whole fused forward+backward:
