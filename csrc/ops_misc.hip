@@ -154,11 +154,24 @@ void softmax_xent_t_kernel(const ushort_t* logits, const long* labels,
     float v[CC ? CC : 16];
     float mx = -1e30f;
     int arg = 0;
+    if (CC == 10 && live) {
+      // one 16B + one 4B load per row (20B, always 4B-aligned) instead of
+      // 10 scalar u16 loads
+      ushort_t row[10];
+      __builtin_memcpy(&row[0], logits + (size_t)b * 10, 16);
+      __builtin_memcpy(&row[8], logits + (size_t)b * 10 + 8, 4);
 #pragma unroll
-    for (int c = 0; c < (CC ? CC : 16); ++c) {
-      if (!CC && c >= C) break;
-      v[c] = live ? bf2f(logits[(size_t)b * C + c]) : 0.f;
-      if (v[c] > mx) { mx = v[c]; arg = c; }
+      for (int c = 0; c < 10; ++c) {
+        v[c] = bf2f(row[c]);
+        if (v[c] > mx) { mx = v[c]; arg = c; }
+      }
+    } else {
+#pragma unroll
+      for (int c = 0; c < (CC ? CC : 16); ++c) {
+        if (!CC && c >= C) break;
+        v[c] = live ? bf2f(logits[(size_t)b * C + c]) : 0.f;
+        if (v[c] > mx) { mx = v[c]; arg = c; }
+      }
     }
     float se = 0.f;
 #pragma unroll
@@ -170,12 +183,26 @@ void softmax_xent_t_kernel(const ushort_t* logits, const long* labels,
     float inv_se = 1.f / se;
     long lab = live ? labels[b] : 0;
     float invB = 1.f / (float)B;
+    if (CC == 10) {
+      ushort_t rowo[10];
 #pragma unroll
-    for (int c = 0; c < (CC ? CC : 16); ++c) {
-      if (!CC && c >= C) break;
-      float pp = v[c] * inv_se;
-      d[c] = live ? (pp - (c == (int)lab ? 1.f : 0.f)) * invB : 0.f;
-      if (live) dlogits[(size_t)b * C + c] = f2bf(d[c]);
+      for (int c = 0; c < 10; ++c) {
+        float pp = v[c] * inv_se;
+        d[c] = live ? (pp - (c == (int)lab ? 1.f : 0.f)) * invB : 0.f;
+        rowo[c] = f2bf(d[c]);
+      }
+      if (live) {
+        __builtin_memcpy(dlogits + (size_t)b * 10, &rowo[0], 16);
+        __builtin_memcpy(dlogits + (size_t)b * 10 + 8, &rowo[8], 4);
+      }
+    } else {
+#pragma unroll
+      for (int c = 0; c < (CC ? CC : 16); ++c) {
+        if (!CC && c >= C) break;
+        float pp = v[c] * inv_se;
+        d[c] = live ? (pp - (c == (int)lab ? 1.f : 0.f)) * invB : 0.f;
+        if (live) dlogits[(size_t)b * C + c] = f2bf(d[c]);
+      }
     }
     if (live) {
       loss = -(__logf(v[(int)lab] * inv_se)) * invB;
