@@ -34,6 +34,38 @@ def test_bench_single_process():
     assert j["data"] == "synthetic"
 
 
+def test_bench_json_schema_complete():
+    """Every field the driver contract names must be present with the
+    right type/value domain (a missing or mistyped field invalidates the
+    round's BENCH record)."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"), "--steps", "3",
+         "--warmup", "1", "--batch_size", "32"],
+        capture_output=True, text=True, timeout=240, cwd=ROOT)
+    assert out.returncode == 0, out.stderr[-2000:]
+    j = _last_json_line(out.stdout)
+    assert isinstance(j["metric"], str) and j["metric"]
+    assert isinstance(j["value"], (int, float)) and j["value"] > 0
+    assert j["unit"] == "images/sec"
+    assert j["n_gpus"] == 1 and j["steps"] == 3 and j["warmup"] == 1
+    assert isinstance(j["ms_per_step"], (int, float)) and j["ms_per_step"] > 0
+    assert j["higher_is_better"] is True
+    assert j["scaling"] == "weak"
+    assert j["vs_baseline"] is None  # reference publishes no numbers
+    assert j["dtype"] in ("bf16", "fp32")
+    assert j["data"] == "synthetic"
+    cfg = j["config"]
+    assert cfg["model"] == "lenet"
+    assert cfg["global_batch"] == 32
+    assert cfg["seq_len"] is None
+    assert cfg["parallelism"] == "dp1"
+    for k in ("p50_ms_per_step", "p95_ms_per_step", "p99_ms_per_step"):
+        assert isinstance(cfg[k], (int, float)) and cfg[k] > 0
+    # whole-job aggregate identity: value = global_batch * steps / elapsed
+    approx = cfg["global_batch"] / (j["ms_per_step"] / 1000.0)
+    assert abs(j["value"] - approx) < 0.01 * approx
+
+
 @pytest.mark.timeout(300)
 def test_bench_mode_is_real_cdf():
     """VERDICT round-1 task 1: --mode must map onto the trainer flags, not

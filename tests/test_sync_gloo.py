@@ -307,3 +307,26 @@ def _run_trainer_e2e(rank, world, port, q):
 def test_trainer_e2e_params_identical_across_ranks():
     res = _spawn(_run_trainer_e2e)
     assert res[0] == res[1], "ranks diverged after 5 sync steps"
+
+
+def _run_k_of_n_tie(rank, world, port, q):
+    _init(rank, world, port)
+    g = torch.full((10,), float(rank + 1))
+    eng = SyncEngine(g, mode="k_of_n", replicas_to_aggregate=1,
+                     rank=rank, world_size=world)
+    eng.step_begin(0)
+    # EQUAL compute times: the (time, rank) sort key must break the tie
+    # identically on every rank (rank 0 wins) or the contribute decisions
+    # diverge and the renormalization count is wrong
+    applied, grad, contributors = eng.reduce(0, compute_time_s=0.5)
+    q.put((rank, applied, grad[0].item(), contributors))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_k_of_n_tie_break_deterministic():
+    res = _spawn(_run_k_of_n_tie)
+    for r in (0, 1):
+        applied, gval, contributors = res[r]
+        assert applied and contributors == 1
+        assert gval == 1.0  # rank 0 wins the tie on both ranks
