@@ -112,3 +112,45 @@ def test_set_step_clears_device_counter():
     assert m._step_dev is sentinel
     m.set_step(7)
     assert m._step_dev is None and m._step == 7
+
+
+def test_stale_index_falls_back_to_glob(tmp_path):
+    """The `checkpoint` index can reference a file the GC already removed
+    (reference get_checkpoint_state semantics): latest_checkpoint must fall
+    back to the newest on-disk model.ckpt-* instead of returning nothing."""
+    import torch
+
+    from distributedmnist_amd.engine.supervisor import Supervisor
+    td = tmp_path / "td"
+    td.mkdir()
+    torch.save({"flat_master": torch.zeros(3), "step": 7}, td / "model.ckpt-7")
+    (td / "checkpoint").write_text(
+        'model_checkpoint_path: "model.ckpt-99"\n')  # points at a ghost
+    latest = Supervisor.latest_checkpoint(str(td))
+    assert latest is not None
+    step, path = latest
+    assert step == 7 and path.endswith("model.ckpt-7")
+    step2, payload = Supervisor.restore(str(td))
+    assert step2 == 7 and payload["step"] == 7
+
+
+def test_restore_rejects_pickled_code(tmp_path):
+    """weights_only=True: a checkpoint carrying arbitrary pickled objects
+    must be REJECTED, not executed (round-1 ADVICE: torch.load RCE)."""
+    import pickle
+
+    import pytest as _pytest
+
+    from distributedmnist_amd.engine.supervisor import Supervisor
+    td = tmp_path / "td"
+    td.mkdir()
+
+    class Evil:
+        def __reduce__(self):
+            return (print, ("pwned",))
+
+    with open(td / "model.ckpt-3", "wb") as f:
+        pickle.dump({"flat_master": Evil()}, f)
+    (td / "checkpoint").write_text('model_checkpoint_path: "model.ckpt-3"\n')
+    with _pytest.raises(Exception):
+        Supervisor.restore(str(td))
