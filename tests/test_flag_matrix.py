@@ -76,3 +76,53 @@ def test_flag_matrix_world2(extra, must_match):
         assert p.exitcode == 0, (extra, p.exitcode)
     if must_match:
         assert res[0] == res[1], (extra, res)
+
+
+def _run_wire_traj(rank, world, port, grad_dtype, q):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from distributedmnist_amd.engine.train import Trainer, make_dataset
+    from distributedmnist_amd.utils.flags import build_train_parser
+    flags = build_train_parser().parse_args(
+        ["--synthetic_data", "--train_dir", f"/tmp/dmnist_wt_{port}",
+         "--batch_size", "16", "--max_steps", "20", "--model", "mlp",
+         "--device", "cpu", "--save_interval_secs", "100000",
+         "--grad_dtype", grad_dtype])
+    t = Trainer(flags, rank=rank, world=world, local_rank=rank)
+    ds = make_dataset(flags, rank, world, t.device, t.compute_dtype)
+    t.train(ds)
+    q.put((rank, t.fp.flat_master.clone()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_bf16_wire_trajectory_tracks_fp32():
+    """20-step 2-rank trajectories with fp32 vs bf16 gradient wire must
+    stay close (bf16 rounds each reduce; fp32 master updates otherwise
+    identical) — the numerics guard for --grad_dtype bf16."""
+    import torch as _torch
+
+    def run(dtype):
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        port = free_port()
+        ps = [ctx.Process(target=_run_wire_traj, args=(r, 2, port, dtype, q))
+              for r in range(2)]
+        for p in ps:
+            p.start()
+        res = {}
+        for _ in range(2):
+            r, w = q.get()
+            res[r] = w
+        for p in ps:
+            p.join(120)
+            assert p.exitcode == 0
+        assert _torch.equal(res[0], res[1])
+        return res[0]
+
+    w32 = run("fp32")
+    w16 = run("bf16")
+    rel = float((w32 - w16).norm() / w32.norm())
+    assert rel < 0.01, f"bf16-wire trajectory diverged: rel {rel}"
