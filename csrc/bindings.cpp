@@ -530,6 +530,15 @@ void conv_dw_into(torch::Tensor x, torch::Tensor dact, torch::Tensor dw_out) {
     int want = p.K <= 262144 ? 1024 : 2048;
     if (const char* e = getenv("DMNIST_DW_BLOCKS")) want = atoi(e);
     p.splitk = std::min(cdiv(p.K, 64), std::max(1, want / tiles));
+    if (getenv("DMNIST_DW_SPREAD")) {
+      // contention probe: atomics land in a throwaway 16-plane scratch
+      static torch::Tensor scratch;
+      int64_t need = 16LL * p.M * p.N;
+      if (!scratch.defined() || scratch.numel() < need)
+        scratch = torch::zeros({need}, dact.options().dtype(at::kFloat));
+      p.C = scratch.data_ptr();
+      p.offset = 1;
+    }
     if (dw_tr_enabled() && Cin % 8 == 0 && Cout % 64 == 0)
       conv_dw_tr(p, s);
     else

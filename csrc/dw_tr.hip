@@ -270,6 +270,11 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
 
   const int frow = (lane >> 4) * 4;
   const int fcol = lane & 15;
+  float* outp = reinterpret_cast<float*>(p.C);
+  // p.offset != 0: DIAGNOSTIC contention probe — spread each k-slice's
+  // atomics over 16 separate planes (outputs meaningless; isolates the
+  // same-address atomic serialization cost in the flush)
+  if (p.offset) outp += (size_t)(bz & 15) * p.M * p.ldc;
 #pragma unroll
   for (int mi = 0; mi < MI; ++mi) {
 #pragma unroll
@@ -280,8 +285,7 @@ __global__ __launch_bounds__(NT) void dw_tr_kernel(GemmParams p) {
       for (int r = 0; r < 4; ++r) {
         int gr = m0 + wr * WM + mi * 16 + frow + r;
         if (gr >= p.M) continue;
-        atomicAdd(reinterpret_cast<float*>(p.C) + (size_t)gr * p.ldc + gc,
-                  acc[mi][ni][r]);
+        atomicAdd(outp + (size_t)gr * p.ldc + gc, acc[mi][ni][r]);
       }
     }
   }
