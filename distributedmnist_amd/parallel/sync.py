@@ -236,9 +236,25 @@ class SyncEngine:
             self._next_fire += self.interval_s
         return self._poll_apply(step)
 
-    def _post_generation(self, step: int):
+    def _post_generation(self, step: int, block: bool = True) -> bool:
+        """Post one aggregation generation.  block=False refuses instead of
+        waiting when the ring is exhausted — REQUIRED during the shutdown
+        phase-1 schedule servicing: a rank ahead of the schedule that
+        hard-blocked there could wait on a generation its (already
+        barrier-complete, no-longer-firing) peer will never post, while
+        that peer blocks in the phase-2 MAX — mutual deadlock (found by the
+        flag-matrix stress test).  In the training loop and in phase 2
+        blocking is safe: every other rank is still servicing the schedule
+        (training/phase-1) or known to be posting its owed generations."""
         n = self.flat_grad.numel()
         if not self._free:
+            # completed-but-unapplied work still occupies slots (nothing
+            # calls _poll_apply during shutdown phase 1): recycle it into
+            # the deferred-apply queue before deciding to refuse/block
+            self._recycle_completed()
+        if not self._free:
+            if not block:
+                return False
             # backpressure: ring exhausted (this rank is > RING intervals
             # ahead of the slowest) — block on the oldest in-flight, move
             # its result to the deferred-apply queue, recycle its slot
@@ -261,6 +277,25 @@ class SyncEngine:
         self._gen_posted += 1
         self._accum.zero_()
         self._accum_count = 0
+        return True
+
+    def _recycle_completed(self):
+        """Move every COMPLETED in-flight aggregation (oldest-first) into
+        the deferred-apply queue, freeing its ring slot.  Keeps the ring
+        live for ranks that are posting without applying (the shutdown
+        phase-1 schedule servicing)."""
+        n = self.flat_grad.numel()
+        while self._pending:
+            gen, work, idx = self._pending[0]
+            if work is not None and not work.is_completed():
+                break
+            if work is not None:
+                work.wait()  # ordering/stream registration; already done
+            self._pending.pop(0)
+            buf_ = self._ring[idx]
+            contributors = max(1, int(round(float(buf_[n].item()))))
+            self._deferred.append((buf_[:n].clone(), contributors))
+            self._free.append(idx)
 
     def _poll_apply(self, step: int, force: bool = False):
         """Apply the oldest completed aggregation, if any (at most one per
@@ -316,7 +351,11 @@ class SyncEngine:
                 now = time.time()
                 while (self._next_fire is not None and self.interval_s > 0
                        and now >= self._next_fire):
-                    self._post_generation(step=-1)
+                    # NON-blocking post: a full ring here must yield back
+                    # to the barrier poll, never hard-block (deadlock
+                    # against a peer already waiting in the phase-2 MAX)
+                    if not self._post_generation(step=-1, block=False):
+                        break
                     self._next_fire += self.interval_s
                 time.sleep(0.002)
             g = torch.tensor([self._gen_posted], dtype=torch.int64)

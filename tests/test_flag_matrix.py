@@ -15,7 +15,7 @@ import torch.multiprocessing as mp
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-from test_sync_gloo import free_port  # noqa: E402
+from mp_utils import free_port, init_pg, spawn_collect  # noqa: E402,F401
 
 CASES = [
     # (extra argv, params_must_match)
@@ -37,10 +37,7 @@ CASES = [
 
 
 def _run(rank, world, port, extra, q):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
-                      RANK=str(rank), WORLD_SIZE=str(world),
-                      LOCAL_RANK=str(rank))
-    dist.init_process_group("gloo", rank=rank, world_size=world)
+    init_pg(rank, world, port)
     from distributedmnist_amd.engine.train import Trainer, make_dataset
     from distributedmnist_amd.utils.flags import build_train_parser
     flags = build_train_parser().parse_args(
@@ -60,29 +57,14 @@ def _run(rank, world, port, extra, q):
                          CASES, ids=[" ".join(c[0]) or "default"
                                      for c in CASES])
 def test_flag_matrix_world2(extra, must_match):
-    ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
-    port = free_port()
-    ps = [ctx.Process(target=_run, args=(r, 2, port, extra, q))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    res = {}
-    for _ in range(2):
-        r, s, head = q.get()
-        res[r] = (s, tuple(head))
-    for p in ps:
-        p.join(120)
-        assert p.exitcode == 0, (extra, p.exitcode)
+    res = spawn_collect(_run, 2, args=(extra,))
+    res = {r: (v[0], tuple(v[1])) for r, v in res.items()}
     if must_match:
         assert res[0] == res[1], (extra, res)
 
 
 def _run_wire_traj(rank, world, port, grad_dtype, q):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
-                      RANK=str(rank), WORLD_SIZE=str(world),
-                      LOCAL_RANK=str(rank))
-    dist.init_process_group("gloo", rank=rank, world_size=world)
+    init_pg(rank, world, port)
     from distributedmnist_amd.engine.train import Trainer, make_dataset
     from distributedmnist_amd.utils.flags import build_train_parser
     flags = build_train_parser().parse_args(
@@ -105,22 +87,9 @@ def test_bf16_wire_trajectory_tracks_fp32():
     import torch as _torch
 
     def run(dtype):
-        ctx = mp.get_context("spawn")
-        q = ctx.SimpleQueue()
-        port = free_port()
-        ps = [ctx.Process(target=_run_wire_traj, args=(r, 2, port, dtype, q))
-              for r in range(2)]
-        for p in ps:
-            p.start()
-        res = {}
-        for _ in range(2):
-            r, w = q.get()
-            res[r] = w
-        for p in ps:
-            p.join(120)
-            assert p.exitcode == 0
-        assert _torch.equal(res[0], res[1])
-        return res[0]
+        res = spawn_collect(_run_wire_traj, 2, args=(dtype,))
+        assert _torch.equal(res[0][0], res[1][0])
+        return res[0][0]
 
     w32 = run("fp32")
     w16 = run("bf16")

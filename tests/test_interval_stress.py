@@ -18,13 +18,11 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from distributedmnist_amd.parallel.sync import SyncEngine  # noqa: E402
 
-from test_sync_gloo import free_port  # noqa: E402
+from mp_utils import init_pg, spawn_collect  # noqa: E402
 
 
 def _run(rank, world, port, interval_ms, steps, seed, q):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
-                      RANK=str(rank), WORLD_SIZE=str(world))
-    dist.init_process_group("gloo", rank=rank, world_size=world)
+    init_pg(rank, world, port)
     rng = random.Random(seed * 100 + rank)
     g = torch.ones(256)
     eng = SyncEngine(g, mode="interval", interval_ms=interval_ms,
@@ -44,21 +42,8 @@ def _run(rank, world, port, interval_ms, steps, seed, q):
 
 
 def _trial(world, interval_ms, steps, seed):
-    ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
-    port = free_port()
-    ps = [ctx.Process(target=_run,
-                      args=(r, world, port, interval_ms, steps, seed, q))
-          for r in range(world)]
-    for p in ps:
-        p.start()
-    res = {}
-    for _ in range(world):
-        r, gen, posted, fs = q.get()
-        res[r] = (gen, posted, fs)
-    for p in ps:
-        p.join(120)
-        assert p.exitcode == 0, p.exitcode
+    res = spawn_collect(_run, world, args=(interval_ms, steps, seed))
+    res = {r: tuple(v) for r, v in res.items()}
     assert len(set(res.values())) == 1, res
     return res[0]
 

@@ -2,33 +2,21 @@
 distributed path is made correct-by-construction here on CPU; the same code
 runs over RCCL on the 8-GPU node (SURVEY.md section 4 item c)."""
 
-import os
-import socket
+import os  # noqa: F401  (children use env)
 
 import numpy as np
 import pytest
 import torch
 import torch.distributed as dist
-import torch.multiprocessing as mp
+import torch.multiprocessing as mp  # noqa: F401
 
 from distributedmnist_amd.parallel.sync import SyncEngine
 
-
-def free_port():
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    p = s.getsockname()[1]
-    s.close()
-    return p
+from mp_utils import free_port, init_pg, spawn_collect  # noqa: F401
 
 
 def _init(rank, world, port):
-    os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(port)
-    os.environ["RANK"] = str(rank)
-    os.environ["WORLD_SIZE"] = str(world)
-    os.environ["LOCAL_RANK"] = str(rank)
-    dist.init_process_group("gloo", rank=rank, world_size=world)
+    init_pg(rank, world, port)
 
 
 def _run_full_sync(rank, world, port, q):
@@ -79,21 +67,7 @@ def _run_timeout(rank, world, port, q):
 
 
 def _spawn(fn, world=2):
-    ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
-    port = free_port()
-    procs = [ctx.Process(target=fn, args=(r, world, port, q))
-             for r in range(world)]
-    for p in procs:
-        p.start()
-    results = {}
-    for _ in range(world):
-        r, *vals = q.get()
-        results[r] = vals
-    for p in procs:
-        p.join(60)
-        assert p.exitcode == 0
-    return results
+    return spawn_collect(fn, world)
 
 
 @pytest.mark.timeout(120)
