@@ -57,3 +57,38 @@ def _trial(world, interval_ms, steps, seed):
 def test_interval_stress(world, interval_ms, steps):
     gen, posted, _ = _trial(world, interval_ms, steps, seed=world)
     assert gen == posted > 0
+
+
+def _run_staggered(rank, world, port, q):
+    # recreate the shutdown-deadlock geometry: rank 1 starts LATE (spawn
+    # stagger), rank 0 finishes its few steps immediately and sits in
+    # finalize phase 1 firing every 10 ms until its ring fills
+    time.sleep(rank * 2.0)
+    init_pg(rank, world, port)
+    g = torch.ones(128)
+    eng = SyncEngine(g, mode="interval", interval_ms=10.0,
+                     rank=rank, world_size=world)
+    for s in range(4):
+        eng.step_begin(s)
+        time.sleep(0.01)
+        eng.reduce(s, 0.001)
+    for _ in eng.finalize_interval():
+        pass
+    q.put((rank, eng.generation, eng._gen_posted))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_interval_shutdown_staggered_no_deadlock():
+    """Regression: a rank ahead of the fire schedule used to hard-block in
+    ring backpressure during finalize phase 1 (waiting on generations its
+    stopped peer would never post) while the peer blocked in the phase-2
+    MAX; a second shape had the phase-1 rank yielding forever because
+    completed ring slots were never recycled.  Both fixed in sync.py."""
+    res = _trial_raw(_run_staggered, 2)
+    assert res[0] == res[1]
+
+
+def _trial_raw(fn, world):
+    res = spawn_collect(fn, world, collect_s=200)
+    return {r: tuple(v) for r, v in res.items()}
