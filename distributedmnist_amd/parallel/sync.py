@@ -368,6 +368,10 @@ class SyncEngine:
             if not applied:
                 break
             yield grad, contributors
+        # a later training run on this engine renegotiates its own fire
+        # origin (a stale origin would owe one catch-up per elapsed
+        # interval since THIS run ended)
+        self._next_fire = None
 
     # ------------------------------------------------------------------
     def _maybe_log_cdf(self, step: int):
