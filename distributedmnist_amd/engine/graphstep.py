@@ -43,8 +43,10 @@ class GraphedStep:
         t = trainer
         self.split = split
         assert t.device.type == "cuda", "GraphedStep requires a GPU"
-        assert t.mode == "full_sync" and t.engine.timeout_s is None, \
-            "only the full_sync hot path is graph-captured"
+        assert t.engine.timeout_s is None and (
+            t.mode == "full_sync" or (t.mode == "cdf" and t.world == 1)), \
+            "graph-captured: full_sync, or cdf at world=1 (identical apply " \
+            "semantics — contributors == world, no drop path)"
         self.t = t
         ext = _C.ext()
         dev = t.device

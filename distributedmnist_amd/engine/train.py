@@ -150,7 +150,13 @@ class Trainer:
             return None
         self._graph_tried = True
         flags = self.flags
-        eligible = (self.device.type == "cuda" and self.mode == "full_sync"
+        # cdf at world=1 is graphable too: no collective can pollute the
+        # per-worker timing, so hipEvents around the replay measure exactly
+        # what the eager path measures (at world>1 cdf stays eager — the
+        # straggler CDF must be COMPUTE time, not compute+comm)
+        graphable_mode = (self.mode == "full_sync"
+                          or (self.mode == "cdf" and self.world == 1))
+        eligible = (self.device.type == "cuda" and graphable_mode
                     and self.engine.timeout_s is None
                     and flags.inject_slow_rank < 0
                     and getattr(flags, "hip_graph", "auto") != "off")
@@ -197,6 +203,14 @@ class Trainer:
         """Preferred step entry: graph replay when available."""
         g = self.get_graph(images)
         if g is not None:
+            if self.mode == "cdf":
+                step = self.step
+                self.engine.step_begin(step)
+                self.timer.start()
+                loss, acc = g.run(images, labels)
+                ct = self.timer.stop()
+                self.engine.record_cdf(step, ct)
+                return True, loss, acc, ct
             loss, acc = g.run(images, labels)
             return True, loss, acc, 0.0
         return self.train_step(images, labels)

@@ -389,6 +389,15 @@ class SyncEngine:
         log.info("ELAPSED TIMES %s", str(elapsed))
         log.info("ITERATION TIMES %s", str(iter_times))
 
+    def record_cdf(self, step: int, compute_time_s: float):
+        """CDF bookkeeping for steps whose compute ran OUTSIDE reduce()
+        (the graph-replayed cdf path at world=1): gather per-rank times,
+        append, emit the scraper report — mirrors reduce()'s cdf block."""
+        times = self._all_gather_times(compute_time_s, step)
+        for w, tm in enumerate(times):
+            self._compute_times.append((tm, w, step))
+        self._maybe_log_cdf(step)
+
     def compute_time_percentiles(self):
         """Percentile stats over collected per-rank compute times
         (benchmark.py:97-111 shape)."""
