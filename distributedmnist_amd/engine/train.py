@@ -206,14 +206,39 @@ class Trainer:
             if self.mode == "cdf":
                 step = self.step
                 self.engine.step_begin(step)
-                self.timer.start()
+                e0, e1 = self._cdf_event_pair()
+                e0.record()
                 loss, acc = g.run(images, labels)
-                ct = self.timer.stop()
-                self.engine.record_cdf(step, ct)
-                return True, loss, acc, ct
+                e1.record()
+                self._cdf_steps.append(step)
+                if len(self._cdf_steps) >= self.engine.cdf_log_every:
+                    self._flush_cdf()
+                return True, loss, acc, 0.0
             loss, acc = g.run(images, labels)
             return True, loss, acc, 0.0
         return self.train_step(images, labels)
+
+    # -- batched CDF timing for the graph path -------------------------
+    def _cdf_event_pair(self):
+        if not hasattr(self, "_cdf_events"):
+            n = self.engine.cdf_log_every
+            self._cdf_events = [(torch.cuda.Event(enable_timing=True),
+                                 torch.cuda.Event(enable_timing=True))
+                                for _ in range(n)]
+            self._cdf_steps = []
+        return self._cdf_events[len(self._cdf_steps)]
+
+    def _flush_cdf(self):
+        """One synchronize for the whole window, then one gather+report
+        (identical content to the per-step form; ~70 us/step cheaper)."""
+        if not getattr(self, "_cdf_steps", None):
+            return
+        n = len(self._cdf_steps)
+        self._cdf_events[n - 1][1].synchronize()
+        times = [self._cdf_events[i][0].elapsed_time(
+                     self._cdf_events[i][1]) / 1000.0 for i in range(n)]
+        self.engine.record_cdf_batch(self._cdf_steps, times)
+        self._cdf_steps = []
 
     def _get_eager_fused(self):
         """FusedLeNetStep for the eager path (non-graphable modes), or None
@@ -379,6 +404,8 @@ class Trainer:
             # ranks leave with identical parameters
             for grad, contributors in self.engine.finalize_interval():
                 self._apply_update(grad, contributors)
+        if self.mode == "cdf":
+            self._flush_cdf()  # tail of the last (partial) timing window
         if writer is not None:
             writer.close()
         if self.is_chief:

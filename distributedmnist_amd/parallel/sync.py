@@ -374,12 +374,14 @@ class SyncEngine:
         self._next_fire = None
 
     # ------------------------------------------------------------------
-    def _maybe_log_cdf(self, step: int):
+    def _maybe_log_cdf(self, step: int, force: bool = False):
         """Scraper-compatible CDF report (timeout_manager.py:63-70 format,
-        consumed by benchmark.py extract_compute_times/iteration_times)."""
+        consumed by benchmark.py extract_compute_times/iteration_times).
+        force=True skips the every-N-steps gate (the batched path flushes
+        exactly once per window, at whatever step ends it)."""
         if self.rank != 0 or step <= self.cdf_start_tracking:
             return
-        if step % self.cdf_log_every != 0:
+        if not force and step % self.cdf_log_every != 0:
             return
         elapsed = sorted((t, w, i) for (t, w, i) in self._compute_times
                          if i > self.cdf_start_tracking)
@@ -397,6 +399,28 @@ class SyncEngine:
         for w, tm in enumerate(times):
             self._compute_times.append((tm, w, step))
         self._maybe_log_cdf(step)
+
+    def record_cdf_batch(self, steps, my_times):
+        """Batched form of record_cdf: one gather for a WINDOW of steps
+        (the graph path times every replay with hipEvent pairs but only
+        synchronizes once per report window — per-step event.synchronize
+        was the cdf mode's bottleneck, ~70 us/step of host round-trips).
+        Report content is identical to the per-step form."""
+        if not steps:
+            return
+        if self.distributed:
+            t = torch.tensor(my_times, dtype=torch.float64,
+                             device=self.flat_grad.device)
+            out = torch.zeros(self.world * len(my_times),
+                              dtype=torch.float64, device=t.device)
+            dist.all_gather_into_tensor(out, t, group=self.group)
+            allt = out.view(self.world, -1).tolist()
+        else:
+            allt = [my_times]
+        for w, row in enumerate(allt):
+            for s, tm in zip(steps, row):
+                self._compute_times.append((tm, w, s))
+        self._maybe_log_cdf(steps[-1], force=True)
 
     def compute_time_percentiles(self):
         """Percentile stats over collected per-rank compute times
