@@ -383,6 +383,11 @@ class SyncEngine:
             return
         if not force and step % self.cdf_log_every != 0:
             return
+        if not log.isEnabledFor(logging.INFO):
+            # the report sorts + str()ifies the WHOLE history (reference
+            # semantics, timeout_manager.py:63-70) — skip the O(N log N)
+            # formatting when nothing consumes it (bench runs)
+            return
         elapsed = sorted((t, w, i) for (t, w, i) in self._compute_times
                          if i > self.cdf_start_tracking)
         starts = [t for (s, t) in self._iter_start_times
