@@ -148,9 +148,19 @@ class SyncEngine:
         if self.mode == "interval":
             return self._reduce_interval(step)
 
+        if self.mode == "cdf":
+            # instrumentation first (the per-rank time gather), then the
+            # same fast reduce as full_sync below: cdf never drops a
+            # contribution, so the flag-buffer staging path (two extra
+            # full-bucket copies) is pure overhead for it
+            times = self._all_gather_times(compute_time_s, step)
+            for w, tm in enumerate(times):
+                self._compute_times.append((tm, w, step))
+            self._maybe_log_cdf(step)
+
         # hot path: pure synchronous all-reduce, no staging copies, no
         # host round-trips — the collective IS the barrier
-        if self.mode == "full_sync" and self.timeout_s is None:
+        if self.mode in ("full_sync", "cdf") and self.timeout_s is None:
             if self.distributed:
                 if self.wire_dtype is None:
                     dist.all_reduce(self.flat_grad, op=dist.ReduceOp.SUM,
@@ -175,12 +185,6 @@ class SyncEngine:
             order = sorted(range(self.world), key=lambda r: (times[r], r))
             if self.rank not in order[:self.K]:
                 contribute = False
-
-        if self.mode == "cdf":
-            times = self._all_gather_times(compute_time_s, step)
-            for w, tm in enumerate(times):
-                self._compute_times.append((tm, w, step))
-            self._maybe_log_cdf(step)
 
         n = self.flat_grad.numel()
         if not self.distributed:
